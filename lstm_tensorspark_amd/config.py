@@ -1,0 +1,191 @@
+"""Typed configuration for the MI355X-native LSTM training engine.
+
+Replaces the reference's spark-submit/argparse CLI flag set (SURVEY.md §5,
+"Config / flag system" row: dataset path, hidden size, layers, epochs,
+learning rate, #partitions -> here: dataclass config + YAML + CLI overrides).
+The five BASELINE.json configs are checked-in presets (PRESETS below).
+"""
+from __future__ import annotations
+
+import dataclasses
+import json
+from dataclasses import dataclass, field
+from typing import Any, Optional
+
+
+@dataclass
+class ModelConfig:
+    input_size: int = 128
+    hidden_size: int = 128
+    num_layers: int = 1
+    bidirectional: bool = False
+    dropout: float = 0.0          # inter-layer dropout (not on recurrent path)
+    head: str = "seq_classifier"  # "seq_classifier" | "classifier" | "lm" | "none"
+    num_classes: int = 128        # for classifier heads
+    vocab_size: int = 0           # for head == "lm" (embedding + vocab projection)
+
+    @property
+    def num_directions(self) -> int:
+        return 2 if self.bidirectional else 1
+
+
+@dataclass
+class TrainConfig:
+    batch_size: int = 32          # per-rank batch size (weak scaling)
+    seq_len: int = 32
+    bptt: int = 0                 # truncated-BPTT window; 0 => full backprop through seq_len
+    lr: float = 0.1
+    momentum: float = 0.9
+    weight_decay: float = 0.0
+    optimizer: str = "sgd"        # "sgd" | "adam"
+    adam_betas: tuple = (0.9, 0.999)
+    adam_eps: float = 1e-8
+    grad_clip: float = 0.0        # global-norm clip; 0 => off
+    dtype: str = "fp32"           # "fp32" | "bf16" (bf16 compute + fp32 master weights)
+    steps: int = 100
+    log_interval: int = 10
+    ckpt_interval: int = 0        # 0 => only at end if ckpt_dir set
+    ckpt_dir: str = ""
+    seed: int = 1234
+
+
+@dataclass
+class DistConfig:
+    backend: str = "auto"         # "auto" | "gloo" | "nccl" (nccl IS RCCL on ROCm)
+    bucket_mb: int = 16           # gradient all-reduce bucket size, sized for xGMI links
+
+
+@dataclass
+class DataConfig:
+    task: str = "random"          # "random" | "copy" | "lm"
+    copy_delay: int = 4           # for the learnable copy task
+
+
+@dataclass
+class Config:
+    model: ModelConfig = field(default_factory=ModelConfig)
+    train: TrainConfig = field(default_factory=TrainConfig)
+    dist: DistConfig = field(default_factory=DistConfig)
+    data: DataConfig = field(default_factory=DataConfig)
+
+    # ------------------------------------------------------------------ io
+    def to_dict(self) -> dict:
+        return dataclasses.asdict(self)
+
+    @classmethod
+    def from_dict(cls, d: dict) -> "Config":
+        def sub(klass, key):
+            kw = dict(d.get(key, {}) or {})
+            fields = {f.name for f in dataclasses.fields(klass)}
+            unknown = set(kw) - fields
+            if unknown:
+                raise ValueError(f"unknown {key} config keys: {sorted(unknown)}")
+            if klass is TrainConfig and "adam_betas" in kw:
+                kw["adam_betas"] = tuple(kw["adam_betas"])
+            return klass(**kw)
+
+        unknown = set(d) - {"model", "train", "dist", "data"}
+        if unknown:
+            raise ValueError(f"unknown config sections: {sorted(unknown)}")
+        return cls(
+            model=sub(ModelConfig, "model"),
+            train=sub(TrainConfig, "train"),
+            dist=sub(DistConfig, "dist"),
+            data=sub(DataConfig, "data"),
+        )
+
+    def to_json(self) -> str:
+        return json.dumps(self.to_dict(), indent=2)
+
+    @classmethod
+    def from_yaml(cls, path: str) -> "Config":
+        import yaml
+
+        with open(path) as f:
+            return cls.from_dict(yaml.safe_load(f) or {})
+
+    def apply_overrides(self, overrides: list) -> "Config":
+        """Apply 'section.key=value' CLI overrides, returning a new Config."""
+        d = self.to_dict()
+        for ov in overrides:
+            key, _, val = ov.partition("=")
+            sec, _, name = key.partition(".")
+            if sec not in d or name not in d[sec]:
+                raise ValueError(f"unknown config override: {ov}")
+            cur = d[sec][name]
+            if isinstance(cur, bool):
+                d[sec][name] = val.lower() in ("1", "true", "yes")
+            elif isinstance(cur, int):
+                d[sec][name] = int(val)
+            elif isinstance(cur, float):
+                d[sec][name] = float(val)
+            elif isinstance(cur, (list, tuple)):
+                d[sec][name] = json.loads(val)
+            else:
+                d[sec][name] = val
+        return Config.from_dict(d)
+
+
+# ---------------------------------------------------------------------------
+# The five BASELINE.json configs as presets (BASELINE.md measurement ladder).
+# ---------------------------------------------------------------------------
+def _preset0() -> Config:
+    """configs[0]: 1-layer LSTM h=128 seq=32 on CPU, world_size=1 (plumbing)."""
+    return Config(
+        model=ModelConfig(input_size=128, hidden_size=128, num_layers=1),
+        train=TrainConfig(batch_size=32, seq_len=32, dtype="fp32", lr=0.05),
+    )
+
+
+def _preset1() -> Config:
+    """configs[1]: 2-layer LSTM h=1024 seq=128 bf16 on one MI355X."""
+    return Config(
+        model=ModelConfig(input_size=1024, hidden_size=1024, num_layers=2,
+                          num_classes=128),
+        train=TrainConfig(batch_size=1024, seq_len=128, dtype="bf16", lr=0.05),
+    )
+
+
+def _preset2() -> Config:
+    """configs[2]: 2-layer LSTM h=1024 seq=128, DP over RCCL/xGMI."""
+    c = _preset1()
+    c.dist.backend = "nccl"
+    c.dist.bucket_mb = 16
+    return c
+
+
+def _preset3() -> Config:
+    """configs[3]: 4-layer bidirectional LSTM h=2048 seq=512 DP=8."""
+    return Config(
+        model=ModelConfig(input_size=2048, hidden_size=2048, num_layers=4,
+                          bidirectional=True, dropout=0.1, num_classes=128),
+        train=TrainConfig(batch_size=128, seq_len=512, dtype="bf16", lr=0.05),
+        dist=DistConfig(backend="nccl", bucket_mb=32),
+    )
+
+
+def _preset4() -> Config:
+    """configs[4]: LSTM-LM h=4096 vocab=50k, truncated BPTT=256, grad-clip, DP=8."""
+    return Config(
+        model=ModelConfig(input_size=4096, hidden_size=4096, num_layers=2,
+                          head="lm", vocab_size=50000, dropout=0.1),
+        train=TrainConfig(batch_size=64, seq_len=512, bptt=256, dtype="bf16",
+                          lr=1e-3, optimizer="adam", grad_clip=1.0),
+        dist=DistConfig(backend="nccl", bucket_mb=32),
+        data=DataConfig(task="lm"),
+    )
+
+
+PRESETS = {
+    "plumbing-cpu": _preset0,
+    "mi355x-1gpu": _preset1,
+    "mi355x-dp8": _preset2,
+    "bidir-4l": _preset3,
+    "lm-50k": _preset4,
+}
+
+
+def get_preset(name: str) -> Config:
+    if name not in PRESETS:
+        raise KeyError(f"unknown preset {name!r}; available: {sorted(PRESETS)}")
+    return PRESETS[name]()

@@ -1,0 +1,90 @@
+"""Synthetic sequence data, generated per-rank from a rank-seeded RNG.
+
+Replaces the reference's Spark RDD dataset handling (SURVEY.md §2.1 "Dataset
+handling": file reader + sc.parallelize into partitions). There is no
+network in this environment, so data is synthetic by contract
+(BASELINE.json: "synthetic sequence data / random-init weights"); the
+per-rank generator plays the role of one RDD partition: deterministic,
+disjoint-by-seed, and resumable (the generator state is (seed, step)).
+"""
+from __future__ import annotations
+
+import torch
+
+from .config import Config
+
+
+class SyntheticSeqData:
+    """Deterministic per-rank synthetic batches.
+
+    Each (rank, step) pair maps to a unique RNG stream so that
+    - two ranks never see the same batch (data parallelism is real work),
+    - resume from checkpoint regenerates the identical batch sequence,
+    - tests can reproduce any batch in isolation.
+    """
+
+    def __init__(self, cfg: Config, rank: int = 0, world_size: int = 1,
+                 device: str = "cpu", dtype: torch.dtype = torch.float32):
+        self.cfg = cfg
+        self.rank = rank
+        self.world_size = world_size
+        self.device = torch.device(device)
+        self.dtype = dtype
+        self.task = cfg.data.task
+        m, t = cfg.model, cfg.train
+        self.batch_size = t.batch_size
+        self.seq_len = t.seq_len
+        self.input_size = m.input_size
+        self.base_seed = t.seed
+
+    def _gen(self, step: int) -> torch.Generator:
+        g = torch.Generator(device="cpu")
+        # distinct stream per (seed, rank, step)
+        g.manual_seed(self.base_seed * 1000003 + self.rank * 7919 + step)
+        return g
+
+    def batch(self, step: int):
+        """Return (x, y) for a training step.
+
+        x: (B, T, input_size) float for feature tasks, (B, T) long for "lm".
+        y: per-head target -- see models/heads.
+        """
+        g = self._gen(step)
+        B, T = self.batch_size, self.seq_len
+        task = self.task
+        if task == "random":
+            x = torch.randn(B, T, self.input_size, generator=g)
+            head = self.cfg.model.head
+            if head == "classifier":
+                y = torch.randint(0, self.cfg.model.num_classes, (B,), generator=g)
+            elif head == "seq_classifier":
+                y = torch.randint(0, self.cfg.model.num_classes, (B, T), generator=g)
+            else:
+                y = torch.randn(B, T, self.cfg.model.hidden_size, generator=g)
+            return (x.to(self.device, self.dtype), y.to(self.device))
+        if task == "copy":
+            # Learnable task: predict the input class seen `delay` steps ago.
+            # Input: one-hot of a random class sequence; target: the shifted
+            # sequence (first `delay` targets are class 0 = "blank").
+            C = self.cfg.model.num_classes
+            d = self.cfg.data.copy_delay
+            toks = torch.randint(1, min(C, self.input_size), (B, T), generator=g)
+            x = torch.zeros(B, T, self.input_size)
+            x.scatter_(2, toks.unsqueeze(-1), 1.0)
+            y = torch.zeros(B, T, dtype=torch.long)
+            if d < T:
+                y[:, d:] = toks[:, : T - d]
+            return (x.to(self.device, self.dtype), y.to(self.device))
+        if task == "lm":
+            V = self.cfg.model.vocab_size
+            assert V > 1, "lm task requires model.vocab_size"
+            stream = torch.randint(0, V, (B, T + 1), generator=g)
+            x, y = stream[:, :-1], stream[:, 1:]
+            return (x.to(self.device), y.to(self.device))
+        raise ValueError(f"unknown data task {task!r}")
+
+    def __iter__(self):
+        step = 0
+        while True:
+            yield self.batch(step)
+            step += 1

@@ -1,0 +1,166 @@
+"""Stacked (optionally bidirectional) LSTM model with task heads.
+
+Replaces the reference's hand-rolled TF-graph LSTM cell/model module
+(SURVEY.md §2.1 "LSTM model / cell": weights 4 gates x {W_x, W_h, b},
+stepped over the sequence, truncated BPTT, loss, optimizer update).
+
+Weight naming follows the checkpoint layout contract (SURVEY.md §5
+Checkpoint row): a flat named-weight dict with keys
+    layer{i}/W_x, layer{i}/W_h, layer{i}/b           (forward direction)
+    layer{i}_rev/W_x, ...                            (reverse direction)
+    embed/W, head/W, head/b                          (task head)
+Gate order along the 4H axis is blocked [i|f|g|o].
+"""
+from __future__ import annotations
+
+import math
+from typing import Optional
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from ..config import ModelConfig
+from ..ops.lstm_layer import lstm_layer
+
+
+def _pname(layer: int, rev: bool, which: str) -> str:
+    return f"layer{layer}{'_rev' if rev else ''}/{which}"
+
+
+class LSTMModel(nn.Module):
+    """Multi-layer (bi)LSTM + head. Compute dtype follows parameter dtype;
+    cell state is always fp32."""
+
+    def __init__(self, cfg: ModelConfig):
+        super().__init__()
+        self.cfg = cfg
+        H = cfg.hidden_size
+        D = cfg.num_directions
+        self._params: dict[str, nn.Parameter] = {}
+
+        if cfg.head == "lm":
+            assert cfg.vocab_size > 0
+            self._register("embed/W", torch.empty(cfg.vocab_size, cfg.input_size))
+
+        for l in range(cfg.num_layers):
+            in_size = cfg.input_size if l == 0 else H * D
+            for rev in ([False, True] if cfg.bidirectional else [False]):
+                self._register(_pname(l, rev, "W_x"), torch.empty(in_size, 4 * H))
+                self._register(_pname(l, rev, "W_h"), torch.empty(H, 4 * H))
+                self._register(_pname(l, rev, "b"), torch.empty(4 * H))
+
+        feat = H * D
+        if cfg.head in ("classifier", "seq_classifier"):
+            self._register("head/W", torch.empty(feat, cfg.num_classes))
+            self._register("head/b", torch.empty(cfg.num_classes))
+        elif cfg.head == "lm":
+            self._register("head/W", torch.empty(feat, cfg.vocab_size))
+            self._register("head/b", torch.empty(cfg.vocab_size))
+        self.reset_parameters()
+
+    # -- flat named-weight dict plumbing ----------------------------------
+    def _register(self, name: str, t: torch.Tensor):
+        p = nn.Parameter(t)
+        self.register_parameter(name.replace("/", "__"), p)
+        self._params[name] = p
+
+    def weight(self, name: str) -> nn.Parameter:
+        return self._params[name]
+
+    def weight_dict(self) -> dict[str, nn.Parameter]:
+        """The flat named-weight dict (checkpoint layout contract)."""
+        return dict(self._params)
+
+    def reset_parameters(self, seed: Optional[int] = None):
+        g = None
+        if seed is not None:
+            g = torch.Generator(device="cpu")
+            g.manual_seed(seed)
+        H = self.cfg.hidden_size
+        std = 1.0 / math.sqrt(H)
+        with torch.no_grad():
+            for name, p in self._params.items():
+                if name == "embed/W":
+                    src = torch.randn(p.shape, generator=g) * 0.02
+                elif name.endswith("/b"):
+                    src = torch.zeros(p.shape)
+                    if name.startswith("layer"):
+                        # forget-gate bias = 1 (standard LSTM practice)
+                        src[H : 2 * H] = 1.0
+                else:
+                    src = (torch.rand(p.shape, generator=g) * 2 - 1) * std
+                p.copy_(src)
+
+    # -- forward -----------------------------------------------------------
+    def _layer_io(self, x, layer: int, rev: bool, state=None):
+        B = x.shape[0]
+        H = self.cfg.hidden_size
+        w_x = self.weight(_pname(layer, rev, "W_x"))
+        w_h = self.weight(_pname(layer, rev, "W_h"))
+        b = self.weight(_pname(layer, rev, "b"))
+        if state is None:
+            h0 = torch.zeros(B, H, dtype=x.dtype, device=x.device)
+            c0 = torch.zeros(B, H, dtype=torch.float32, device=x.device)
+        else:
+            h0, c0 = state
+        return lstm_layer(x, h0, c0, w_x, w_h, b, reverse=rev)
+
+    def backbone(self, x: torch.Tensor, state: Optional[list] = None):
+        """Run embedding (if LM) + all LSTM layers.
+
+        x: (B,T,input_size) features, or (B,T) long tokens for LM.
+        state: per-(layer,dir) list of (h0, c0) carry (truncated BPTT), or
+               None for zero state.
+        Returns (features (B,T,H*D), new_state list).
+        """
+        cfg = self.cfg
+        if cfg.head == "lm":
+            x = F.embedding(x, self.weight("embed/W"))
+        new_state = []
+        si = 0
+        for l in range(cfg.num_layers):
+            outs = []
+            for rev in ([False, True] if cfg.bidirectional else [False]):
+                st = state[si] if state is not None else None
+                h_seq, hT, cT = self._layer_io(x, l, rev, st)
+                outs.append(h_seq)
+                new_state.append((hT, cT))
+                si += 1
+            x = outs[0] if len(outs) == 1 else torch.cat(outs, dim=2)
+            if cfg.dropout > 0 and self.training and l < cfg.num_layers - 1:
+                x = F.dropout(x, p=cfg.dropout)
+        return x, new_state
+
+    def forward(self, x, state: Optional[list] = None):
+        """Returns (output, new_state).
+
+        output: logits -- (B,C) for "classifier", (B,T,C) for
+        "seq_classifier", (B,T,V) for "lm", features (B,T,H*D) for "none".
+        """
+        feats, new_state = self.backbone(x, state)
+        cfg = self.cfg
+        if cfg.head == "none":
+            return feats, new_state
+        w, b = self.weight("head/W"), self.weight("head/b")
+        if cfg.head == "classifier":
+            return torch.addmm(b, feats[:, -1], w), new_state
+        B, T, F_ = feats.shape
+        logits = torch.addmm(b, feats.reshape(B * T, F_), w).view(B, T, -1)
+        return logits, new_state
+
+    def loss(self, output: torch.Tensor, y: torch.Tensor) -> torch.Tensor:
+        head = self.cfg.head
+        if head == "classifier":
+            return F.cross_entropy(output.float(), y)
+        if head in ("seq_classifier", "lm"):
+            C = output.shape[-1]
+            return F.cross_entropy(output.reshape(-1, C).float(), y.reshape(-1))
+        # regression on features
+        return F.mse_loss(output.float(), y.float())
+
+    def detach_state(self, state):
+        """Detach the (h, c) carry for truncated BPTT."""
+        if state is None:
+            return None
+        return [(h.detach(), c.detach()) for h, c in state]

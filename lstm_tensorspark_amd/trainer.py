@@ -1,0 +1,172 @@
+"""Trainer: the train()/fit() API of the engine.
+
+Replaces the reference's driver epoch loop: the Spark
+distribute->train->aggregate round becomes, per step and per rank,
+
+    synth batch (rank-seeded)                      [data.py]
+    forward (fused HIP kernels on GPU)             [ops/lstm_layer.py]
+    backward -> bucketed async all-reduce over RCCL [parallel/ddp.py]
+    grad clip (K6) -> fused optimizer (K7)         [ops/optim.py]
+    rank0: log / checkpoint at interval
+
+Truncated BPTT (SURVEY.md §5 long-context row): when train.bptt > 0 and
+< seq_len, each batch is streamed in windows of bptt steps with the (h, c)
+carry detached between windows -- activation memory is O(bptt*B*H), not
+O(T*B*H).
+"""
+from __future__ import annotations
+
+import json
+import os
+import time
+from typing import Optional
+
+import torch
+import torch.distributed as dist
+
+from .checkpoint import load_checkpoint, save_checkpoint
+from .config import Config
+from .data import SyntheticSeqData
+from .models.lstm import LSTMModel
+from .ops.optim import build_optimizer
+from .parallel.ddp import GradBucketer, broadcast_parameters
+from .parallel.launch import DistContext
+from .utils.logging import get_logger
+
+
+def compute_dtype(cfg: Config, device: torch.device) -> torch.dtype:
+    if cfg.train.dtype == "bf16" and device.type == "cuda":
+        return torch.bfloat16
+    return torch.float32
+
+
+class Trainer:
+    def __init__(self, cfg: Config, model: Optional[LSTMModel] = None,
+                 dist_ctx: Optional[DistContext] = None, device=None):
+        self.cfg = cfg
+        self.ctx = dist_ctx or DistContext()
+        self.device = torch.device(device) if device is not None else self.ctx.device
+        self.dtype = compute_dtype(cfg, self.device)
+        self.log = get_logger(rank=self.ctx.rank)
+
+        torch.manual_seed(cfg.train.seed)  # same init on every rank
+        self.model = model if model is not None else LSTMModel(cfg.model)
+        self.model = self.model.to(self.device)
+        if self.dtype != torch.float32:
+            self.model = self.model.to(self.dtype)
+        broadcast_parameters(self.model)  # belt-and-braces weight sync at init
+
+        self.optimizer = build_optimizer(cfg.train, self.model.parameters())
+        self.bucketer = None
+        if self.ctx.world_size > 1:
+            self.bucketer = GradBucketer(self.model.parameters(),
+                                         bucket_mb=cfg.dist.bucket_mb)
+        self.data = SyntheticSeqData(cfg, rank=self.ctx.rank,
+                                     world_size=self.ctx.world_size,
+                                     device=self.device, dtype=self.dtype)
+        self.step = 0
+        self.history: list[dict] = []
+
+    # ------------------------------------------------------------------
+    def _bptt_windows(self, x, y):
+        """Split (x, y) along time into truncated-BPTT windows."""
+        T = x.shape[1]
+        k = self.cfg.train.bptt
+        if k <= 0 or k >= T:
+            return [(x, y)]
+        per_step_y = y.dim() >= 2 and y.shape[1] == T
+        wins = []
+        for s in range(0, T, k):
+            e = min(s + k, T)
+            wins.append((x[:, s:e], y[:, s:e] if per_step_y else y))
+        return wins
+
+    def train_step(self, batch) -> dict:
+        cfg = self.cfg
+        x, y = batch
+        if self.bucketer is not None:
+            self.bucketer.zero_grad()
+        else:
+            self.optimizer.zero_grad()
+
+        total_loss = 0.0
+        wins = self._bptt_windows(x, y)
+        state = None
+        for wi, (wx, wy) in enumerate(wins):
+            out, state = self.model(wx, state)
+            state = self.model.detach_state(state)
+            loss = self.model.loss(out, wy)
+            if len(wins) > 1:
+                loss = loss / len(wins)
+            if self.bucketer is not None:
+                # grads accumulate across windows into the bucket buffers;
+                # only the LAST window's backward launches the all-reduces.
+                self.bucketer.start_backward(will_reduce=(wi == len(wins) - 1))
+            loss.backward()
+            total_loss += float(loss.detach())
+        if self.bucketer is not None:
+            self.bucketer.finalize()
+
+        norm = None
+        if cfg.train.grad_clip > 0:
+            norm = self.optimizer.clip_grad_norm(cfg.train.grad_clip)
+        self.optimizer.step()
+        return {"loss": total_loss,
+                "grad_norm": float(norm) if norm is not None else None}
+
+    def fit(self, steps: Optional[int] = None) -> list:
+        """Train for `steps` optimizer steps (default: config.train.steps)."""
+        cfg = self.cfg
+        steps = steps if steps is not None else cfg.train.steps
+        t_last = time.perf_counter()
+        seqs_since = 0
+        start_step = self.step
+        while self.step < start_step + steps:
+            batch = self.data.batch(self.step)
+            info = self.train_step(batch)
+            self.step += 1
+            seqs_since += cfg.train.batch_size * self.ctx.world_size
+            if cfg.train.log_interval and self.step % cfg.train.log_interval == 0:
+                if self.device.type == "cuda":
+                    torch.cuda.synchronize()
+                now = time.perf_counter()
+                rec = {
+                    "step": self.step,
+                    "loss": round(info["loss"], 6),
+                    "grad_norm": info["grad_norm"],
+                    "seq_per_sec": round(seqs_since / (now - t_last), 2),
+                    "ms_per_step": round(1e3 * (now - t_last)
+                                         / max(1, cfg.train.log_interval), 3),
+                }
+                self.history.append(rec)
+                if self.ctx.is_main:
+                    self.log.info(json.dumps(rec))
+                t_last, seqs_since = now, 0
+            if (cfg.train.ckpt_interval and cfg.train.ckpt_dir
+                    and self.step % cfg.train.ckpt_interval == 0):
+                self.save(os.path.join(cfg.train.ckpt_dir,
+                                       f"ckpt_step{self.step}.pt"))
+        if cfg.train.ckpt_dir:
+            self.save(os.path.join(cfg.train.ckpt_dir, "ckpt_final.pt"))
+        return self.history
+
+    # alias matching the reference's API surface
+    train = fit
+
+    # ------------------------------------------------------------------
+    def save(self, path: str):
+        if not self.ctx.is_main:
+            if dist.is_initialized():
+                dist.barrier()
+            return
+        save_checkpoint(path, self.model, self.optimizer, step=self.step,
+                        config=self.cfg.to_dict())
+        self.log.info(f"checkpoint saved: {path} (step {self.step})")
+        if dist.is_initialized():
+            dist.barrier()
+
+    def load(self, path: str):
+        payload = load_checkpoint(path, self.model, self.optimizer)
+        self.step = payload["step"]
+        self.log.info(f"checkpoint loaded: {path} (step {self.step})")
+        return payload

@@ -1,0 +1,88 @@
+"""Checkpoint layout + bitwise save->load->resume (SURVEY.md §4.2 plumbing)."""
+import os
+
+import torch
+
+from lstm_tensorspark_amd.config import Config, ModelConfig, TrainConfig, DataConfig
+from lstm_tensorspark_amd.checkpoint import load_checkpoint, save_checkpoint
+from lstm_tensorspark_amd.trainer import Trainer
+
+
+def small_cfg(tmp, **train_kw):
+    kw = dict(batch_size=4, seq_len=8, lr=0.05, momentum=0.9, steps=4,
+              log_interval=0, ckpt_dir=str(tmp))
+    kw.update(train_kw)
+    return Config(
+        model=ModelConfig(input_size=12, hidden_size=10, num_layers=2,
+                          num_classes=7),
+        train=TrainConfig(**kw),
+        data=DataConfig(task="copy"),
+    )
+
+
+def test_layout_flat_named_weight_dict(tmp_path):
+    cfg = small_cfg(tmp_path)
+    t = Trainer(cfg)
+    path = os.path.join(str(tmp_path), "c.pt")
+    save_checkpoint(path, t.model, t.optimizer, step=3, config=cfg.to_dict())
+    payload = torch.load(path, map_location="cpu", weights_only=False)
+    keys = set(payload["weights"])
+    assert {"layer0/W_x", "layer0/W_h", "layer0/b",
+            "layer1/W_x", "layer1/W_h", "layer1/b",
+            "head/W", "head/b"} == keys
+    assert payload["step"] == 3
+    assert payload["optimizer"] is not None
+    assert "rng" in payload and "torch" in payload["rng"]
+    assert payload["config"]["model"]["hidden_size"] == 10
+
+
+def test_bitwise_resume(tmp_path):
+    # Run A: 3 steps, save, 3 more steps.
+    cfg = small_cfg(tmp_path)
+    tA = Trainer(cfg)
+    tA.fit(3)
+    mid = os.path.join(str(tmp_path), "mid.pt")
+    tA.save(mid)
+    tA.fit(3)
+    wA = {k: v.detach().clone() for k, v in tA.model.weight_dict().items()}
+
+    # Run B: fresh trainer, load the mid checkpoint, 3 steps.
+    tB = Trainer(small_cfg(tmp_path))
+    tB.load(mid)
+    assert tB.step == 3
+    tB.fit(3)
+    for k, vA in wA.items():
+        vB = tB.model.weight_dict()[k]
+        assert torch.equal(vA, vB), f"resume not bitwise for {k}"
+    # optimizer masters identical too
+    for mA, mB in zip(tA.optimizer.masters, tB.optimizer.masters):
+        assert torch.equal(mA, mB)
+
+
+def test_bitwise_resume_adam_bptt(tmp_path):
+    cfg = small_cfg(tmp_path, optimizer="adam", lr=1e-3, bptt=3, seq_len=8,
+                    grad_clip=1.0)
+    tA = Trainer(cfg)
+    tA.fit(2)
+    mid = os.path.join(str(tmp_path), "mid.pt")
+    tA.save(mid)
+    tA.fit(2)
+    tB = Trainer(small_cfg(tmp_path, optimizer="adam", lr=1e-3, bptt=3,
+                           seq_len=8, grad_clip=1.0))
+    tB.load(mid)
+    tB.fit(2)
+    for k, vA in tA.model.weight_dict().items():
+        assert torch.equal(vA, tB.model.weight_dict()[k]), k
+
+
+def test_mismatched_names_rejected(tmp_path):
+    cfg = small_cfg(tmp_path)
+    t = Trainer(cfg)
+    p = os.path.join(str(tmp_path), "c.pt")
+    save_checkpoint(p, t.model)
+    cfg2 = small_cfg(tmp_path)
+    cfg2.model.num_layers = 1
+    t2 = Trainer(cfg2)
+    import pytest
+    with pytest.raises(ValueError):
+        load_checkpoint(p, t2.model)

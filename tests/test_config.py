@@ -1,0 +1,60 @@
+import json
+
+import pytest
+
+from lstm_tensorspark_amd.config import Config, PRESETS, get_preset
+
+
+def test_default_roundtrip():
+    c = Config()
+    d = c.to_dict()
+    c2 = Config.from_dict(d)
+    assert c2.to_dict() == d
+
+
+def test_presets_cover_baseline_configs():
+    # one preset per BASELINE.json config
+    assert len(PRESETS) == 5
+    p1 = get_preset("mi355x-1gpu")
+    assert p1.model.hidden_size == 1024
+    assert p1.model.num_layers == 2
+    assert p1.train.seq_len == 128
+    assert p1.train.dtype == "bf16"
+    p3 = get_preset("bidir-4l")
+    assert p3.model.bidirectional and p3.model.num_layers == 4
+    assert p3.model.hidden_size == 2048 and p3.train.seq_len == 512
+    p4 = get_preset("lm-50k")
+    assert p4.model.head == "lm" and p4.model.vocab_size == 50000
+    assert p4.train.bptt == 256 and p4.train.grad_clip > 0
+    assert p4.model.hidden_size == 4096
+
+
+def test_overrides():
+    c = Config().apply_overrides(
+        ["train.lr=0.5", "model.hidden_size=64", "model.bidirectional=true",
+         "dist.backend=gloo"])
+    assert c.train.lr == 0.5
+    assert c.model.hidden_size == 64
+    assert c.model.bidirectional is True
+    assert c.dist.backend == "gloo"
+
+
+def test_bad_override_rejected():
+    with pytest.raises(ValueError):
+        Config().apply_overrides(["train.nope=1"])
+
+
+def test_unknown_key_rejected():
+    with pytest.raises(ValueError):
+        Config.from_dict({"model": {"hidden_sizee": 8}})
+
+
+def test_yaml_load(tmp_path):
+    p = tmp_path / "c.yaml"
+    p.write_text("model:\n  hidden_size: 96\ntrain:\n  seq_len: 17\n")
+    c = Config.from_yaml(str(p))
+    assert c.model.hidden_size == 96 and c.train.seq_len == 17
+
+
+def test_json_serializable():
+    json.loads(Config().to_json())
