@@ -1,0 +1,279 @@
+// C++ sequence drivers + pybind bindings for the MI355X LSTM kernels.
+//
+// The time loops run HERE (one Python call per layer per direction), not in
+// Python: at ~3.5us host launch overhead per kernel, a 128-step recurrent
+// loop must not round-trip through the interpreter per step (SURVEY.md §7
+// hard part #1).  Library GEMMs (recurrent dh, and the batched dW/dx in
+// ops/lstm_layer.py) go through rocBLAS/hipBLASLt via at::mm.
+
+#include <torch/extension.h>
+#include <ATen/ATen.h>
+#include <c10/hip/HIPStream.h>
+#include <hip/hip_runtime.h>
+
+#include <vector>
+
+extern "C" {
+void launch_lstm_rec_fwd(const void*, long, const void*, const void*, long,
+                         const float*, long, void*, long, float*, long,
+                         void*, long, int, int, int, hipStream_t);
+void launch_lstm_pw_bwd(const void*, long, const void*, long, const void*,
+                        long, const float*, long, const float*, long, void*,
+                        long, float*, long, int, int, int, const float*, long,
+                        hipStream_t);
+void launch_l2norm(const void*, long, float*, int, hipStream_t);
+void launch_clip_scale(void*, long, const float*, float, int, hipStream_t);
+void launch_sgd(void*, float*, const void*, float*, long, float, float, float,
+                int, hipStream_t);
+void launch_adam(void*, float*, const void*, float*, float*, long, float,
+                 float, float, float, float, float, float, int, hipStream_t);
+}
+
+namespace {
+
+using at::Tensor;
+
+hipStream_t cur_stream() {
+    return c10::hip::getCurrentHIPStream().stream();
+}
+
+void check_bf16_cuda(const Tensor& t, const char* name) {
+    TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
+    TORCH_CHECK(t.scalar_type() == at::kBFloat16, name, " must be bf16");
+}
+
+// ---------------------------------------------------------------------
+// Forward: whole-sequence driver.  pre (B,T,4H) bf16, w_h (H,4H) bf16,
+// h0 (B,H) bf16, c0 (B,H) fp32 -> (h_seq, c_seq, acts, hT, cT)
+// ---------------------------------------------------------------------
+std::vector<Tensor> lstm_seq_fwd(Tensor pre, Tensor w_h, Tensor h0, Tensor c0,
+                                 bool reverse) {
+    check_bf16_cuda(pre, "pre");
+    check_bf16_cuda(w_h, "w_h");
+    check_bf16_cuda(h0, "h0");
+    TORCH_CHECK(c0.scalar_type() == at::kFloat && c0.is_cuda(), "c0 fp32 GPU");
+    TORCH_CHECK(pre.dim() == 3, "pre must be (B,T,4H)");
+    const long B = pre.size(0), T = pre.size(1), H4 = pre.size(2);
+    const long H = H4 / 4;
+    TORCH_CHECK(w_h.size(0) == H && w_h.size(1) == H4, "w_h must be (H,4H)");
+    TORCH_CHECK(H % 64 == 0, "fused kernel requires H % 64 == 0, got ", H);
+    pre = pre.contiguous();
+    auto h0c = h0.contiguous();
+    auto c0c = c0.contiguous();
+    // W_h^T (4H, H): one transposed copy per layer call so each lane's MFMA
+    // B-fragment is a contiguous 16-byte load.
+    auto w_t = w_h.t().contiguous();
+
+    auto opts_bf = pre.options();
+    auto opts_f32 = pre.options().dtype(at::kFloat);
+    auto h_seq = at::empty({B, T, H}, opts_bf);
+    auto c_seq = at::empty({B, T, H}, opts_f32);
+    auto acts = at::empty({B, T, H4}, opts_bf);
+
+    auto* pre_p = (const char*)pre.data_ptr();
+    auto* hs_p = (char*)h_seq.data_ptr();
+    auto* cs_p = (float*)c_seq.data_ptr();
+    auto* act_p = (char*)acts.data_ptr();
+    auto stream = cur_stream();
+
+    const void* hp = h0c.data_ptr();
+    long hp_stride = H;
+    const float* cp = (const float*)c0c.data_ptr();
+    long cp_stride = H;
+    const long hs_stride = T * H, cs_stride = T * H;
+    const long pre_stride = T * H4, act_stride = T * H4;
+
+    for (long k = 0; k < T; ++k) {
+        const long t = reverse ? (T - 1 - k) : k;
+        launch_lstm_rec_fwd(
+            hp, hp_stride, w_t.data_ptr(),
+            pre_p + 2 * t * H4, pre_stride,  // bf16 = 2 bytes
+            cp, cp_stride,
+            hs_p + 2 * t * H, hs_stride,
+            cs_p + t * H, cs_stride,
+            act_p + 2 * t * H4, act_stride,
+            (int)B, (int)H, /*raw_debug=*/0, stream);
+        hp = hs_p + 2 * t * H;
+        hp_stride = hs_stride;
+        cp = cs_p + t * H;
+        cp_stride = cs_stride;
+    }
+    const long last = reverse ? 0 : T - 1;
+    auto hT = h_seq.select(1, last).clone();
+    auto cT = c_seq.select(1, last).clone();
+    return {h_seq, c_seq, acts, hT, cT};
+}
+
+// Single fused step with optional raw-gate debug output (numerics tests).
+std::vector<Tensor> lstm_rec_fwd_step(Tensor h_prev, Tensor w_h, Tensor pre_t,
+                                      Tensor c_prev, bool raw_debug) {
+    check_bf16_cuda(h_prev, "h_prev");
+    const long B = pre_t.size(0), H4 = pre_t.size(1), H = H4 / 4;
+    TORCH_CHECK(H % 64 == 0, "H % 64 != 0");
+    auto w_t = w_h.t().contiguous();
+    auto hp = h_prev.contiguous();
+    auto pr = pre_t.contiguous();
+    auto cp = c_prev.contiguous();
+    auto h_out = at::empty({B, H}, pre_t.options());
+    auto c_out = at::empty({B, H}, pre_t.options().dtype(at::kFloat));
+    auto acts = at::empty({B, H4}, pre_t.options());
+    launch_lstm_rec_fwd(hp.data_ptr(), H, w_t.data_ptr(), pr.data_ptr(), H4,
+                        (const float*)cp.data_ptr(), H, h_out.data_ptr(), H,
+                        (float*)c_out.data_ptr(), H, acts.data_ptr(), H4,
+                        (int)B, (int)H, raw_debug ? 1 : 0, cur_stream());
+    return {h_out, c_out, acts};
+}
+
+// ---------------------------------------------------------------------
+// Backward: whole-sequence driver.
+// ---------------------------------------------------------------------
+std::vector<Tensor> lstm_seq_bwd(c10::optional<Tensor> dh_seq_opt,
+                                 c10::optional<Tensor> d_hT_opt,
+                                 c10::optional<Tensor> d_cT_opt,
+                                 Tensor w_h, Tensor acts, Tensor c_seq,
+                                 Tensor c0, bool reverse) {
+    check_bf16_cuda(acts, "acts");
+    const long B = acts.size(0), T = acts.size(1), H4 = acts.size(2);
+    const long H = H4 / 4;
+    TORCH_CHECK(H % 4 == 0, "H % 4 != 0");
+    acts = acts.contiguous();
+    c_seq = c_seq.contiguous();
+    auto c0c = c0.contiguous();
+    Tensor dh_seq;
+    const bool has_up = dh_seq_opt.has_value();
+    if (has_up) dh_seq = dh_seq_opt->contiguous();
+    Tensor d_hT;
+    const bool has_hT = d_hT_opt.has_value();
+    if (has_hT) d_hT = d_hT_opt->contiguous();
+    Tensor d_cT;
+    const bool has_cT = d_cT_opt.has_value();
+    if (has_cT) d_cT = d_cT_opt->contiguous();
+
+    auto dgates = at::empty({B, T, H4}, acts.options());
+    auto dc = at::empty({B, H}, acts.options().dtype(at::kFloat));
+    auto dh_buf = at::empty({B, H}, acts.options());
+    // W_h^T view for the recurrent grad GEMM (rocBLAS handles the transpose)
+    auto w_h_t = w_h.t();
+
+    auto* act_p = (const char*)acts.data_ptr();
+    auto* cs_p = (const float*)c_seq.data_ptr();
+    auto* dg_p = (char*)dgates.data_ptr();
+    auto* up_p = has_up ? (const char*)dh_seq.data_ptr() : nullptr;
+    auto stream = cur_stream();
+    const long act_stride = T * H4, cs_stride = T * H, up_stride = T * H;
+
+    for (long k = T - 1; k >= 0; --k) {
+        const long t = reverse ? (T - 1 - k) : k;
+        const long t_prev = reverse ? (T - k) : (k - 1);  // index of c_{t-1}
+        const bool first = (k == T - 1);
+        const float* cprev_p;
+        long cprev_stride;
+        if (k > 0) {
+            cprev_p = cs_p + t_prev * H;
+            cprev_stride = cs_stride;
+        } else {
+            cprev_p = (const float*)c0c.data_ptr();
+            cprev_stride = H;
+        }
+        const void* carry_p = nullptr;
+        long carry_stride = H;
+        if (first) {
+            if (has_hT) carry_p = d_hT.data_ptr();
+        } else {
+            carry_p = dh_buf.data_ptr();
+        }
+        launch_lstm_pw_bwd(
+            has_up ? up_p + 2 * t * H : nullptr, up_stride,
+            carry_p, carry_stride,
+            act_p + 2 * t * H4, act_stride,
+            cs_p + t * H, cs_stride, cprev_p, cprev_stride,
+            dg_p + 2 * t * H4, act_stride,
+            (float*)dc.data_ptr(), H, (int)B, (int)H,
+            first ? 1 : 0,
+            (first && has_cT) ? (const float*)d_cT.data_ptr() : nullptr, H,
+            stream);
+        // recurrent hidden-grad GEMM [K5a]: dh_{t-1} = dgates_t @ W_h^T
+        at::mm_out(dh_buf, dgates.select(1, t), w_h_t);
+    }
+    return {dgates, dh_buf, dc};  // (dgates, dh0, dc0)
+}
+
+// ---------------------------------------------------------------------
+// K6/K7: multi-tensor clip + optimizers
+// ---------------------------------------------------------------------
+Tensor multi_tensor_l2norm(std::vector<Tensor> grads) {
+    TORCH_CHECK(!grads.empty());
+    auto out = at::zeros({}, grads[0].options().dtype(at::kFloat));
+    auto stream = cur_stream();
+    for (auto& g : grads) {
+        auto gc = g.contiguous();
+        TORCH_CHECK(gc.is_cuda());
+        launch_l2norm(gc.data_ptr(), gc.numel(), (float*)out.data_ptr(),
+                      gc.scalar_type() == at::kBFloat16 ? 1 : 0, stream);
+    }
+    return out.sqrt_();
+}
+
+void multi_tensor_clip_scale(std::vector<Tensor> grads, Tensor norm,
+                             double max_norm) {
+    auto stream = cur_stream();
+    for (auto& g : grads) {
+        TORCH_CHECK(g.is_contiguous(), "grad must be contiguous for clip");
+        launch_clip_scale(g.data_ptr(), g.numel(),
+                          (const float*)norm.data_ptr(), (float)max_norm,
+                          g.scalar_type() == at::kBFloat16 ? 1 : 0, stream);
+    }
+}
+
+void fused_sgd(std::vector<Tensor> params, std::vector<Tensor> masters,
+               std::vector<Tensor> grads, std::vector<Tensor> bufs,
+               double lr, double mom, double wd) {
+    auto stream = cur_stream();
+    const bool has_mom = !bufs.empty();
+    for (size_t i = 0; i < params.size(); ++i) {
+        auto& p = params[i];
+        auto g = grads[i].contiguous();
+        TORCH_CHECK(p.is_contiguous() && masters[i].is_contiguous());
+        TORCH_CHECK(g.numel() == p.numel() && masters[i].numel() == p.numel());
+        TORCH_CHECK(g.scalar_type() == p.scalar_type(), "grad/param dtype mismatch");
+        launch_sgd(p.data_ptr(), (float*)masters[i].data_ptr(), g.data_ptr(),
+                   has_mom ? (float*)bufs[i].data_ptr() : nullptr, p.numel(),
+                   (float)lr, (float)mom, (float)wd,
+                   p.scalar_type() == at::kBFloat16 ? 1 : 0, stream);
+    }
+}
+
+void fused_adam(std::vector<Tensor> params, std::vector<Tensor> masters,
+                std::vector<Tensor> grads, std::vector<Tensor> ea,
+                std::vector<Tensor> eas, double lr, double b1, double b2,
+                double eps, double wd, double bc1, double bc2) {
+    auto stream = cur_stream();
+    for (size_t i = 0; i < params.size(); ++i) {
+        auto& p = params[i];
+        auto g = grads[i].contiguous();
+        TORCH_CHECK(g.scalar_type() == p.scalar_type(), "grad/param dtype mismatch");
+        launch_adam(p.data_ptr(), (float*)masters[i].data_ptr(), g.data_ptr(),
+                    (float*)ea[i].data_ptr(), (float*)eas[i].data_ptr(),
+                    p.numel(), (float)lr, (float)b1, (float)b2, (float)eps,
+                    (float)wd, (float)bc1, (float)bc2,
+                    p.scalar_type() == at::kBFloat16 ? 1 : 0, stream);
+    }
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+    m.doc() = "MI355X (gfx950) HIP kernels for lstm_tensorspark_amd";
+    m.def("lstm_seq_fwd", &lstm_seq_fwd, "fused LSTM sequence forward");
+    m.def("lstm_seq_bwd", &lstm_seq_bwd, "fused LSTM sequence backward",
+          py::arg("dh_seq") = py::none(), py::arg("d_hT") = py::none(),
+          py::arg("d_cT") = py::none(), py::arg("w_h"), py::arg("acts"),
+          py::arg("c_seq"), py::arg("c0"), py::arg("reverse"));
+    m.def("lstm_rec_fwd_step", &lstm_rec_fwd_step,
+          "single fused recurrent step (raw_debug for GEMM-mapping tests)");
+    m.def("multi_tensor_l2norm", &multi_tensor_l2norm);
+    m.def("multi_tensor_clip_scale", &multi_tensor_clip_scale);
+    m.def("fused_sgd", &fused_sgd);
+    m.def("fused_adam", &fused_adam);
+    m.attr("BUILT_FOR") = "gfx950";
+}

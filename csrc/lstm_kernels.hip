@@ -1,0 +1,400 @@
+// MI355X (gfx950 / CDNA4) kernels for the LSTM cell hot path.
+//
+// Replaces the reference's per-timestep TF graph ops:
+//   k_lstm_rec_fwd : fused recurrent-GEMM + pointwise cell update [K1b+K2]
+//                    gates = pre_t + h_{t-1} @ W_h  (MFMA bf16, fp32 accum)
+//                    i,f,o = sigmoid, g = tanh; c_t = f*c + i*g;
+//                    h_t = o*tanh(c_t)  -- one kernel per timestep.
+//   k_lstm_pw_bwd  : fused pointwise backward [K4] -> raw-gate grads dgates
+//   (recurrent dh GEMM [K5a] and the batched dW/dx GEMMs [K5b] go through
+//    rocBLAS/hipBLASLt from the C++ sequence driver in bindings.cpp)
+//   k_l2norm / k_clip_scale [K6], k_sgd / k_adam [K7]: optimizer path.
+//
+// Layout contract (must match lstm_tensorspark_amd/ops/functional.py):
+//   * gate order i,f,g,o BLOCKED along the 4H axis;
+//   * W_h passed TRANSPOSED as (4H, H) row-major so each lane's MFMA
+//     B-fragment (8 k-contiguous bf16) is one contiguous 16-byte load;
+//   * cell state fp32, h/activations bf16, MFMA accumulation fp32.
+//
+// Wavefront size is 64 (CDNA), block = 256 threads = 4 waves.
+
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+
+typedef __bf16 bf16x8 __attribute__((ext_vector_type(8)));
+typedef __bf16 bf16x4 __attribute__((ext_vector_type(4)));
+typedef float f32x4 __attribute__((ext_vector_type(4)));
+
+#define DEVINL __device__ __forceinline__
+
+DEVINL float sigmoidf_(float x) { return 1.0f / (1.0f + __expf(-x)); }
+// tanh via exp2-free formulation: tanhf is fine on VALU (transcendental)
+DEVINL float tanhf_(float x) { return tanhf(x); }
+
+// ---------------------------------------------------------------------------
+// K1b+K2: fused recurrent GEMM + LSTM pointwise, one timestep.
+//
+// Grid: (ceil(B/BM), H/64).  Block: 256 threads = 4 waves; wave w owns
+// h-columns [blockIdx.y*64 + w*16, +16) and computes FOUR 16-col MFMA
+// accumulators (one per gate) over BM rows, so the cell update needs no
+// cross-lane traffic: after the K-loop, lane (l) holds, for each of its
+// 4 fragment rows, the i/f/g/o raw-gate values of ONE (row, h-column).
+//
+// MFMA: v_mfma_f32_16x16x32_bf16.  A-fragment: lane l = h_prev[row=l&15]
+// [k=(l>>4)*8 ..+8]; B-fragment: w_t[n=l&15][same k] with w_t=(4H,H)=W_h^T;
+// C/D: col=l&15, row=(l>>4)*4+r.
+// TAIL HANDLING: rows (batch) bounds-checked; requires H % 64 == 0.
+// ---------------------------------------------------------------------------
+template <int BM, bool RAW_DEBUG>
+__global__ __launch_bounds__(256) void k_lstm_rec_fwd(
+    const __hip_bfloat16* __restrict__ h_prev, long hp_stride,
+    const __hip_bfloat16* __restrict__ w_t,     // (4H, H) row-major = W_h^T
+    const __hip_bfloat16* __restrict__ pre, long pre_stride,   // (B, 4H)
+    const float* __restrict__ c_prev, long cp_stride,
+    __hip_bfloat16* __restrict__ h_out, long ho_stride,
+    float* __restrict__ c_out, long co_stride,
+    __hip_bfloat16* __restrict__ acts, long act_stride,        // (B, 4H)
+    int B, int H)
+{
+    constexpr int MF = BM / 16;
+    const int tid  = threadIdx.x;
+    const int wave = tid >> 6;
+    const int lane = tid & 63;
+    const int lr   = lane & 15;   // fragment row (A) / col (B, C/D)
+    const int lk   = lane >> 4;   // k-group 0..3
+    const int m_base = blockIdx.x * BM;
+    const int n_base = blockIdx.y * 64 + wave * 16;  // h-column base
+
+    f32x4 acc[4][MF];
+#pragma unroll
+    for (int g = 0; g < 4; ++g)
+#pragma unroll
+        for (int mf = 0; mf < MF; ++mf) acc[g][mf] = f32x4{0, 0, 0, 0};
+
+    const long koff0 = (long)lk * 8;
+    for (int k0 = 0; k0 < H; k0 += 32) {
+        bf16x8 a[MF];
+#pragma unroll
+        for (int mf = 0; mf < MF; ++mf) {
+            const int row = m_base + mf * 16 + lr;
+            a[mf] = (row < B)
+                ? *(const bf16x8*)(h_prev + (long)row * hp_stride + k0 + koff0)
+                : bf16x8{};
+        }
+        bf16x8 bfr[4];
+#pragma unroll
+        for (int g = 0; g < 4; ++g) {
+            const long n = (long)g * H + n_base + lr;
+            bfr[g] = *(const bf16x8*)(w_t + n * H + k0 + koff0);
+        }
+#pragma unroll
+        for (int g = 0; g < 4; ++g)
+#pragma unroll
+            for (int mf = 0; mf < MF; ++mf)
+                acc[g][mf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    a[mf], bfr[g], acc[g][mf], 0, 0, 0);
+    }
+
+    // Epilogue: fused bias/pre add + activations + cell update.
+    const int hcol = n_base + lr;
+#pragma unroll
+    for (int mf = 0; mf < MF; ++mf) {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+            const int row = m_base + mf * 16 + lk * 4 + r;
+            if (row >= B) continue;
+            float raw[4];
+#pragma unroll
+            for (int g = 0; g < 4; ++g)
+                raw[g] = acc[g][mf][r] +
+                         __bfloat162float(pre[(long)row * pre_stride + (long)g * H + hcol]);
+            if (RAW_DEBUG) {
+                // debug/numerics mode: write raw gate pre-activations only
+#pragma unroll
+                for (int g = 0; g < 4; ++g)
+                    acts[(long)row * act_stride + (long)g * H + hcol] =
+                        __float2bfloat16(raw[g]);
+                continue;
+            }
+            const float gi = sigmoidf_(raw[0]);
+            const float gf = sigmoidf_(raw[1]);
+            const float gg = tanhf_(raw[2]);
+            const float go = sigmoidf_(raw[3]);
+            const float cp = c_prev[(long)row * cp_stride + hcol];
+            const float c  = gf * cp + gi * gg;
+            const float h  = go * tanhf_(c);
+            h_out[(long)row * ho_stride + hcol] = __float2bfloat16(h);
+            c_out[(long)row * co_stride + hcol] = c;
+            __hip_bfloat16* ap = acts + (long)row * act_stride + hcol;
+            ap[0]           = __float2bfloat16(gi);
+            ap[(long)H]     = __float2bfloat16(gf);
+            ap[(long)2 * H] = __float2bfloat16(gg);
+            ap[(long)3 * H] = __float2bfloat16(go);
+        }
+    }
+}
+
+// ---------------------------------------------------------------------------
+// K4: fused pointwise backward for one timestep (elementwise over B x H).
+// Produces raw-gate grads dgates_t (bf16) and the fp32 cell-grad carry
+// IN PLACE (dc reads+writes the same (row,col)).
+// dh_total = dh_up[t] + dh_carry (either may be null = zero).
+// Vectorized x4 per lane (G13); H % 4 == 0 required.
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(256) void k_lstm_pw_bwd(
+    const __hip_bfloat16* __restrict__ dh_up, long up_stride,   // nullable
+    const __hip_bfloat16* __restrict__ dh_carry, long hc_stride, // nullable
+    const __hip_bfloat16* __restrict__ acts, long act_stride,   // (B,4H)
+    const float* __restrict__ c_cur, long cc_stride,
+    const float* __restrict__ c_prev, long cp_stride,
+    __hip_bfloat16* __restrict__ dgates, long dg_stride,        // (B,4H)
+    float* __restrict__ dc, long dc_stride,                     // in-place carry
+    int B, int H, int first_step /* dc uninitialised: treat as dcT or 0 */,
+    const float* __restrict__ dcT, long dcT_stride)             // nullable
+{
+    const int row = blockIdx.y;
+    if (row >= B) return;
+    const int col0 = (blockIdx.x * blockDim.x + threadIdx.x) * 4;
+    if (col0 >= H) return;
+
+    float dh[4] = {0, 0, 0, 0};
+    if (dh_up) {
+        bf16x4 v = *(const bf16x4*)(dh_up + (long)row * up_stride + col0);
+#pragma unroll
+        for (int j = 0; j < 4; ++j) dh[j] += (float)v[j];
+    }
+    if (dh_carry) {
+        bf16x4 v = *(const bf16x4*)(dh_carry + (long)row * hc_stride + col0);
+#pragma unroll
+        for (int j = 0; j < 4; ++j) dh[j] += (float)v[j];
+    }
+    f32x4 dcv;
+    if (first_step) {
+        if (dcT) dcv = *(const f32x4*)(dcT + (long)row * dcT_stride + col0);
+        else dcv = f32x4{0, 0, 0, 0};
+    } else {
+        dcv = *(const f32x4*)(dc + (long)row * dc_stride + col0);
+    }
+    const f32x4 cc = *(const f32x4*)(c_cur + (long)row * cc_stride + col0);
+    const f32x4 cp = *(const f32x4*)(c_prev + (long)row * cp_stride + col0);
+    const __hip_bfloat16* ap = acts + (long)row * act_stride + col0;
+    bf16x4 vi = *(const bf16x4*)(ap);
+    bf16x4 vf = *(const bf16x4*)(ap + (long)H);
+    bf16x4 vg = *(const bf16x4*)(ap + (long)2 * H);
+    bf16x4 vo = *(const bf16x4*)(ap + (long)3 * H);
+
+    bf16x4 out_i, out_f, out_g, out_o;
+    f32x4 dc_out;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+        const float i = (float)vi[j], f = (float)vf[j];
+        const float g = (float)vg[j], o = (float)vo[j];
+        const float tc = tanhf_(cc[j]);
+        const float d_o = dh[j] * tc;
+        float d_c = dcv[j] + dh[j] * o * (1.0f - tc * tc);
+        const float d_i = d_c * g;
+        const float d_g = d_c * i;
+        const float d_f = d_c * cp[j];
+        dc_out[j] = d_c * f;
+        out_i[j] = (__bf16)(d_i * i * (1.0f - i));
+        out_f[j] = (__bf16)(d_f * f * (1.0f - f));
+        out_g[j] = (__bf16)(d_g * (1.0f - g * g));
+        out_o[j] = (__bf16)(d_o * o * (1.0f - o));
+    }
+    __hip_bfloat16* dgp = dgates + (long)row * dg_stride + col0;
+    *(bf16x4*)(dgp)               = out_i;
+    *(bf16x4*)(dgp + (long)H)     = out_f;
+    *(bf16x4*)(dgp + (long)2 * H) = out_g;
+    *(bf16x4*)(dgp + (long)3 * H) = out_o;
+    *(f32x4*)(dc + (long)row * dc_stride + col0) = dc_out;
+}
+
+// ---------------------------------------------------------------------------
+// K6: global-norm clip.  Per-tensor partial sum-of-squares -> atomicAdd into
+// a single fp32 accumulator; then scale-if-needed reading the final norm.
+// ---------------------------------------------------------------------------
+template <typename T>
+__global__ __launch_bounds__(256) void k_l2norm_partial(
+    const T* __restrict__ x, long n, float* __restrict__ out_sumsq)
+{
+    __shared__ float red[4];  // one partial per wave
+    float s = 0.f;
+    for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+         i += (long)gridDim.x * blockDim.x) {
+        const float v = (float)x[i];
+        s += v * v;
+    }
+#pragma unroll
+    for (int off = 32; off > 0; off >>= 1) s += __shfl_down(s, off, 64);
+    const int lane = threadIdx.x & 63, wave = threadIdx.x >> 6;
+    if (lane == 0) red[wave] = s;
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        float t = red[0] + red[1] + red[2] + red[3];
+        atomicAdd(out_sumsq, t);
+    }
+}
+
+template <typename T>
+__global__ __launch_bounds__(256) void k_clip_scale(
+    T* __restrict__ x, long n, const float* __restrict__ norm, float max_norm)
+{
+    const float nv = *norm;
+    const float coef = max_norm / (nv + 1e-6f);
+    if (coef >= 1.0f) return;
+    for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+         i += (long)gridDim.x * blockDim.x)
+        x[i] = (T)((float)x[i] * coef);
+}
+
+// ---------------------------------------------------------------------------
+// K7: fused optimizer apply (per tensor; fp32 master weights).
+// Math matches ops/optim.py exactly (the CPU oracle).
+// ---------------------------------------------------------------------------
+template <typename PT>
+__global__ __launch_bounds__(256) void k_sgd(
+    PT* __restrict__ p, float* __restrict__ m, const PT* __restrict__ g,
+    float* __restrict__ buf /* null if momentum==0 */, long n,
+    float lr, float mom, float wd)
+{
+    for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+         i += (long)gridDim.x * blockDim.x) {
+        float gv = (float)g[i];
+        float mv = m[i];
+        if (wd != 0.f) gv += wd * mv;
+        if (buf) {
+            const float b = buf[i] * mom + gv;
+            buf[i] = b;
+            gv = b;
+        }
+        mv -= lr * gv;
+        m[i] = mv;
+        p[i] = (PT)mv;
+    }
+}
+
+template <typename PT>
+__global__ __launch_bounds__(256) void k_adam(
+    PT* __restrict__ p, float* __restrict__ m, const PT* __restrict__ g,
+    float* __restrict__ ea, float* __restrict__ eas, long n,
+    float lr, float b1, float b2, float eps, float wd, float bc1, float bc2)
+{
+    for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+         i += (long)gridDim.x * blockDim.x) {
+        float gv = (float)g[i];
+        float mv = m[i];
+        if (wd != 0.f) gv += wd * mv;
+        const float a1 = ea[i] * b1 + (1.f - b1) * gv;
+        const float a2 = eas[i] * b2 + (1.f - b2) * gv * gv;
+        ea[i] = a1;
+        eas[i] = a2;
+        const float denom = sqrtf(a2 / bc2) + eps;
+        mv -= lr * (a1 / bc1) / denom;
+        m[i] = mv;
+        p[i] = (PT)mv;
+    }
+}
+
+// ---------------------------------------------------------------------------
+// C-linkage launchers (called from bindings.cpp)
+// ---------------------------------------------------------------------------
+extern "C" {
+
+void launch_lstm_rec_fwd(
+    const void* h_prev, long hp_stride, const void* w_t,
+    const void* pre, long pre_stride, const float* c_prev, long cp_stride,
+    void* h_out, long ho_stride, float* c_out, long co_stride,
+    void* acts, long act_stride, int B, int H, int raw_debug,
+    hipStream_t stream)
+{
+    constexpr int BM = 64;
+    dim3 grid((B + BM - 1) / BM, H / 64);
+    dim3 block(256);
+    if (raw_debug)
+        hipLaunchKernelGGL((k_lstm_rec_fwd<BM, true>), grid, block, 0, stream,
+            (const __hip_bfloat16*)h_prev, hp_stride, (const __hip_bfloat16*)w_t,
+            (const __hip_bfloat16*)pre, pre_stride, c_prev, cp_stride,
+            (__hip_bfloat16*)h_out, ho_stride, c_out, co_stride,
+            (__hip_bfloat16*)acts, act_stride, B, H);
+    else
+        hipLaunchKernelGGL((k_lstm_rec_fwd<BM, false>), grid, block, 0, stream,
+            (const __hip_bfloat16*)h_prev, hp_stride, (const __hip_bfloat16*)w_t,
+            (const __hip_bfloat16*)pre, pre_stride, c_prev, cp_stride,
+            (__hip_bfloat16*)h_out, ho_stride, c_out, co_stride,
+            (__hip_bfloat16*)acts, act_stride, B, H);
+}
+
+void launch_lstm_pw_bwd(
+    const void* dh_up, long up_stride, const void* dh_carry, long hc_stride,
+    const void* acts, long act_stride, const float* c_cur, long cc_stride,
+    const float* c_prev, long cp_stride, void* dgates, long dg_stride,
+    float* dc, long dc_stride, int B, int H, int first_step,
+    const float* dcT, long dcT_stride, hipStream_t stream)
+{
+    dim3 block(256);
+    dim3 grid((H / 4 + 255) / 256, B);
+    hipLaunchKernelGGL(k_lstm_pw_bwd, grid, block, 0, stream,
+        (const __hip_bfloat16*)dh_up, up_stride,
+        (const __hip_bfloat16*)dh_carry, hc_stride,
+        (const __hip_bfloat16*)acts, act_stride, c_cur, cc_stride,
+        c_prev, cp_stride, (__hip_bfloat16*)dgates, dg_stride,
+        dc, dc_stride, B, H, first_step, dcT, dcT_stride);
+}
+
+static inline int norm_grid(long n) {
+    long g = (n + 255) / 256;
+    return (int)(g > 2048 ? 2048 : g);
+}
+
+void launch_l2norm(const void* x, long n, float* out, int is_bf16,
+                   hipStream_t stream)
+{
+    dim3 grid(norm_grid(n)), block(256);
+    if (is_bf16)
+        hipLaunchKernelGGL(k_l2norm_partial<__hip_bfloat16>, grid, block, 0,
+                           stream, (const __hip_bfloat16*)x, n, out);
+    else
+        hipLaunchKernelGGL(k_l2norm_partial<float>, grid, block, 0, stream,
+                           (const float*)x, n, out);
+}
+
+void launch_clip_scale(void* x, long n, const float* norm, float max_norm,
+                       int is_bf16, hipStream_t stream)
+{
+    dim3 grid(norm_grid(n)), block(256);
+    if (is_bf16)
+        hipLaunchKernelGGL(k_clip_scale<__hip_bfloat16>, grid, block, 0,
+                           stream, (__hip_bfloat16*)x, n, norm, max_norm);
+    else
+        hipLaunchKernelGGL(k_clip_scale<float>, grid, block, 0, stream,
+                           (float*)x, n, norm, max_norm);
+}
+
+void launch_sgd(void* p, float* m, const void* g, float* buf, long n,
+                float lr, float mom, float wd, int is_bf16, hipStream_t stream)
+{
+    dim3 grid(norm_grid(n)), block(256);
+    if (is_bf16)
+        hipLaunchKernelGGL(k_sgd<__hip_bfloat16>, grid, block, 0, stream,
+            (__hip_bfloat16*)p, m, (const __hip_bfloat16*)g, buf, n, lr, mom, wd);
+    else
+        hipLaunchKernelGGL(k_sgd<float>, grid, block, 0, stream,
+            (float*)p, m, (const float*)g, buf, n, lr, mom, wd);
+}
+
+void launch_adam(void* p, float* m, const void* g, float* ea, float* eas,
+                 long n, float lr, float b1, float b2, float eps, float wd,
+                 float bc1, float bc2, int is_bf16, hipStream_t stream)
+{
+    dim3 grid(norm_grid(n)), block(256);
+    if (is_bf16)
+        hipLaunchKernelGGL(k_adam<__hip_bfloat16>, grid, block, 0, stream,
+            (__hip_bfloat16*)p, m, (const __hip_bfloat16*)g, ea, eas, n,
+            lr, b1, b2, eps, wd, bc1, bc2);
+    else
+        hipLaunchKernelGGL(k_adam<float>, grid, block, 0, stream,
+            (float*)p, m, (const float*)g, ea, eas, n,
+            lr, b1, b2, eps, wd, bc1, bc2);
+}
+
+}  // extern "C"

@@ -56,12 +56,16 @@ def force_ref() -> bool:
     return os.environ.get("LSTM_TS_FORCE_REF", "0") == "1"
 
 
-def use_hip(device: torch.device | str) -> bool:
-    """Decide the backend for a given device: HIP kernels on GPU, ref on CPU."""
+def use_hip(device: torch.device | str,
+            dtype: "torch.dtype | None" = None) -> bool:
+    """Decide the backend: HIP kernels on GPU for bf16 (the flagship compute
+    dtype), reference ops on CPU or for fp32-on-GPU (numerics/debug runs)."""
     dev = torch.device(device)
     if dev.type != "cuda":
         return False
     if force_ref():
+        return False
+    if dtype is not None and dtype != torch.bfloat16:
         return False
     # GPU path: the extension is mandatory (fail loudly if missing).
     get_extension()

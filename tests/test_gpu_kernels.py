@@ -1,0 +1,243 @@
+"""GPU numerics: every HIP kernel vs the plain-PyTorch fp32 reference
+(functional.py / optim.py CPU oracle) on the same device.  All tests here
+run in bf16 (the flagship compute dtype) against fp32 references built
+from the SAME bf16-quantized inputs, so tolerances cover accumulation-
+order differences only.
+"""
+import os
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+requires_gpu = pytest.mark.skipif(not torch.cuda.is_available(), reason="no GPU")
+
+
+@pytest.fixture(scope="module")
+def ext():
+    from lstm_tensorspark_amd.ops import extension
+
+    return extension.get_extension()
+
+
+def make_step_inputs(B, H, seed=0, dev="cuda"):
+    g = torch.Generator(device="cpu").manual_seed(seed)
+
+    def r(*s, scale=1.0):
+        return (torch.randn(*s, generator=g) * scale).to(dev)
+
+    h_prev = r(B, H, scale=0.5).bfloat16()
+    w_h = r(H, 4 * H, scale=1.0 / H ** 0.5).bfloat16()
+    pre = r(B, 4 * H, scale=1.0).bfloat16()
+    c_prev = r(B, H, scale=0.5)  # fp32
+    return h_prev, w_h, pre, c_prev
+
+
+# ---------------------------------------------------------------- K1b GEMM
+@pytest.mark.parametrize("B,H", [(64, 64), (128, 128), (100, 256), (256, 1024)])
+def test_rec_gemm_mapping_raw(ext, B, H):
+    """MFMA fragment mapping: raw gates == pre + h @ W_h.  Asymmetric random
+    operands (transpose-detecting, guide §5.4 rule 16)."""
+    h_prev, w_h, pre, c_prev = make_step_inputs(B, H)
+    h_out, c_out, acts = ext.lstm_rec_fwd_step(h_prev, w_h, pre, c_prev, True)
+    ref = pre.float() + h_prev.float() @ w_h.float()
+    err = (acts.float() - ref).abs().max().item()
+    scale = ref.abs().max().item()
+    assert err < 0.02 * max(scale, 1.0), f"raw-gate mismatch: {err} (scale {scale})"
+
+
+@pytest.mark.parametrize("B,H", [(64, 64), (96, 128), (256, 1024)])
+def test_rec_fwd_step_full(ext, B, H):
+    """Fused step: activations + cell update vs fp32 reference."""
+    h_prev, w_h, pre, c_prev = make_step_inputs(B, H, seed=1)
+    h_out, c_out, acts = ext.lstm_rec_fwd_step(h_prev, w_h, pre, c_prev, False)
+    raw = pre.float() + h_prev.float() @ w_h.float()
+    i, f, g, o = raw.chunk(4, dim=1)
+    i, f, g, o = i.sigmoid(), f.sigmoid(), g.tanh(), o.sigmoid()
+    c_ref = f * c_prev + i * g
+    h_ref = o * torch.tanh(c_ref)
+    assert (c_out - c_ref).abs().max().item() < 0.03
+    assert (h_out.float() - h_ref).abs().max().item() < 0.02
+    acts_ref = torch.cat([i, f, g, o], dim=1)
+    assert (acts.float() - acts_ref).abs().max().item() < 0.02
+
+
+# ---------------------------------------------------------------- sequence
+def seq_inputs(B, T, H, seed=0, dev="cuda"):
+    g = torch.Generator(device="cpu").manual_seed(seed)
+
+    def r(*s, scale=1.0):
+        return (torch.randn(*s, generator=g) * scale).to(dev)
+
+    pre = r(B, T, 4 * H).bfloat16()
+    w_h = r(H, 4 * H, scale=1.0 / H ** 0.5).bfloat16()
+    h0 = r(B, H, scale=0.5).bfloat16()
+    c0 = r(B, H, scale=0.5)
+    return pre, w_h, h0, c0
+
+
+@pytest.mark.parametrize("reverse", [False, True])
+@pytest.mark.parametrize("B,T,H", [(64, 5, 64), (256, 16, 128)])
+def test_seq_fwd_vs_ref(ext, reverse, B, T, H):
+    from lstm_tensorspark_amd.ops import functional
+
+    pre, w_h, h0, c0 = seq_inputs(B, T, H)
+    h_seq, c_seq, acts, hT, cT = ext.lstm_seq_fwd(pre, w_h, h0, c0, reverse)
+    h_r, c_r, a_r, hT_r, cT_r = functional.ref_seq_fwd(pre, w_h, h0, c0, reverse)
+    # recurrence compounds bf16 rounding; tolerance grows mildly with T
+    tol = 0.02 + 0.005 * T
+    assert (h_seq.float() - h_r.float()).abs().max().item() < tol
+    assert (c_seq - c_r).abs().max().item() < 2 * tol
+    assert (hT.float() - hT_r.float()).abs().max().item() < tol
+    assert (cT - cT_r).abs().max().item() < 2 * tol
+
+
+@pytest.mark.parametrize("reverse", [False, True])
+def test_seq_bwd_vs_ref(ext, reverse):
+    """Backward kernels on IDENTICAL saved tensors as the reference."""
+    from lstm_tensorspark_amd.ops import functional
+
+    B, T, H = 64, 8, 128
+    pre, w_h, h0, c0 = seq_inputs(B, T, H, seed=2)
+    # use the reference forward's saved tensors for BOTH backward paths
+    h_r, c_r, a_r, hT_r, cT_r = functional.ref_seq_fwd(pre, w_h, h0, c0, reverse)
+    g = torch.Generator(device="cpu").manual_seed(3)
+    d_hseq = (torch.randn(B, T, H, generator=g) * 0.1).to("cuda").bfloat16()
+    d_hT = (torch.randn(B, H, generator=g) * 0.1).to("cuda").bfloat16()
+    d_cT = (torch.randn(B, H, generator=g) * 0.1).to("cuda")
+
+    dg, dh0, dc0 = ext.lstm_seq_bwd(d_hseq, d_hT, d_cT, w_h, a_r, c_r, c0, reverse)
+    dg_r, dh0_r, dc0_r = functional.ref_seq_bwd(d_hseq, d_hT, d_cT, w_h, a_r,
+                                                c_r, c0, reverse)
+    assert (dg.float() - dg_r.float()).abs().max().item() < 0.05
+    assert (dh0.float() - dh0_r.float()).abs().max().item() < 0.05
+    assert (dc0 - dc0_r).abs().max().item() < 0.05
+
+
+def test_seq_bwd_none_grads(ext):
+    from lstm_tensorspark_amd.ops import functional
+
+    B, T, H = 32, 4, 64
+    pre, w_h, h0, c0 = seq_inputs(B, T, H, seed=4)
+    h_r, c_r, a_r, _, _ = functional.ref_seq_fwd(pre, w_h, h0, c0, False)
+    d_hT = (torch.randn(B, H) * 0.1).to("cuda").bfloat16()
+    dg, dh0, dc0 = ext.lstm_seq_bwd(None, d_hT, None, w_h, a_r, c_r, c0, False)
+    dg_r, dh0_r, dc0_r = functional.ref_seq_bwd(None, d_hT, None, w_h, a_r,
+                                                c_r, c0, False)
+    assert (dg.float() - dg_r.float()).abs().max().item() < 0.05
+    assert (dc0 - dc0_r).abs().max().item() < 0.05
+
+
+# ------------------------------------------------------------- full layer
+def test_layer_autograd_gpu_vs_cpu_fp32():
+    """End-to-end layer fwd+bwd: GPU bf16 fused path vs CPU fp32 reference."""
+    from lstm_tensorspark_amd.ops.lstm_layer import lstm_layer
+
+    B, T, X, H = 32, 8, 128, 128
+    g = torch.Generator(device="cpu").manual_seed(5)
+    x32 = torch.randn(B, T, X, generator=g) * 0.5
+    h032 = torch.randn(B, H, generator=g) * 0.5
+    c032 = torch.randn(B, H, generator=g) * 0.5
+    wx32 = torch.randn(X, 4 * H, generator=g) / X ** 0.5
+    wh32 = torch.randn(H, 4 * H, generator=g) / H ** 0.5
+    b32 = torch.randn(4 * H, generator=g) * 0.1
+    gout = torch.randn(B, T, H, generator=g) * 0.1
+
+    # CPU fp32 reference
+    cpu = [t.clone().requires_grad_(True) for t in (x32, h032, c032, wx32, wh32, b32)]
+    h_seq_c, _, _ = lstm_layer(*cpu)
+    h_seq_c.backward(gout)
+
+    # GPU bf16 fused
+    gpu = [t.to("cuda").bfloat16().requires_grad_(True) for t in (x32, h032, wx32, wh32, b32)]
+    c0g = c032.to("cuda").requires_grad_(True)
+    xg, h0g, wxg, whg, bg = gpu
+    h_seq_g, _, _ = lstm_layer(xg, h0g, c0g, wxg, whg, bg)
+    h_seq_g.backward(gout.to("cuda").bfloat16())
+
+    assert (h_seq_g.float().cpu() - h_seq_c.detach()).abs().max().item() < 0.1
+    pairs = [(xg, cpu[0]), (h0g, cpu[1]), (c0g, cpu[2]), (wxg, cpu[3]),
+             (whg, cpu[4]), (bg, cpu[5])]
+    for gt, ct in pairs:
+        gg, cg = gt.grad.float().cpu(), ct.grad
+        denom = cg.abs().max().item() + 1e-3
+        rel = (gg - cg).abs().max().item() / denom
+        assert rel < 0.12, f"grad rel err {rel}"
+
+
+# ---------------------------------------------------------------- K6 / K7
+def test_l2norm_and_clip(ext):
+    gs = [torch.randn(1000, device="cuda").bfloat16(),
+          torch.randn(64, 33, device="cuda"),
+          torch.randn(7, device="cuda").bfloat16()]
+    norm = ext.multi_tensor_l2norm(list(gs))
+    want = torch.sqrt(sum(g.float().pow(2).sum() for g in gs))
+    assert abs(norm.item() - want.item()) / want.item() < 1e-2
+    ext.multi_tensor_clip_scale(list(gs), norm, 0.5)
+    after = torch.sqrt(sum(g.float().pow(2).sum() for g in gs))
+    assert abs(after.item() - 0.5) < 0.02
+
+
+def test_fused_sgd_vs_oracle(ext):
+    torch.manual_seed(0)
+    shapes = [(128, 64), (4096,), (33,)]
+    ps = [torch.randn(s, device="cuda").bfloat16() for s in shapes]
+    ms = [p.float() for p in ps]
+    bufs = [torch.zeros_like(m) for m in ms]
+    ps_c = [p.clone() for p in ps]
+    ms_c = [m.clone() for m in ms]
+    bufs_c = [b.clone() for b in bufs]
+    for step in range(3):
+        gs = [torch.randn(s, device="cuda").bfloat16() * 0.1 for s in shapes]
+        ext.fused_sgd(list(ps), list(ms), list(gs), list(bufs), 0.1, 0.9, 0.01)
+        # oracle (fp32 math, same formulas as ops/optim.py CPU path)
+        for p, m, b, g in zip(ps_c, ms_c, bufs_c, gs):
+            gv = g.float() + 0.01 * m
+            b.mul_(0.9).add_(gv)
+            m.add_(b, alpha=-0.1)
+            p.copy_(m.bfloat16())
+    for a, b in zip(ms, ms_c):
+        assert (a - b).abs().max().item() < 1e-5
+    for a, b in zip(ps, ps_c):
+        assert torch.equal(a, b)
+
+
+def test_fused_adam_vs_oracle(ext):
+    torch.manual_seed(1)
+    shapes = [(256, 32), (999,)]
+    ps = [torch.randn(s, device="cuda").bfloat16() for s in shapes]
+    ms = [p.float() for p in ps]
+    ea = [torch.zeros_like(m) for m in ms]
+    eas = [torch.zeros_like(m) for m in ms]
+    ps_c = [p.clone() for p in ps]
+    ms_c = [m.clone() for m in ms]
+    ea_c = [e.clone() for e in ea]
+    eas_c = [e.clone() for e in eas]
+    b1, b2, eps, lr = 0.9, 0.999, 1e-8, 1e-3
+    for step in range(1, 4):
+        bc1, bc2 = 1 - b1 ** step, 1 - b2 ** step
+        gs = [torch.randn(s, device="cuda").bfloat16() * 0.1 for s in shapes]
+        ext.fused_adam(list(ps), list(ms), list(gs), list(ea), list(eas),
+                       lr, b1, b2, eps, 0.0, bc1, bc2)
+        for p, m, e1, e2, g in zip(ps_c, ms_c, ea_c, eas_c, gs):
+            gv = g.float()
+            e1.mul_(b1).add_(gv, alpha=1 - b1)
+            e2.mul_(b2).addcmul_(gv, gv, value=1 - b2)
+            denom = (e2 / bc2).sqrt().add_(eps)
+            m.addcdiv_(e1 / bc1, denom, value=-lr)
+            p.copy_(m.bfloat16())
+    for a, b in zip(ms, ms_c):
+        assert (a - b).abs().max().item() < 1e-5
+
+
+# -------------------------------------------------- native-code-is-loaded
+def test_hip_path_actually_dispatches():
+    """The flagship path must run the HIP extension, not eager fallback."""
+    from lstm_tensorspark_amd.ops import extension
+    from lstm_tensorspark_amd.ops.lstm_layer import lstm_layer
+
+    assert extension.extension_available(), "extension must be importable"
+    assert extension.use_hip(torch.device("cuda"), torch.bfloat16)
+    so = extension.get_extension().__file__
+    assert "lstm_tensorspark_amd" in so  # in-tree .so, not site-packages

@@ -1,0 +1,77 @@
+"""GPU integration: single-GPU bf16 training through the fused kernels
+(SURVEY.md §4.2 "Integration: single-GPU train" tier)."""
+import os
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+from lstm_tensorspark_amd.config import Config, ModelConfig, TrainConfig, DataConfig
+from lstm_tensorspark_amd.trainer import Trainer
+
+
+def copy_cfg(**train_kw):
+    kw = dict(batch_size=64, seq_len=32, lr=0.5, momentum=0.9, steps=40,
+              log_interval=0, seed=7, dtype="bf16")
+    kw.update(train_kw)
+    return Config(
+        model=ModelConfig(input_size=64, hidden_size=128, num_layers=1,
+                          num_classes=8),
+        train=TrainConfig(**kw),
+        data=DataConfig(task="copy", copy_delay=1),
+    )
+
+
+def test_gpu_loss_decreases_on_copy_task():
+    t = Trainer(copy_cfg(), device="cuda:0")
+    first = t.train_step(t.data.batch(0))["loss"]
+    t.fit(40)
+    last = t.train_step(t.data.batch(41))["loss"]
+    assert last < first * 0.6, (first, last)
+
+
+def test_gpu_bidirectional_multilayer_step():
+    cfg = copy_cfg(steps=2)
+    cfg.model.bidirectional = True
+    cfg.model.num_layers = 2
+    cfg.model.dropout = 0.1
+    t = Trainer(cfg, device="cuda:0")
+    info = t.train_step(t.data.batch(0))
+    assert info["loss"] == info["loss"]
+
+
+def test_gpu_lm_head_adam_clip_bptt():
+    cfg = Config(
+        model=ModelConfig(input_size=128, hidden_size=128, num_layers=2,
+                          head="lm", vocab_size=512),
+        train=TrainConfig(batch_size=16, seq_len=64, bptt=32, lr=1e-3,
+                          optimizer="adam", grad_clip=1.0, steps=2,
+                          log_interval=0, dtype="bf16"),
+        data=DataConfig(task="lm"),
+    )
+    t = Trainer(cfg, device="cuda:0")
+    t.fit(2)
+    assert t.step == 2
+
+
+def test_gpu_checkpoint_resume_bitwise(tmp_path):
+    cfg = copy_cfg(steps=3)
+    cfg.train.ckpt_dir = str(tmp_path)
+    tA = Trainer(cfg, device="cuda:0")
+    tA.fit(2)
+    mid = os.path.join(str(tmp_path), "mid.pt")
+    tA.save(mid)
+    tA.fit(2)
+    tB = Trainer(copy_cfg(steps=3), device="cuda:0")
+    tB.load(mid)
+    tB.fit(2)
+    for k, vA in tA.model.weight_dict().items():
+        vB = tB.model.weight_dict()[k]
+        assert torch.equal(vA, vB), f"resume not bitwise for {k}"
+
+
+def test_graft_smoke_entry():
+    import __graft_entry__
+
+    __graft_entry__.smoke()
