@@ -1,0 +1,128 @@
+#!/usr/bin/env python3
+"""Microbenchmark of the hot kernels in isolation (GPU box).
+
+Prints one JSON line per timing so the results can be committed under
+profiles/ and compared across kernel revisions.
+"""
+import argparse
+import json
+import os
+import sys
+
+import torch
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+
+def time_op(fn, iters=50, warmup=10):
+    for _ in range(warmup):
+        fn()
+    torch.cuda.synchronize()
+    start = torch.cuda.Event(enable_timing=True)
+    end = torch.cuda.Event(enable_timing=True)
+    start.record()
+    for _ in range(iters):
+        fn()
+    end.record()
+    torch.cuda.synchronize()
+    return start.elapsed_time(end) / iters  # ms
+
+
+def report(name, ms, flops=None, bytes_=None, **extra):
+    rec = {"name": name, "ms": round(ms, 4)}
+    if flops:
+        rec["tflops"] = round(flops / ms / 1e9, 1)
+    if bytes_:
+        rec["gbps"] = round(bytes_ / ms / 1e6, 1)
+    rec.update(extra)
+    print(json.dumps(rec), flush=True)
+    return rec
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--B", type=int, default=1024)
+    ap.add_argument("--H", type=int, default=1024)
+    ap.add_argument("--T", type=int, default=128)
+    ap.add_argument("--iters", type=int, default=50)
+    args = ap.parse_args()
+    B, H, T = args.B, args.H, args.T
+
+    from lstm_tensorspark_amd.ops import extension
+    ext = extension.get_extension()
+    dev = "cuda"
+    g = torch.Generator(device=dev).manual_seed(0)
+
+    h_prev = torch.randn(B, H, generator=g, device=dev).bfloat16()
+    w_h = (torch.randn(H, 4 * H, generator=g, device=dev) / H ** 0.5).bfloat16()
+    w_t = w_h.t().contiguous()
+    pre1 = torch.randn(B, 4 * H, generator=g, device=dev).bfloat16()
+    c_prev = torch.randn(B, H, generator=g, device=dev)
+
+    rec_flops = 2 * B * 4 * H * H
+    # single fused recurrent step (includes w_h.t().contiguous() overhead
+    # unless we prepass -- rec_fwd_step transposes internally; measure both)
+    report("rec_fwd_step(incl. transpose)", time_op(
+        lambda: ext.lstm_rec_fwd_step(h_prev, w_h, pre1, c_prev, False),
+        args.iters), rec_flops, B=B, H=H)
+
+    # sequence forward (transpose amortized over T steps)
+    pre = torch.randn(B, T, 4 * H, generator=g, device=dev).bfloat16()
+    h0 = torch.randn(B, H, generator=g, device=dev).bfloat16()
+    c0 = torch.randn(B, H, generator=g, device=dev)
+    ms = time_op(lambda: ext.lstm_seq_fwd(pre, w_h, h0, c0, False),
+                 max(args.iters // 10, 3))
+    report("seq_fwd", ms, rec_flops * T, B=B, H=H, T=T,
+           us_per_step=round(1e3 * ms / T, 2))
+
+    # sequence backward
+    h_seq, c_seq, acts, hT, cT = ext.lstm_seq_fwd(pre, w_h, h0, c0, False)
+    d_hseq = torch.randn(B, T, H, generator=g, device=dev).bfloat16()
+    ms = time_op(lambda: ext.lstm_seq_bwd(d_hseq, None, None, w_h, acts,
+                                          c_seq, c0, False),
+                 max(args.iters // 10, 3))
+    report("seq_bwd", ms, 2 * rec_flops * T, B=B, H=H, T=T,
+           us_per_step=round(1e3 * ms / T, 2))
+
+    # components: library GEMMs
+    dg1 = torch.randn(B, 4 * H, generator=g, device=dev).bfloat16()
+    report("mm dh=dgates@WhT (per step)", time_op(lambda: dg1 @ w_h.t(), args.iters),
+           rec_flops)
+    x_flat = torch.randn(B * T, H, generator=g, device=dev).bfloat16()
+    w_x = (torch.randn(H, 4 * H, generator=g, device=dev) / H ** 0.5).bfloat16()
+    report("pre-GEMM (BT,X)@(X,4H)", time_op(lambda: x_flat @ w_x, 10),
+           2 * B * T * H * 4 * H)
+    dg_flat = torch.randn(B * T, 4 * H, generator=g, device=dev).bfloat16()
+    report("dW GEMM (H,BT)@(BT,4H)", time_op(lambda: x_flat.t() @ dg_flat, 10),
+           2 * B * T * H * 4 * H)
+
+    # full layer autograd fwd+bwd
+    from lstm_tensorspark_amd.ops.lstm_layer import lstm_layer
+    x = torch.randn(B, T, H, generator=g, device=dev).bfloat16()
+    wx_p = w_x.clone().requires_grad_(True)
+    wh_p = w_h.clone().requires_grad_(True)
+    b_p = torch.zeros(4 * H, device=dev).bfloat16().requires_grad_(True)
+    h0z = torch.zeros(B, H, device=dev).bfloat16()
+    c0z = torch.zeros(B, H, device=dev)
+    gout = torch.randn(B, T, H, generator=g, device=dev).bfloat16()
+
+    def layer_fb():
+        hs, _, _ = lstm_layer(x, h0z, c0z, wx_p, wh_p, b_p)
+        hs.backward(gout)
+        wx_p.grad = wh_p.grad = b_p.grad = None
+
+    total_flops = (2 * B * T * 4 * H * H) * 2 + rec_flops * T * 3 + 2 * B * T * H * 4 * H * 2
+    report("layer fwd+bwd", time_op(layer_fb, 5, warmup=3), total_flops,
+           B=B, H=H, T=T)
+
+    # pointwise bwd kernel isolated
+    dc = torch.zeros(B, H, device=dev)
+    dg_out = torch.empty(B, 4 * H, device=dev).bfloat16()
+    report("pw_bwd isolated (via seq T=1)", time_op(
+        lambda: ext.lstm_seq_bwd(None, d_hseq[:, 0].contiguous(), None, w_h,
+                                 acts[:, :1].contiguous(), c_seq[:, :1].contiguous(),
+                                 c0, False), args.iters))
+
+
+if __name__ == "__main__":
+    main()
