@@ -27,6 +27,8 @@ void launch_sgd(void*, float*, const void*, float*, long, float, float, float,
                 int, hipStream_t);
 void launch_adam(void*, float*, const void*, float*, float*, long, float,
                  float, float, float, float, float, float, int, hipStream_t);
+void set_fwd_variant(int);
+int get_fwd_variant();
 }
 
 namespace {
@@ -271,6 +273,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           py::arg("c_seq"), py::arg("c0"), py::arg("reverse"));
     m.def("lstm_rec_fwd_step", &lstm_rec_fwd_step,
           "single fused recurrent step (raw_debug for GEMM-mapping tests)");
+    m.def("set_fwd_variant", &set_fwd_variant,
+          "select rec_fwd kernel variant (bit0 STAGE_A, bit1 XCD_SWZ, bit2 BM32)");
+    m.def("get_fwd_variant", &get_fwd_variant);
     m.def("multi_tensor_l2norm", &multi_tensor_l2norm);
     m.def("multi_tensor_clip_scale", &multi_tensor_clip_scale);
     m.def("fused_sgd", &fused_sgd);

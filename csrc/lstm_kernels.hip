@@ -44,8 +44,17 @@ DEVINL float tanhf_(float x) { return tanhf(x); }
 // [k=(l>>4)*8 ..+8]; B-fragment: w_t[n=l&15][same k] with w_t=(4H,H)=W_h^T;
 // C/D: col=l&15, row=(l>>4)*4+r.
 // TAIL HANDLING: rows (batch) bounds-checked; requires H % 64 == 0.
+//
+// Variants (within-probe A/B sweep, tools/kernel_bench.py):
+//   STAGE_A: stage the h tile in LDS once per BK=64 K-step (the 4 waves
+//            otherwise each load identical A fragments = 4x redundant L2
+//            traffic).  LDS rows padded +8 elems so the ds_read_b128
+//            fragment reads are bank-conflict-free (16 lanes' start banks
+//            4*(9r mod 16) all distinct).
+//   XCD_SWZ: bijective blockIdx remap so blocks sharing a W_h panel
+//            (same n-block) land on the same XCD's private L2 (T1).
 // ---------------------------------------------------------------------------
-template <int BM, bool RAW_DEBUG>
+template <int BM, bool RAW_DEBUG, bool STAGE_A, bool XCD_SWZ>
 __global__ __launch_bounds__(256) void k_lstm_rec_fwd(
     const __hip_bfloat16* __restrict__ h_prev, long hp_stride,
     const __hip_bfloat16* __restrict__ w_t,     // (4H, H) row-major = W_h^T
@@ -57,13 +66,27 @@ __global__ __launch_bounds__(256) void k_lstm_rec_fwd(
     int B, int H)
 {
     constexpr int MF = BM / 16;
+    constexpr int LDA = 72;  // 64 + 8 pad: conflict-free b128 column reads
+    __shared__ __hip_bfloat16 a_lds[STAGE_A ? BM : 1][STAGE_A ? LDA : 1];
+
     const int tid  = threadIdx.x;
     const int wave = tid >> 6;
     const int lane = tid & 63;
     const int lr   = lane & 15;   // fragment row (A) / col (B, C/D)
     const int lk   = lane >> 4;   // k-group 0..3
-    const int m_base = blockIdx.x * BM;
-    const int n_base = blockIdx.y * 64 + wave * 16;  // h-column base
+
+    int m_blk = blockIdx.x, n_blk = blockIdx.y;
+    if (XCD_SWZ) {
+        const int nm = gridDim.x, nwg = gridDim.x * gridDim.y;
+        const int bid = blockIdx.x + blockIdx.y * nm;
+        const int q = nwg >> 3, r = nwg & 7, xcd = bid & 7, loc = bid >> 3;
+        const int swz = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q)
+                        + loc;
+        m_blk = swz % nm;      // m fastest: one XCD covers few W panels
+        n_blk = swz / nm;
+    }
+    const int m_base = m_blk * BM;
+    const int n_base = n_blk * 64 + wave * 16;  // h-column base
 
     f32x4 acc[4][MF];
 #pragma unroll
@@ -72,27 +95,88 @@ __global__ __launch_bounds__(256) void k_lstm_rec_fwd(
         for (int mf = 0; mf < MF; ++mf) acc[g][mf] = f32x4{0, 0, 0, 0};
 
     const long koff0 = (long)lk * 8;
-    for (int k0 = 0; k0 < H; k0 += 32) {
-        bf16x8 a[MF];
-#pragma unroll
-        for (int mf = 0; mf < MF; ++mf) {
-            const int row = m_base + mf * 16 + lr;
-            a[mf] = (row < B)
-                ? *(const bf16x8*)(h_prev + (long)row * hp_stride + k0 + koff0)
-                : bf16x8{};
-        }
-        bf16x8 bfr[4];
-#pragma unroll
-        for (int g = 0; g < 4; ++g) {
-            const long n = (long)g * H + n_base + lr;
-            bfr[g] = *(const bf16x8*)(w_t + n * H + k0 + koff0);
-        }
+
+    if (STAGE_A) {
+        // BK = 64 K-steps; A staged in LDS, B register-prefetched one step
+        // ahead (fragment-shaped direct global loads, L2-resident W panel).
+        constexpr int BK = 64;
+        // staging map: thread -> (row = tid/8, col8 = tid%8); BM=64 needs
+        // 1 slice of 256 threads per 64x64 tile half... (BM*BK/8) / 256
+        constexpr int SLICES = (BM * BK / 8 + 255) / 256;
+        bf16x8 b_nxt[4][2];
 #pragma unroll
         for (int g = 0; g < 4; ++g)
 #pragma unroll
-            for (int mf = 0; mf < MF; ++mf)
-                acc[g][mf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                    a[mf], bfr[g], acc[g][mf], 0, 0, 0);
+            for (int s = 0; s < 2; ++s)
+                b_nxt[g][s] = *(const bf16x8*)(
+                    w_t + ((long)g * H + n_base + lr) * H + s * 32 + koff0);
+        for (int k0 = 0; k0 < H; k0 += BK) {
+            // stage A tile (BM x BK) -- one 16B load+store per thread slice
+            __syncthreads();
+#pragma unroll
+            for (int s = 0; s < SLICES; ++s) {
+                const int t = tid + s * 256;
+                const int ar = t >> 3, ac = (t & 7) * 8;
+                if (ar < BM) {
+                    const int row = m_base + ar;
+                    bf16x8 v = (row < B)
+                        ? *(const bf16x8*)(h_prev + (long)row * hp_stride + k0 + ac)
+                        : bf16x8{};
+                    *(bf16x8*)(&a_lds[ar][ac]) = v;
+                }
+            }
+            __syncthreads();
+            bf16x8 b_cur[4][2];
+#pragma unroll
+            for (int g = 0; g < 4; ++g)
+#pragma unroll
+                for (int s = 0; s < 2; ++s) b_cur[g][s] = b_nxt[g][s];
+            if (k0 + BK < H) {
+#pragma unroll
+                for (int g = 0; g < 4; ++g)
+#pragma unroll
+                    for (int s = 0; s < 2; ++s)
+                        b_nxt[g][s] = *(const bf16x8*)(
+                            w_t + ((long)g * H + n_base + lr) * H
+                            + k0 + BK + s * 32 + koff0);
+            }
+#pragma unroll
+            for (int s = 0; s < 2; ++s) {
+                bf16x8 a[MF];
+#pragma unroll
+                for (int mf = 0; mf < MF; ++mf)
+                    a[mf] = *(const bf16x8*)(&a_lds[mf * 16 + lr][s * 32 + lk * 8]);
+#pragma unroll
+                for (int g = 0; g < 4; ++g)
+#pragma unroll
+                    for (int mf = 0; mf < MF; ++mf)
+                        acc[g][mf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                            a[mf], b_cur[g][s], acc[g][mf], 0, 0, 0);
+            }
+        }
+    } else {
+        for (int k0 = 0; k0 < H; k0 += 32) {
+            bf16x8 a[MF];
+#pragma unroll
+            for (int mf = 0; mf < MF; ++mf) {
+                const int row = m_base + mf * 16 + lr;
+                a[mf] = (row < B)
+                    ? *(const bf16x8*)(h_prev + (long)row * hp_stride + k0 + koff0)
+                    : bf16x8{};
+            }
+            bf16x8 bfr[4];
+#pragma unroll
+            for (int g = 0; g < 4; ++g) {
+                const long n = (long)g * H + n_base + lr;
+                bfr[g] = *(const bf16x8*)(w_t + n * H + k0 + koff0);
+            }
+#pragma unroll
+            for (int g = 0; g < 4; ++g)
+#pragma unroll
+                for (int mf = 0; mf < MF; ++mf)
+                    acc[g][mf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        a[mf], bfr[g], acc[g][mf], 0, 0, 0);
+        }
     }
 
     // Epilogue: fused bias/pre add + activations + cell update.
@@ -300,6 +384,47 @@ __global__ __launch_bounds__(256) void k_adam(
 // ---------------------------------------------------------------------------
 extern "C" {
 
+// variant encoding: bit0 = STAGE_A, bit1 = XCD_SWZ, bit2 = BM32 (else BM64)
+void launch_lstm_rec_fwd_v(
+    const void* h_prev, long hp_stride, const void* w_t,
+    const void* pre, long pre_stride, const float* c_prev, long cp_stride,
+    void* h_out, long ho_stride, float* c_out, long co_stride,
+    void* acts, long act_stride, int B, int H, int raw_debug, int variant,
+    hipStream_t stream)
+{
+    const int BM = (variant & 4) ? 32 : 64;
+    dim3 grid((B + BM - 1) / BM, H / 64);
+    dim3 block(256);
+#define ARGS                                                                  \
+    (const __hip_bfloat16*)h_prev, hp_stride, (const __hip_bfloat16*)w_t,     \
+    (const __hip_bfloat16*)pre, pre_stride, c_prev, cp_stride,                \
+    (__hip_bfloat16*)h_out, ho_stride, c_out, co_stride,                      \
+    (__hip_bfloat16*)acts, act_stride, B, H
+#define LAUNCH(BM_, RAW_, ST_, SW_)                                           \
+    hipLaunchKernelGGL((k_lstm_rec_fwd<BM_, RAW_, ST_, SW_>), grid, block, 0, \
+                       stream, ARGS)
+#define DISPATCH(RAW_)                                                        \
+    switch (variant & 7) {                                                    \
+        case 0: LAUNCH(64, RAW_, false, false); break;                        \
+        case 1: LAUNCH(64, RAW_, true, false); break;                         \
+        case 2: LAUNCH(64, RAW_, false, true); break;                         \
+        case 3: LAUNCH(64, RAW_, true, true); break;                          \
+        case 4: LAUNCH(32, RAW_, false, false); break;                        \
+        case 5: LAUNCH(32, RAW_, true, false); break;                         \
+        case 6: LAUNCH(32, RAW_, false, true); break;                         \
+        case 7: LAUNCH(32, RAW_, true, true); break;                          \
+    }
+    if (raw_debug) { DISPATCH(true); } else { DISPATCH(false); }
+#undef DISPATCH
+#undef LAUNCH
+#undef ARGS
+}
+
+static int g_fwd_variant = 3;  // default: BM64 + STAGE_A + XCD_SWZ
+
+void set_fwd_variant(int v) { g_fwd_variant = v; }
+int get_fwd_variant() { return g_fwd_variant; }
+
 void launch_lstm_rec_fwd(
     const void* h_prev, long hp_stride, const void* w_t,
     const void* pre, long pre_stride, const float* c_prev, long cp_stride,
@@ -307,21 +432,9 @@ void launch_lstm_rec_fwd(
     void* acts, long act_stride, int B, int H, int raw_debug,
     hipStream_t stream)
 {
-    constexpr int BM = 64;
-    dim3 grid((B + BM - 1) / BM, H / 64);
-    dim3 block(256);
-    if (raw_debug)
-        hipLaunchKernelGGL((k_lstm_rec_fwd<BM, true>), grid, block, 0, stream,
-            (const __hip_bfloat16*)h_prev, hp_stride, (const __hip_bfloat16*)w_t,
-            (const __hip_bfloat16*)pre, pre_stride, c_prev, cp_stride,
-            (__hip_bfloat16*)h_out, ho_stride, c_out, co_stride,
-            (__hip_bfloat16*)acts, act_stride, B, H);
-    else
-        hipLaunchKernelGGL((k_lstm_rec_fwd<BM, false>), grid, block, 0, stream,
-            (const __hip_bfloat16*)h_prev, hp_stride, (const __hip_bfloat16*)w_t,
-            (const __hip_bfloat16*)pre, pre_stride, c_prev, cp_stride,
-            (__hip_bfloat16*)h_out, ho_stride, c_out, co_stride,
-            (__hip_bfloat16*)acts, act_stride, B, H);
+    launch_lstm_rec_fwd_v(h_prev, hp_stride, w_t, pre, pre_stride, c_prev,
+                          cp_stride, h_out, ho_stride, c_out, co_stride, acts,
+                          act_stride, B, H, raw_debug, g_fwd_variant, stream);
 }
 
 void launch_lstm_pw_bwd(
