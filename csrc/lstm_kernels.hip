@@ -67,7 +67,7 @@ __global__ __launch_bounds__(256) void k_lstm_rec_fwd(
 {
     constexpr int MF = BM / 16;
     constexpr int LDA = 72;  // 64 + 8 pad: conflict-free b128 column reads
-    __shared__ __hip_bfloat16 a_lds[STAGE_A ? BM : 1][STAGE_A ? LDA : 1];
+    __shared__ __hip_bfloat16 a_lds[STAGE_A ? 2 * BM : 1][STAGE_A ? LDA : 1];  // double-buffered
 
     const int tid  = threadIdx.x;
     const int wave = tid >> 6;
@@ -97,61 +97,78 @@ __global__ __launch_bounds__(256) void k_lstm_rec_fwd(
     const long koff0 = (long)lk * 8;
 
     if (STAGE_A) {
-        // BK = 64 K-steps; A staged in LDS, B register-prefetched one step
-        // ahead (fragment-shaped direct global loads, L2-resident W panel).
+        // BK=64 K-steps, double-buffered LDS A tile, ONE barrier per step.
+        // A and B are both register-prefetched one K-step ahead so the
+        // global-load latency hides under the previous step's MFMAs (T14
+        // issue-early / write-late).
         constexpr int BK = 64;
-        // staging map: thread -> (row = tid/8, col8 = tid%8); BM=64 needs
-        // 1 slice of 256 threads per 64x64 tile half... (BM*BK/8) / 256
-        constexpr int SLICES = (BM * BK / 8 + 255) / 256;
-        bf16x8 b_nxt[4][2];
-#pragma unroll
-        for (int g = 0; g < 4; ++g)
-#pragma unroll
-            for (int s = 0; s < 2; ++s)
-                b_nxt[g][s] = *(const bf16x8*)(
-                    w_t + ((long)g * H + n_base + lr) * H + s * 32 + koff0);
-        for (int k0 = 0; k0 < H; k0 += BK) {
-            // stage A tile (BM x BK) -- one 16B load+store per thread slice
-            __syncthreads();
+        constexpr int SLICES = BM * BK / 8 / 256;  // 16B loads per thread
+        static_assert(SLICES >= 1, "block too small for tile");
+        const int ar = tid >> 3, ac = (tid & 7) * 8;  // staging coords
+        const int nsteps = H / BK;
+
+        auto load_a = [&](int k0, bf16x8 (&a_reg)[SLICES]) {
 #pragma unroll
             for (int s = 0; s < SLICES; ++s) {
-                const int t = tid + s * 256;
-                const int ar = t >> 3, ac = (t & 7) * 8;
-                if (ar < BM) {
-                    const int row = m_base + ar;
-                    bf16x8 v = (row < B)
-                        ? *(const bf16x8*)(h_prev + (long)row * hp_stride + k0 + ac)
-                        : bf16x8{};
-                    *(bf16x8*)(&a_lds[ar][ac]) = v;
-                }
+                const int row = m_base + ar + s * (256 / 8);
+                a_reg[s] = (row < B)
+                    ? *(const bf16x8*)(h_prev + (long)row * hp_stride + k0 + ac)
+                    : bf16x8{};
             }
-            __syncthreads();
-            bf16x8 b_cur[4][2];
+        };
+        auto store_a = [&](int buf, const bf16x8 (&a_reg)[SLICES]) {
+#pragma unroll
+            for (int s = 0; s < SLICES; ++s)
+                *(bf16x8*)(&a_lds[buf * BM + ar + s * (256 / 8)][ac]) = a_reg[s];
+        };
+        auto load_b = [&](int k0, bf16x8 (&b_reg)[4][2]) {
 #pragma unroll
             for (int g = 0; g < 4; ++g)
 #pragma unroll
-                for (int s = 0; s < 2; ++s) b_cur[g][s] = b_nxt[g][s];
-            if (k0 + BK < H) {
-#pragma unroll
-                for (int g = 0; g < 4; ++g)
-#pragma unroll
-                    for (int s = 0; s < 2; ++s)
-                        b_nxt[g][s] = *(const bf16x8*)(
-                            w_t + ((long)g * H + n_base + lr) * H
-                            + k0 + BK + s * 32 + koff0);
-            }
+                for (int s = 0; s < 2; ++s)
+                    b_reg[g][s] = *(const bf16x8*)(
+                        w_t + ((long)g * H + n_base + lr) * H + k0 + s * 32 + koff0);
+        };
+
+        bf16x8 a_reg[SLICES], b_cur[4][2], b_nxt[4][2];
+        load_a(0, a_reg);
+        load_b(0, b_cur);
+        store_a(0, a_reg);
+        if (nsteps > 1) {
+            load_a(BK, a_reg);   // tile 1 -> a_reg (stored at bottom of j=0)
+            load_b(BK, b_nxt);
+        }
+        __syncthreads();
+        for (int j = 0; j < nsteps; ++j) {
+            const int buf = j & 1;
 #pragma unroll
             for (int s = 0; s < 2; ++s) {
                 bf16x8 a[MF];
 #pragma unroll
                 for (int mf = 0; mf < MF; ++mf)
-                    a[mf] = *(const bf16x8*)(&a_lds[mf * 16 + lr][s * 32 + lk * 8]);
+                    a[mf] = *(const bf16x8*)(
+                        &a_lds[buf * BM + mf * 16 + lr][s * 32 + lk * 8]);
 #pragma unroll
                 for (int g = 0; g < 4; ++g)
 #pragma unroll
                     for (int mf = 0; mf < MF; ++mf)
                         acc[g][mf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                             a[mf], b_cur[g][s], acc[g][mf], 0, 0, 0);
+            }
+            if (j + 1 < nsteps) {
+                // write tile j+1 (its loads were issued one MFMA phase ago,
+                // so the vmcnt wait here is already satisfied), then issue
+                // tile j+2's loads to hide under the NEXT MFMA phase.
+                store_a(buf ^ 1, a_reg);
+#pragma unroll
+                for (int g = 0; g < 4; ++g)
+#pragma unroll
+                    for (int s = 0; s < 2; ++s) b_cur[g][s] = b_nxt[g][s];
+                if (j + 2 < nsteps) {
+                    load_a((j + 2) * BK, a_reg);
+                    load_b((j + 2) * BK, b_nxt);
+                }
+                __syncthreads();
             }
         }
     } else {
