@@ -11,6 +11,7 @@
 #include <c10/hip/HIPStream.h>
 #include <hip/hip_runtime.h>
 
+#include <set>
 #include <vector>
 
 extern "C" {
@@ -29,11 +30,17 @@ void launch_adam(void*, float*, const void*, float*, float*, long, float,
                  float, float, float, float, float, float, int, hipStream_t);
 void set_fwd_variant(int);
 int get_fwd_variant();
+int launch_lstm_seq_fwd_persistent(const void*, const void*, const void*,
+                                   const float*, void*, float*, void*,
+                                   unsigned*, int, int, int, int, hipStream_t);
 }
 
 namespace {
 
 using at::Tensor;
+
+bool g_persistent_enabled = true;
+std::set<long> g_validated_grids;
 
 hipStream_t cur_stream() {
     return c10::hip::getCurrentHIPStream().stream();
@@ -77,6 +84,32 @@ std::vector<Tensor> lstm_seq_fwd(Tensor pre, Tensor w_h, Tensor h0, Tensor c0,
     auto* cs_p = (float*)c_seq.data_ptr();
     auto* act_p = (char*)acts.data_ptr();
     auto stream = cur_stream();
+
+    // Persistent whole-sequence path (W_h resident in LDS across all T
+    // steps, grid barrier per step).  First use of each grid shape is
+    // validated synchronously against the bounded-spin timeout flag.
+    if (g_persistent_enabled) {
+        auto cnt = at::zeros({T + 1}, pre.options().dtype(at::kInt));
+        int rc = launch_lstm_seq_fwd_persistent(
+            h0c.data_ptr(), w_t.data_ptr(), pre.data_ptr(),
+            (const float*)c0c.data_ptr(), h_seq.data_ptr(),
+            (float*)c_seq.data_ptr(), acts.data_ptr(),
+            (unsigned*)cnt.data_ptr(), (int)B, (int)H, (int)T, reverse ? 1 : 0,
+            stream);
+        if (rc == 0) {
+            const long key = (B / 256) * 1000000 + H;
+            if (!g_validated_grids.count(key)) {
+                const int flag = cnt[T].item<int>();  // syncs once per shape
+                TORCH_CHECK(flag == 0,
+                            "persistent LSTM kernel timed out (grid not "
+                            "resident?) B=", B, " H=", H, " T=", T);
+                g_validated_grids.insert(key);
+            }
+            const long last = reverse ? 0 : T - 1;
+            return {h_seq, c_seq, acts, h_seq.select(1, last).clone(),
+                    c_seq.select(1, last).clone()};
+        }
+    }
 
     const void* hp = h0c.data_ptr();
     long hp_stride = H;
@@ -273,6 +306,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           py::arg("c_seq"), py::arg("c0"), py::arg("reverse"));
     m.def("lstm_rec_fwd_step", &lstm_rec_fwd_step,
           "single fused recurrent step (raw_debug for GEMM-mapping tests)");
+    m.def("set_persistent_fwd", [](bool v) { g_persistent_enabled = v; });
+    m.def("get_persistent_fwd", []() { return g_persistent_enabled; });
     m.def("set_fwd_variant", &set_fwd_variant,
           "select rec_fwd kernel variant (bit0 STAGE_A, bit1 XCD_SWZ, bit2 BM32)");
     m.def("get_fwd_variant", &get_fwd_variant);

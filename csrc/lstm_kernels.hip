@@ -236,6 +236,192 @@ __global__ __launch_bounds__(256) void k_lstm_rec_fwd(
 }
 
 // ---------------------------------------------------------------------------
+// PERSISTENT whole-sequence forward: ONE kernel for the entire T-step loop.
+//
+// The recurrent weight slice stays RESIDENT in LDS for all T timesteps and
+// the cell state stays in registers; only h crosses steps (through L2,
+// published with an agent-scope release / acquire grid barrier per step --
+// guide §6 Guideline 16).  This is the SURVEY.md §7 "persistent-kernel
+// fusion across timesteps" design:
+//   - block = 256 threads = 4 waves; block tile = 256 batch rows x 16
+//     h-columns (= 64 gate-columns);  wave w owns rows [w*64, +64).
+//   - W_lds[64][H+8]: the block's 64 gate-columns of W_h^T, padded +8
+//     elements/row so ds_read_b128 B-fragments are bank-conflict-free
+//     (start banks 4*gcol mod 64, distinct within each 16-lane group).
+//   - c kept in registers (same lane owns the same (row,hcol) across steps
+//     because the C/D fragment map is step-invariant).
+//   - grid = (B/256) x (H/16) <= 256 blocks -> fully resident at 1
+//     block/CU (LDS-limited).  Natural dispatch places all n-blocks of one
+//     m-block on one XCD (grid.x = m), so each XCD's L2 caches its own
+//     256-row h panel.
+//   - per-step grid barrier: plain h stores -> per-wave vmcnt(0) drain ->
+//     __syncthreads -> one-lane agent RELEASE fence (+ restated asm wait,
+//     G16 pitfall 12) -> relaxed arrival count on cnt[t];  consumer polls
+//     relaxed with s_sleep + bounded spin (NO hang: on timeout sets
+//     cnt[T] flag and every block returns), then ONE agent ACQUIRE.
+// Eligibility: B % 256 == 0, H % 64 == 0, (B/256)*(H/16) <= 256,
+//              64*(H+8)*2 <= 160 KiB (H <= 1144).  Others use the
+//              per-step kernel path.
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(256) void k_lstm_seq_fwd_persistent(
+    const __hip_bfloat16* __restrict__ h0,      // (B,H)
+    const __hip_bfloat16* __restrict__ w_t,     // (4H,H) = W_h^T
+    const __hip_bfloat16* __restrict__ pre,     // (B,T,4H)
+    const float* __restrict__ c0,               // (B,H)
+    __hip_bfloat16* __restrict__ h_seq,         // (B,T,H)
+    float* __restrict__ c_seq,                  // (B,T,H)
+    __hip_bfloat16* __restrict__ acts,          // (B,T,4H)
+    unsigned* __restrict__ cnt,                 // [T+1] zeroed; [T] = timeout flag
+    int B, int H, int T, int reverse)
+{
+    extern __shared__ __attribute__((aligned(16))) char smem[];
+    __hip_bfloat16* w_lds = (__hip_bfloat16*)smem;   // [64][H+8]
+    const int ldw = H + 8;
+
+    const int tid = threadIdx.x;
+    const int wave = tid >> 6;
+    const int lane = tid & 63;
+    const int lr = lane & 15;
+    const int lk = lane >> 4;
+    const int m_blk = blockIdx.x;   // fast dim -> XCD gets one m-panel
+    const int n_blk = blockIdx.y;
+    const int n0 = n_blk * 16;                  // first h-column
+    const int m_base = m_blk * 256 + wave * 64; // wave's first batch row
+    const int nblocks = gridDim.x * gridDim.y;
+    const long HS = (long)T * H, AS = (long)T * 4 * H;  // row strides
+
+    // ---- fill the resident W slice: local gate-col gc = g*16+c  <-
+    //      global w_t row g*H + n0 + c, all K
+    for (int idx = tid; idx < 64 * (H / 8); idx += 256) {
+        const int gc = idx / (H / 8);
+        const int kc = (idx % (H / 8)) * 8;
+        const long grow = (long)(gc >> 4) * H + n0 + (gc & 15);
+        *(bf16x8*)(w_lds + gc * ldw + kc) = *(const bf16x8*)(w_t + grow * H + kc);
+    }
+
+    // ---- load this lane's resident cell state: rows mf*16 + lk*4 + r
+    const int hcol = n0 + lr;
+    float c_reg[4][4];
+#pragma unroll
+    for (int mf = 0; mf < 4; ++mf)
+#pragma unroll
+        for (int r = 0; r < 4; ++r)
+            c_reg[mf][r] = c0[(long)(m_base + mf * 16 + lk * 4 + r) * H + hcol];
+
+    __syncthreads();   // W_lds ready (block-local)
+
+    const long koff0 = (long)lk * 8;
+    int tmo = 0;
+    for (int step = 0; step < T; ++step) {
+        const int t = reverse ? (T - 1 - step) : step;
+        const __hip_bfloat16* hsrc;
+        long hstride;
+        if (step == 0) { hsrc = h0; hstride = H; }
+        else {
+            const int tp = reverse ? (t + 1) : (t - 1);
+            hsrc = h_seq + (long)tp * H;   // + row*HS below
+            hstride = HS;
+        }
+
+        f32x4 acc[4][4];
+#pragma unroll
+        for (int g = 0; g < 4; ++g)
+#pragma unroll
+            for (int mf = 0; mf < 4; ++mf) acc[g][mf] = f32x4{0, 0, 0, 0};
+
+        // K loop: A fragments direct (L2-resident h panel), B from LDS.
+        // A is register-prefetched one k-step ahead.
+        bf16x8 a_cur[4], a_nxt[4];
+#pragma unroll
+        for (int mf = 0; mf < 4; ++mf)
+            a_cur[mf] = *(const bf16x8*)(
+                hsrc + (long)(m_base + mf * 16 + lr) * hstride + koff0);
+        for (int k0 = 0; k0 < H; k0 += 32) {
+            if (k0 + 32 < H) {
+#pragma unroll
+                for (int mf = 0; mf < 4; ++mf)
+                    a_nxt[mf] = *(const bf16x8*)(
+                        hsrc + (long)(m_base + mf * 16 + lr) * hstride
+                        + k0 + 32 + koff0);
+            }
+            bf16x8 bfr[4];
+#pragma unroll
+            for (int g = 0; g < 4; ++g)
+                bfr[g] = *(const bf16x8*)(
+                    w_lds + (g * 16 + lr) * ldw + k0 + koff0);
+#pragma unroll
+            for (int g = 0; g < 4; ++g)
+#pragma unroll
+                for (int mf = 0; mf < 4; ++mf)
+                    acc[g][mf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        a_cur[mf], bfr[g], acc[g][mf], 0, 0, 0);
+#pragma unroll
+            for (int mf = 0; mf < 4; ++mf) a_cur[mf] = a_nxt[mf];
+        }
+
+        // ---- fused epilogue: pre + activations + in-register cell update
+        const __hip_bfloat16* pre_t = pre + (long)t * 4 * H;
+        __hip_bfloat16* h_t = h_seq + (long)t * H;
+        float* c_t = c_seq + (long)t * H;
+        __hip_bfloat16* act_t = acts + (long)t * 4 * H;
+#pragma unroll
+        for (int mf = 0; mf < 4; ++mf) {
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                const long row = m_base + mf * 16 + lk * 4 + r;
+                float raw[4];
+#pragma unroll
+                for (int g = 0; g < 4; ++g)
+                    raw[g] = acc[g][mf][r] +
+                             __bfloat162float(pre_t[row * AS + (long)g * H + hcol]);
+                const float gi = sigmoidf_(raw[0]);
+                const float gf = sigmoidf_(raw[1]);
+                const float gg = tanhf_(raw[2]);
+                const float go = sigmoidf_(raw[3]);
+                const float c = gf * c_reg[mf][r] + gi * gg;
+                c_reg[mf][r] = c;
+                const float h = go * tanhf_(c);
+                h_t[row * HS + hcol] = __float2bfloat16(h);
+                c_t[row * HS + hcol] = c;
+                __hip_bfloat16* ap = act_t + row * AS + hcol;
+                ap[0] = __float2bfloat16(gi);
+                ap[(long)H] = __float2bfloat16(gf);
+                ap[(long)2 * H] = __float2bfloat16(gg);
+                ap[(long)3 * H] = __float2bfloat16(go);
+            }
+        }
+
+        // ---- grid barrier: publish h_t, wait for all blocks (G16)
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");  // every wave drains
+        __syncthreads();
+        if (tid == 0) {
+            __builtin_amdgcn_fence(__ATOMIC_RELEASE, "agent");
+            asm volatile("s_waitcnt vmcnt(0)" ::: "memory");  // pitfall 12
+            __hip_atomic_fetch_add(&cnt[step], 1u, __ATOMIC_RELAXED,
+                                   __HIP_MEMORY_SCOPE_AGENT);
+            unsigned spins = 0;
+            while (__hip_atomic_load(&cnt[step], __ATOMIC_RELAXED,
+                                     __HIP_MEMORY_SCOPE_AGENT) < (unsigned)nblocks) {
+                if (__hip_atomic_load(&cnt[T], __ATOMIC_RELAXED,
+                                      __HIP_MEMORY_SCOPE_AGENT) != 0u) {
+                    tmo = 1;  // another block timed out: abort too
+                    break;
+                }
+                __builtin_amdgcn_s_sleep(8);
+                if (++spins > 10u * 1000u * 1000u) {  // a few seconds: give up
+                    __hip_atomic_store(&cnt[T], 1u, __ATOMIC_RELAXED,
+                                       __HIP_MEMORY_SCOPE_AGENT);
+                    tmo = 1;
+                    break;
+                }
+            }
+            __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
+        }
+        if (__syncthreads_count(tmo) != 0) return;  // a lane timed out: abort
+    }
+}
+
+// ---------------------------------------------------------------------------
 // K4: fused pointwise backward for one timestep (elementwise over B x H).
 // Produces raw-gate grads dgates_t (bf16) and the fp32 cell-grad carry
 // IN PLACE (dc reads+writes the same (row,col)).
@@ -441,6 +627,35 @@ static int g_fwd_variant = 3;  // default: BM64 + STAGE_A + XCD_SWZ
 
 void set_fwd_variant(int v) { g_fwd_variant = v; }
 int get_fwd_variant() { return g_fwd_variant; }
+
+// Persistent whole-sequence forward.  Returns 0 if launched, -1 if the
+// shape is ineligible (caller falls back to the per-step loop).
+int launch_lstm_seq_fwd_persistent(
+    const void* h0, const void* w_t, const void* pre, const float* c0,
+    void* h_seq, float* c_seq, void* acts, unsigned* cnt,
+    int B, int H, int T, int reverse, hipStream_t stream)
+{
+    if (B % 256 != 0 || H % 64 != 0) return -1;
+    const int nm = B / 256, nn = H / 16;
+    if (nm * nn > 256) return -1;
+    const size_t smem = (size_t)64 * (H + 8) * 2;
+    if (smem > 160 * 1024) return -1;
+    static size_t attr_set_for = 0;
+    if (smem > 64 * 1024 && attr_set_for < smem) {
+        if (hipFuncSetAttribute((const void*)k_lstm_seq_fwd_persistent,
+                                hipFuncAttributeMaxDynamicSharedMemorySize,
+                                (int)smem) != hipSuccess)
+            return -1;
+        attr_set_for = smem;
+    }
+    dim3 grid(nm, nn);  // m fastest: each XCD L2 caches one 256-row h panel
+    dim3 block(256);
+    hipLaunchKernelGGL(k_lstm_seq_fwd_persistent, grid, block, smem, stream,
+        (const __hip_bfloat16*)h0, (const __hip_bfloat16*)w_t,
+        (const __hip_bfloat16*)pre, c0, (__hip_bfloat16*)h_seq, c_seq,
+        (__hip_bfloat16*)acts, cnt, B, H, T, reverse);
+    return 0;
+}
 
 void launch_lstm_rec_fwd(
     const void* h_prev, long hp_stride, const void* w_t,

@@ -89,10 +89,22 @@ def main():
     pre = torch.randn(B, T, 4 * H, generator=g, device=dev).bfloat16()
     h0 = torch.randn(B, H, generator=g, device=dev).bfloat16()
     c0 = torch.randn(B, H, generator=g, device=dev)
-    ms = time_op(lambda: ext.lstm_seq_fwd(pre, w_h, h0, c0, False),
-                 max(args.iters // 10, 3))
-    report("seq_fwd", ms, rec_flops * T, B=B, H=H, T=T,
-           us_per_step=round(1e3 * ms / T, 2))
+    # correctness: persistent vs per-step path on identical inputs
+    ext.set_persistent_fwd(True)
+    outs_p = ext.lstm_seq_fwd(pre, w_h, h0, c0, False)
+    ext.set_persistent_fwd(False)
+    outs_s = ext.lstm_seq_fwd(pre, w_h, h0, c0, False)
+    perr = max((a.float() - b.float()).abs().max().item()
+               for a, b in zip(outs_p, outs_s))
+    for on in (False, True):
+        ext.set_persistent_fwd(on)
+        ms = time_op(lambda: ext.lstm_seq_fwd(pre, w_h, h0, c0, False),
+                     max(args.iters // 10, 3))
+        report(f"seq_fwd persistent={on}", ms, rec_flops * T, B=B, H=H, T=T,
+               us_per_step=round(1e3 * ms / T, 2),
+               vs_per_step_err=round(perr, 4) if on else None)
+    ext.set_persistent_fwd(True)
+    ms = ms
 
     # sequence backward
     h_seq, c_seq, acts, hT, cT = ext.lstm_seq_fwd(pre, w_h, h0, c0, False)
