@@ -27,9 +27,18 @@ typedef float f32x4 __attribute__((ext_vector_type(4)));
 
 #define DEVINL __device__ __forceinline__
 
-DEVINL float sigmoidf_(float x) { return 1.0f / (1.0f + __expf(-x)); }
-// tanh via exp2-free formulation: tanhf is fine on VALU (transcendental)
-DEVINL float tanhf_(float x) { return tanhf(x); }
+// Fast activations: v_exp + v_rcp (1-ulp) -- bf16 outputs cannot see the
+// difference vs libcalls, and __ocml_tanh_f32 is a branchy libcall.
+DEVINL float rcp_(float x) { return __builtin_amdgcn_rcpf(x); }
+DEVINL float sigmoidf_(float x) {
+    x = fminf(fmaxf(x, -30.f), 30.f);
+    return rcp_(1.0f + __expf(-x));
+}
+DEVINL float tanhf_(float x) {
+    x = fminf(fmaxf(x, -15.f), 15.f);
+    const float e = __expf(2.0f * x);
+    return (e - 1.0f) * rcp_(e + 1.0f);
+}
 
 // ---------------------------------------------------------------------------
 // K1b+K2: fused recurrent GEMM + LSTM pointwise, one timestep.

@@ -84,6 +84,11 @@ def main():
     print(json.dumps({"best_variant": best[0], "ms": round(best[1], 4)}),
           flush=True)
     ext.set_fwd_variant(best[0] if best[0] is not None else default_variant)
+    # ablation: RAW_DEBUG = GEMM + pre-add + acts stores, NO activations/
+    # cell update/h/c stores -- isolates the epilogue cost
+    report("rec_fwd_step RAW(no-epilogue)", time_op(
+        lambda: ext.lstm_rec_fwd_step(h_prev, w_h, pre1, c_prev, True),
+        args.iters), rec_flops, B=B, H=H)
 
     # sequence forward (transpose amortized over T steps)
     pre = torch.randn(B, T, 4 * H, generator=g, device=dev).bfloat16()
