@@ -63,7 +63,7 @@ DEVINL float tanhf_(float x) {
 //   XCD_SWZ: bijective blockIdx remap so blocks sharing a W_h panel
 //            (same n-block) land on the same XCD's private L2 (T1).
 // ---------------------------------------------------------------------------
-template <int BM, bool RAW_DEBUG, bool STAGE_A, bool XCD_SWZ>
+template <int BM, bool RAW_DEBUG, int STAGE, bool XCD_SWZ>
 __global__ __launch_bounds__(256) void k_lstm_rec_fwd(
     const __hip_bfloat16* __restrict__ h_prev, long hp_stride,
     const __hip_bfloat16* __restrict__ w_t,     // (4H, H) row-major = W_h^T
@@ -75,9 +75,6 @@ __global__ __launch_bounds__(256) void k_lstm_rec_fwd(
     int B, int H)
 {
     constexpr int MF = BM / 16;
-    constexpr int LDA = 72;  // 64 + 8 pad: conflict-free b128 column reads
-    __shared__ __hip_bfloat16 a_lds[STAGE_A ? 2 * BM : 1][STAGE_A ? LDA : 1];  // double-buffered
-
     const int tid  = threadIdx.x;
     const int wave = tid >> 6;
     const int lane = tid & 63;
@@ -105,7 +102,85 @@ __global__ __launch_bounds__(256) void k_lstm_rec_fwd(
 
     const long koff0 = (long)lk * 8;
 
-    if (STAGE_A) {
+    if constexpr (STAGE == 2) {
+        // All-glds 2-phase: BOTH tiles staged by global_load_lds into a
+        // double-buffered dynamic-LDS image (no VGPR staging, no register
+        // double-buffer copies -- the fix for the accvgpr-shuffle stall).
+        // Images are lane-linear (glds writes base + lane*16); bank
+        // conflicts on the b128 fragment reads are broken by XOR-ing the
+        // 16B-chunk index with (row & 7) on the SOURCE address (rule 21).
+        constexpr int BK = 64;
+        constexpr int ASLICES = BM / 32;       // A tile = BM x 64 x 2B
+        constexpr int ABYTES = BM * 128;
+        extern __shared__ __attribute__((aligned(16))) char smem[];
+        char* const a_base = smem;                       // [2][ABYTES]
+        char* const b_base = smem + 2 * ABYTES;          // [2][32768]
+        const int nsteps = H / BK;
+        const long n_col0 = (long)n_blk * 64;
+
+        auto stage = [&](int buf, int k0) {
+#pragma unroll
+            for (int s2 = 0; s2 < ASLICES; ++s2) {
+                const int ci = s2 * 256 + tid;
+                const int row = ci >> 3, ch = ci & 7;
+                int grow = m_base + row;
+                grow = grow < B ? grow : B - 1;  // clamp: extra rows unused
+                const void* src = h_prev + (long)grow * hp_stride + k0
+                                  + ((ch ^ (row & 7)) * 8);
+                __builtin_amdgcn_global_load_lds(
+                    (const __attribute__((address_space(1))) void*)src,
+                    (__attribute__((address_space(3))) void*)(
+                        a_base + buf * ABYTES + s2 * 4096 + wave * 1024),
+                    16, 0, 0);
+            }
+#pragma unroll
+            for (int s2 = 0; s2 < 8; ++s2) {
+                const int ci = s2 * 256 + tid;
+                const int gc = ci >> 3, ch = ci & 7;
+                const long grow = (long)(gc >> 6) * H + n_col0 + (gc & 63);
+                const void* src = w_t + grow * H + k0 + ((ch ^ (gc & 7)) * 8);
+                __builtin_amdgcn_global_load_lds(
+                    (const __attribute__((address_space(1))) void*)src,
+                    (__attribute__((address_space(3))) void*)(
+                        b_base + buf * 32768 + s2 * 4096 + wave * 1024),
+                    16, 0, 0);
+            }
+        };
+
+        stage(0, 0);
+        __syncthreads();  // carries the vmcnt(0) glds drain
+        for (int j = 0; j < nsteps; ++j) {
+            const int buf = j & 1;
+            if (j + 1 < nsteps) stage(buf ^ 1, (j + 1) * BK);  // issue early
+            const char* ab = a_base + buf * ABYTES;
+            const char* bb = b_base + buf * 32768;
+#pragma unroll
+            for (int s = 0; s < 2; ++s) {
+                bf16x8 a[MF], bfr[4];
+#pragma unroll
+                for (int mf = 0; mf < MF; ++mf) {
+                    const int row = mf * 16 + lr;
+                    a[mf] = *(const bf16x8*)(
+                        ab + row * 128 + (((s * 4 + lk) ^ (row & 7)) * 16));
+                }
+#pragma unroll
+                for (int g = 0; g < 4; ++g) {
+                    const int gc = g * 64 + wave * 16 + lr;
+                    bfr[g] = *(const bf16x8*)(
+                        bb + gc * 128 + (((s * 4 + lk) ^ (gc & 7)) * 16));
+                }
+#pragma unroll
+                for (int g = 0; g < 4; ++g)
+#pragma unroll
+                    for (int mf = 0; mf < MF; ++mf)
+                        acc[g][mf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                            a[mf], bfr[g], acc[g][mf], 0, 0, 0);
+            }
+            __syncthreads();  // drains next tile's glds + protects buf reuse
+        }
+    } else if constexpr (STAGE == 1) {
+        constexpr int LDA = 72;  // 64 + 8 pad: conflict-free b128 reads
+        __shared__ __hip_bfloat16 a_lds[2 * BM][LDA];  // double-buffered
         // BK=64 K-steps, double-buffered LDS A tile, ONE barrier per step.
         // A and B are both register-prefetched one K-step ahead so the
         // global-load latency hides under the previous step's MFMAs (T14
@@ -165,9 +240,8 @@ __global__ __launch_bounds__(256) void k_lstm_rec_fwd(
                             a[mf], b_cur[g][s], acc[g][mf], 0, 0, 0);
             }
             if (j + 1 < nsteps) {
-                // write tile j+1 (its loads were issued one MFMA phase ago,
-                // so the vmcnt wait here is already satisfied), then issue
-                // tile j+2's loads to hide under the NEXT MFMA phase.
+                // write tile j+1 (loads issued one MFMA phase ago), then
+                // issue tile j+2's loads to hide under the NEXT MFMA phase.
                 store_a(buf ^ 1, a_reg);
 #pragma unroll
                 for (int g = 0; g < 4; ++g)
@@ -596,7 +670,8 @@ __global__ __launch_bounds__(256) void k_adam(
 // ---------------------------------------------------------------------------
 extern "C" {
 
-// variant encoding: bit0 = STAGE_A, bit1 = XCD_SWZ, bit2 = BM32 (else BM64)
+// variant encoding: (v & 3) = STAGE {0 direct, 1 LDS-A, 2 all-glds},
+// bit2 = BM32 (else BM64), bit3 = XCD_SWZ
 void launch_lstm_rec_fwd_v(
     const void* h_prev, long hp_stride, const void* w_t,
     const void* pre, long pre_stride, const float* c_prev, long cp_stride,
@@ -605,26 +680,44 @@ void launch_lstm_rec_fwd_v(
     hipStream_t stream)
 {
     const int BM = (variant & 4) ? 32 : 64;
+    const int stage = variant & 3;
     dim3 grid((B + BM - 1) / BM, H / 64);
     dim3 block(256);
+    const size_t smem = (stage == 2) ? (size_t)(2 * BM * 128 + 2 * 32768) : 0;
 #define ARGS                                                                  \
     (const __hip_bfloat16*)h_prev, hp_stride, (const __hip_bfloat16*)w_t,     \
     (const __hip_bfloat16*)pre, pre_stride, c_prev, cp_stride,                \
     (__hip_bfloat16*)h_out, ho_stride, c_out, co_stride,                      \
     (__hip_bfloat16*)acts, act_stride, B, H
 #define LAUNCH(BM_, RAW_, ST_, SW_)                                           \
-    hipLaunchKernelGGL((k_lstm_rec_fwd<BM_, RAW_, ST_, SW_>), grid, block, 0, \
-                       stream, ARGS)
+    do {                                                                      \
+        if (ST_ == 2) {                                                       \
+            static bool attr_done_##BM_##RAW_##SW_ = false;                   \
+            if (!attr_done_##BM_##RAW_##SW_ && smem > 64 * 1024) {            \
+                hipFuncSetAttribute(                                          \
+                    (const void*)(k_lstm_rec_fwd<BM_, RAW_, ST_, SW_>),       \
+                    hipFuncAttributeMaxDynamicSharedMemorySize, (int)smem);   \
+                attr_done_##BM_##RAW_##SW_ = true;                            \
+            }                                                                 \
+        }                                                                     \
+        hipLaunchKernelGGL((k_lstm_rec_fwd<BM_, RAW_, ST_, SW_>), grid,       \
+                           block, smem, stream, ARGS);                        \
+    } while (0)
 #define DISPATCH(RAW_)                                                        \
-    switch (variant & 7) {                                                    \
-        case 0: LAUNCH(64, RAW_, false, false); break;                        \
-        case 1: LAUNCH(64, RAW_, true, false); break;                         \
-        case 2: LAUNCH(64, RAW_, false, true); break;                         \
-        case 3: LAUNCH(64, RAW_, true, true); break;                          \
-        case 4: LAUNCH(32, RAW_, false, false); break;                        \
-        case 5: LAUNCH(32, RAW_, true, false); break;                         \
-        case 6: LAUNCH(32, RAW_, false, true); break;                         \
-        case 7: LAUNCH(32, RAW_, true, true); break;                          \
+    switch (variant & 15) {                                                   \
+        case 0: LAUNCH(64, RAW_, 0, false); break;                            \
+        case 1: LAUNCH(64, RAW_, 1, false); break;                            \
+        case 2: LAUNCH(64, RAW_, 2, false); break;                            \
+        case 4: LAUNCH(32, RAW_, 0, false); break;                            \
+        case 5: LAUNCH(32, RAW_, 1, false); break;                            \
+        case 6: LAUNCH(32, RAW_, 2, false); break;                            \
+        case 8: LAUNCH(64, RAW_, 0, true); break;                             \
+        case 9: LAUNCH(64, RAW_, 1, true); break;                             \
+        case 10: LAUNCH(64, RAW_, 2, true); break;                            \
+        case 12: LAUNCH(32, RAW_, 0, true); break;                            \
+        case 13: LAUNCH(32, RAW_, 1, true); break;                            \
+        case 14: LAUNCH(32, RAW_, 2, true); break;                            \
+        default: LAUNCH(64, RAW_, 2, true); break;                            \
     }
     if (raw_debug) { DISPATCH(true); } else { DISPATCH(false); }
 #undef DISPATCH
@@ -632,7 +725,7 @@ void launch_lstm_rec_fwd_v(
 #undef ARGS
 }
 
-static int g_fwd_variant = 3;  // default: BM64 + STAGE_A + XCD_SWZ
+static int g_fwd_variant = 10;  // default: BM64 + all-glds + XCD_SWZ
 
 void set_fwd_variant(int v) { g_fwd_variant = v; }
 int get_fwd_variant() { return g_fwd_variant; }
