@@ -39,7 +39,7 @@ namespace {
 
 using at::Tensor;
 
-bool g_persistent_enabled = true;
+bool g_persistent_enabled = false;  // per-step glds path is faster (kbench6)
 std::set<long> g_validated_grids;
 
 hipStream_t cur_stream() {

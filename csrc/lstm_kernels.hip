@@ -725,7 +725,7 @@ void launch_lstm_rec_fwd_v(
 #undef ARGS
 }
 
-static int g_fwd_variant = 10;  // default: BM64 + all-glds + XCD_SWZ
+static int g_fwd_variant = 14;  // default: BM32 + all-glds + XCD_SWZ
 
 void set_fwd_variant(int v) { g_fwd_variant = v; }
 int get_fwd_variant() { return g_fwd_variant; }
