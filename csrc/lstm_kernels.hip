@@ -725,7 +725,8 @@ void launch_lstm_rec_fwd_v(
 #undef ARGS
 }
 
-static int g_fwd_variant = 14;  // default: BM32 + all-glds + XCD_SWZ
+static int g_fwd_variant = -1;  // -1 = auto: pick by batch size (kbench
+// sweeps: B >= 2048 -> BM64+glds+swz (10); smaller -> BM32+glds+swz (14))
 
 void set_fwd_variant(int v) { g_fwd_variant = v; }
 int get_fwd_variant() { return g_fwd_variant; }
@@ -766,9 +767,11 @@ void launch_lstm_rec_fwd(
     void* acts, long act_stride, int B, int H, int raw_debug,
     hipStream_t stream)
 {
+    int v = g_fwd_variant;
+    if (v < 0) v = (B >= 2048) ? 10 : 14;
     launch_lstm_rec_fwd_v(h_prev, hp_stride, w_t, pre, pre_stride, c_prev,
                           cp_stride, h_out, ho_stride, c_out, co_stride, acts,
-                          act_stride, B, H, raw_debug, g_fwd_variant, stream);
+                          act_stride, B, H, raw_debug, v, stream);
 }
 
 void launch_lstm_pw_bwd(
