@@ -30,6 +30,10 @@ void launch_adam(void*, float*, const void*, float*, float*, long, float,
                  float, float, float, float, float, float, int, hipStream_t);
 void set_fwd_variant(int);
 int get_fwd_variant();
+void launch_xent_fwd(const void*, const long*, float*, float*, long, int,
+                     hipStream_t);
+void launch_xent_bwd(const void*, const long*, const float*, void*,
+                     const float*, float, long, int, hipStream_t);
 int launch_lstm_seq_fwd_persistent(const void*, const void*, const void*,
                                    const float*, void*, float*, void*,
                                    unsigned*, int, int, int, int, hipStream_t);
@@ -234,6 +238,44 @@ std::vector<Tensor> lstm_seq_bwd(c10::optional<Tensor> dh_seq_opt,
 }
 
 // ---------------------------------------------------------------------
+// K3: fused log-softmax cross-entropy (LM head, vocab up to 50k+)
+// ---------------------------------------------------------------------
+std::vector<Tensor> xent_fwd(Tensor logits, Tensor targets) {
+    check_bf16_cuda(logits, "logits");
+    TORCH_CHECK(logits.dim() == 2 && targets.dim() == 1, "logits (N,V), targets (N,)");
+    TORCH_CHECK(targets.scalar_type() == at::kLong, "targets int64");
+    logits = logits.contiguous();
+    auto tgt = targets.contiguous();
+    const long N = logits.size(0);
+    const int V = (int)logits.size(1);
+    auto loss = at::empty({N}, logits.options().dtype(at::kFloat));
+    auto lse = at::empty({N}, logits.options().dtype(at::kFloat));
+    launch_xent_fwd(logits.data_ptr(), (const long*)tgt.data_ptr(),
+                    (float*)loss.data_ptr(), (float*)lse.data_ptr(), N, V,
+                    cur_stream());
+    return {loss, lse};
+}
+
+Tensor xent_bwd(Tensor logits, Tensor targets, Tensor lse,
+                c10::optional<Tensor> grad_out) {
+    logits = logits.contiguous();
+    auto tgt = targets.contiguous();
+    const long N = logits.size(0);
+    const int V = (int)logits.size(1);
+    auto dlogits = at::empty_like(logits);
+    const float* gp = nullptr;
+    Tensor g;
+    if (grad_out.has_value()) {
+        g = grad_out->to(at::kFloat).contiguous();
+        gp = (const float*)g.data_ptr();
+    }
+    launch_xent_bwd(logits.data_ptr(), (const long*)tgt.data_ptr(),
+                    (const float*)lse.data_ptr(), dlogits.data_ptr(),
+                    gp, 1.0f / (float)N, N, V, cur_stream());
+    return dlogits;
+}
+
+// ---------------------------------------------------------------------
 // K6/K7: multi-tensor clip + optimizers
 // ---------------------------------------------------------------------
 Tensor multi_tensor_l2norm(std::vector<Tensor> grads) {
@@ -311,6 +353,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("set_fwd_variant", &set_fwd_variant,
           "select rec_fwd kernel variant (bit0 STAGE_A, bit1 XCD_SWZ, bit2 BM32)");
     m.def("get_fwd_variant", &get_fwd_variant);
+    m.def("xent_fwd", &xent_fwd);
+    m.def("xent_bwd", &xent_bwd);
     m.def("multi_tensor_l2norm", &multi_tensor_l2norm);
     m.def("multi_tensor_clip_scale", &multi_tensor_clip_scale);
     m.def("fused_sgd", &fused_sgd);

@@ -848,3 +848,118 @@ void launch_adam(void* p, float* m, const void* g, float* ea, float* eas,
 }
 
 }  // extern "C"
+
+// ---------------------------------------------------------------------------
+// K3: fused log-softmax + cross-entropy for the LM head (V up to 50k+).
+// Forward: one workgroup per row, single online pass (running max + sum-exp
+// with rescale), vectorized bf16x8 loads; writes per-row loss and lse.
+// Backward: dlogits = (exp(l - lse) - onehot) * row_scale, grid-stride.
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(256) void k_xent_fwd(
+    const __hip_bfloat16* __restrict__ logits,  // (N, V)
+    const long* __restrict__ targets,           // (N,)
+    float* __restrict__ loss,                   // (N,)
+    float* __restrict__ lse,                    // (N,)
+    long N, int V)
+{
+    const long row = blockIdx.x;
+    if (row >= N) return;
+    const __hip_bfloat16* lp = logits + row * V;
+    const int tid = threadIdx.x;
+
+    float m = -3.0e38f, s = 0.f;
+    const int VV = V & ~7;  // vectorized body
+    for (int v0 = tid * 8; v0 < VV; v0 += 256 * 8) {
+        bf16x8 x = *(const bf16x8*)(lp + v0);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+            const float xv = (float)x[j];
+            if (xv > m) { s *= __expf(m - xv); m = xv; }
+            s += __expf(xv - m);
+        }
+    }
+    for (int v = VV + tid; v < V; v += 256) {
+        const float xv = __bfloat162float(lp[v]);
+        if (xv > m) { s *= __expf(m - xv); m = xv; }
+        s += __expf(xv - m);
+    }
+    // wave reduce (running-max merge), then cross-wave via LDS
+    __shared__ float red_m[4], red_s[4];
+#pragma unroll
+    for (int off = 32; off > 0; off >>= 1) {
+        const float om = __shfl_down(m, off, 64);
+        const float os = __shfl_down(s, off, 64);
+        if (om > m) { s = s * __expf(m - om) + os; m = om; }
+        else        { s = s + os * __expf(om - m); }
+    }
+    const int lane = tid & 63, wave = tid >> 6;
+    if (lane == 0) { red_m[wave] = m; red_s[wave] = s; }
+    __syncthreads();
+    if (tid == 0) {
+        float M = red_m[0], S = red_s[0];
+#pragma unroll
+        for (int w = 1; w < 4; ++w) {
+            const float om = red_m[w], os = red_s[w];
+            if (om > M) { S = S * __expf(M - om) + os; M = om; }
+            else        { S = S + os * __expf(om - M); }
+        }
+        const float l = M + __logf(S);
+        lse[row] = l;
+        loss[row] = l - __bfloat162float(lp[targets[row]]);
+    }
+}
+
+__global__ __launch_bounds__(256) void k_xent_bwd(
+    const __hip_bfloat16* __restrict__ logits, const long* __restrict__ targets,
+    const float* __restrict__ lse, __hip_bfloat16* __restrict__ dlogits,
+    const float* __restrict__ gscale, float inv_n, long N, int V)
+{
+    const long row = blockIdx.y;
+    if (row >= N) return;
+    const float scale = (gscale ? *gscale : 1.0f) * inv_n;
+    const float l = lse[row];
+    const long tgt = targets[row];
+    const __hip_bfloat16* lp = logits + row * V;
+    __hip_bfloat16* dp = dlogits + row * V;
+    for (int v0 = (blockIdx.x * 256 + threadIdx.x) * 8; v0 < V;
+         v0 += gridDim.x * 256 * 8) {
+        if (v0 + 8 <= V) {
+            bf16x8 x = *(const bf16x8*)(lp + v0);
+            bf16x8 o;
+#pragma unroll
+            for (int j = 0; j < 8; ++j) {
+                float p = __expf((float)x[j] - l);
+                if (v0 + j == tgt) p -= 1.0f;
+                o[j] = (__bf16)(p * scale);
+            }
+            *(bf16x8*)(dp + v0) = o;
+        } else {
+            for (int v = v0; v < V; ++v) {
+                float p = __expf(__bfloat162float(lp[v]) - l);
+                if (v == tgt) p -= 1.0f;
+                dp[v] = __float2bfloat16(p * scale);
+            }
+        }
+    }
+}
+
+extern "C" {
+void launch_xent_fwd(const void* logits, const long* targets, float* loss,
+                     float* lse, long N, int V, hipStream_t stream)
+{
+    dim3 grid((unsigned)N), block(256);
+    hipLaunchKernelGGL(k_xent_fwd, grid, block, 0, stream,
+                       (const __hip_bfloat16*)logits, targets, loss, lse, N, V);
+}
+void launch_xent_bwd(const void* logits, const long* targets, const float* lse,
+                     void* dlogits, const float* gscale, float inv_n, long N,
+                     int V, hipStream_t stream)
+{
+    int gx = (V / 8 + 255) / 256;
+    gx = gx > 16 ? 16 : (gx < 1 ? 1 : gx);
+    dim3 grid(gx, (unsigned)N), block(256);
+    hipLaunchKernelGGL(k_xent_bwd, grid, block, 0, stream,
+                       (const __hip_bfloat16*)logits, targets, lse,
+                       (__hip_bfloat16*)dlogits, gscale, inv_n, N, V);
+}
+}  // extern "C"

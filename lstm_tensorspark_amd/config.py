@@ -47,6 +47,10 @@ class TrainConfig:
     ckpt_interval: int = 0        # 0 => only at end if ckpt_dir set
     ckpt_dir: str = ""
     seed: int = 1234
+    # fault injection (failure-detection tests): raise on this rank at this
+    # step; -1 = disabled
+    fault_step: int = -1
+    fault_rank: int = 0
 
 
 @dataclass

@@ -22,6 +22,7 @@ import torch.nn.functional as F
 
 from ..config import ModelConfig
 from ..ops.lstm_layer import lstm_layer
+from ..ops import xent
 
 
 def _pname(layer: int, rev: bool, which: str) -> str:
@@ -152,10 +153,12 @@ class LSTMModel(nn.Module):
     def loss(self, output: torch.Tensor, y: torch.Tensor) -> torch.Tensor:
         head = self.cfg.head
         if head == "classifier":
-            return F.cross_entropy(output.float(), y)
+            return xent.cross_entropy(output, y)
         if head in ("seq_classifier", "lm"):
             C = output.shape[-1]
-            return F.cross_entropy(output.reshape(-1, C).float(), y.reshape(-1))
+            # fused K3 kernel on GPU bf16 (one online-softmax pass, no fp32
+            # (N,V) intermediate); torch CE elsewhere
+            return xent.cross_entropy(output.reshape(-1, C), y.reshape(-1))
         # regression on features
         return F.mse_loss(output.float(), y.float())
 

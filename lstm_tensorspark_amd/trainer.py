@@ -83,6 +83,9 @@ class Trainer:
 
     def train_step(self, batch) -> dict:
         cfg = self.cfg
+        if cfg.train.fault_step == self.step and cfg.train.fault_rank == self.ctx.rank:
+            raise RuntimeError(
+                f"injected fault on rank {self.ctx.rank} at step {self.step}")
         x, y = batch
         if self.bucketer is not None:
             self.bucketer.zero_grad()
@@ -114,8 +117,39 @@ class Trainer:
         return {"loss": total_loss,
                 "grad_norm": float(norm) if norm is not None else None}
 
+    def resume_latest(self) -> bool:
+        """Auto-resume: load the newest checkpoint in ckpt_dir, if any.
+        Part of the failure-recovery story: this engine is
+        single-node/fixed-world -- a failed rank fail-fasts the whole job
+        (ProcessGroup timeout is the watchdog) and the relaunched job calls
+        resume_latest() to continue from the last saved step."""
+        d = self.cfg.train.ckpt_dir
+        if not d or not os.path.isdir(d):
+            return False
+        cands = [os.path.join(d, f) for f in os.listdir(d) if f.endswith(".pt")]
+        if not cands:
+            return False
+        latest = max(cands, key=os.path.getmtime)
+        self.load(latest)
+        return True
+
     def fit(self, steps: Optional[int] = None) -> list:
-        """Train for `steps` optimizer steps (default: config.train.steps)."""
+        """Train for `steps` optimizer steps (default: config.train.steps).
+
+        Fail-fast: any exception tears down the process group before
+        re-raising so surviving ranks see a clean abort instead of a hang.
+        """
+        try:
+            return self._fit(steps)
+        except Exception:
+            if dist.is_initialized():
+                try:
+                    dist.destroy_process_group()
+                except Exception:
+                    pass
+            raise
+
+    def _fit(self, steps: Optional[int] = None) -> list:
         cfg = self.cfg
         steps = steps if steps is not None else cfg.train.steps
         t_last = time.perf_counter()

@@ -86,3 +86,31 @@ def test_mismatched_names_rejected(tmp_path):
     import pytest
     with pytest.raises(ValueError):
         load_checkpoint(p, t2.model)
+
+
+def test_fault_injection_and_auto_resume(tmp_path):
+    """Failure-detection tier: a rank fault aborts the run; a relaunched
+    trainer auto-resumes from the newest checkpoint and reproduces the
+    uninterrupted run bitwise."""
+    import pytest
+
+    # uninterrupted oracle
+    cfgA = small_cfg(tmp_path / "a", ckpt_interval=2, steps=6)
+    tA = Trainer(cfgA)
+    tA.fit(6)
+
+    # faulting run: dies at step 5 (after the step-4 checkpoint)
+    cfgB = small_cfg(tmp_path / "b", ckpt_interval=2, steps=6)
+    cfgB.train.fault_step = 5
+    tB = Trainer(cfgB)
+    with pytest.raises(RuntimeError, match="injected fault"):
+        tB.fit(6)
+    assert tB.step == 5  # 5 steps completed (fault raised entering step 6)
+
+    # relaunch + auto-resume from latest checkpoint, finish the job
+    cfgC = small_cfg(tmp_path / "b", ckpt_interval=2, steps=6)
+    tC = Trainer(cfgC)
+    assert tC.resume_latest()
+    tC.fit(6 - tC.step)
+    for k, vA in tA.model.weight_dict().items():
+        assert torch.equal(vA, tC.model.weight_dict()[k]), k

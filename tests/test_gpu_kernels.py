@@ -241,3 +241,40 @@ def test_hip_path_actually_dispatches():
     assert extension.use_hip(torch.device("cuda"), torch.bfloat16)
     so = extension.get_extension().__file__
     assert "lstm_tensorspark_amd" in so  # in-tree .so, not site-packages
+
+
+# ---------------------------------------------------------------- K3 xent
+@pytest.mark.parametrize("N,V", [(64, 512), (32, 50000), (16, 1000 + 3)])
+def test_fused_xent_vs_torch(ext, N, V):
+    import torch.nn.functional as F
+
+    g = torch.Generator(device="cpu").manual_seed(0)
+    logits = (torch.randn(N, V, generator=g) * 2).to("cuda").bfloat16()
+    targets = torch.randint(0, V, (N,), generator=g).to("cuda")
+    lg = logits.clone().requires_grad_(True)
+    from lstm_tensorspark_amd.ops.xent import FusedCrossEntropyFn
+
+    loss = FusedCrossEntropyFn.apply(lg, targets)
+    loss.backward()
+
+    lr = logits.float().clone().requires_grad_(True)
+    loss_ref = F.cross_entropy(lr, targets)
+    loss_ref.backward()
+
+    assert abs(loss.item() - loss_ref.item()) < 2e-2 * max(1.0, abs(loss_ref.item()))
+    err = (lg.grad.float() - lr.grad).abs().max().item()
+    assert err < 2e-3, err
+
+
+def test_fused_xent_grad_scale(ext):
+    """Upstream grad != 1 must scale dlogits on-device."""
+    from lstm_tensorspark_amd.ops.xent import FusedCrossEntropyFn
+
+    g = torch.Generator(device="cpu").manual_seed(1)
+    logits = torch.randn(8, 128, generator=g).to("cuda").bfloat16()
+    targets = torch.randint(0, 128, (8,), generator=g).to("cuda")
+    l1 = logits.clone().requires_grad_(True)
+    FusedCrossEntropyFn.apply(l1, targets).mul(3.0).backward()
+    l2 = logits.clone().requires_grad_(True)
+    FusedCrossEntropyFn.apply(l2, targets).backward()
+    assert (l1.grad.float() - 3.0 * l2.grad.float()).abs().max().item() < 1e-2
