@@ -178,6 +178,81 @@ __global__ __launch_bounds__(256) void k_lstm_rec_fwd(
             }
             __syncthreads();  // drains next tile's glds + protects buf reuse
         }
+    } else if constexpr (STAGE == 3) {
+        // Counted-vmcnt 3-buffer glds pipeline (T3+T4): BK=32 tiles, two
+        // tiles in flight across RAW barriers (never vmcnt(0) in the main
+        // loop), 60 KiB LDS at BM=64 -> 2 blocks/CU.  Per-wave glds per
+        // tile: A = BM/64, B = 4.
+        constexpr int BK = 32;
+        constexpr int ABYTES = BM * BK * 2;          // 4 KiB at BM=64
+        constexpr int BBYTES = 256 * BK * 2;         // 16 KiB
+        constexpr int ASLICES = ABYTES / 4096;       // glds instrs (>=1)
+        constexpr int TILE_GLDS = (ASLICES < 1 ? 1 : ASLICES) + 4;
+        extern __shared__ __attribute__((aligned(16))) char smem[];
+        const int nsteps = H / BK;
+        const long n_col0 = (long)n_blk * 64;
+
+        auto stage = [&](int buf, int k0) {
+            char* const ab = smem + buf * (ABYTES + BBYTES);
+            char* const bb = ab + ABYTES;
+#pragma unroll
+            for (int s2 = 0; s2 < (ASLICES < 1 ? 1 : ASLICES); ++s2) {
+                const int ci = s2 * 256 + tid;
+                const int row = ci >> 2, ch = ci & 3;   // 4 chunks of 16B per 64-byte row
+                int grow = m_base + row;
+                grow = grow < B ? grow : B - 1;
+                const void* src = h_prev + (long)grow * hp_stride + k0
+                                  + ((ch ^ (row & 3)) * 8);
+                __builtin_amdgcn_global_load_lds(
+                    (const __attribute__((address_space(1))) void*)src,
+                    (__attribute__((address_space(3))) void*)(
+                        ab + s2 * 4096 + wave * 1024), 16, 0, 0);
+            }
+#pragma unroll
+            for (int s2 = 0; s2 < 4; ++s2) {
+                const int ci = s2 * 256 + tid;
+                const int gc = ci >> 2, ch = ci & 3;
+                const long grow = (long)(gc >> 6) * H + n_col0 + (gc & 63);
+                const void* src = w_t + grow * H + k0 + ((ch ^ (gc & 3)) * 8);
+                __builtin_amdgcn_global_load_lds(
+                    (const __attribute__((address_space(1))) void*)src,
+                    (__attribute__((address_space(3))) void*)(
+                        bb + s2 * 4096 + wave * 1024), 16, 0, 0);
+            }
+        };
+
+        stage(0, 0);
+        if (nsteps > 1) stage(1, BK);
+        for (int j = 0; j < nsteps; ++j) {
+            const int buf = j % 3;
+            // tile j landed when at most the NEXT tile's glds are pending
+            if (j + 1 < nsteps)
+                asm volatile("s_waitcnt vmcnt(%0)" :: "i"(TILE_GLDS) : "memory");
+            else
+                asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+            __builtin_amdgcn_s_barrier();
+            // issue tile j+2 into the third buffer (not in use by j, j+1)
+            if (j + 2 < nsteps) stage((j + 2) % 3, (j + 2) * BK);
+            const char* ab = smem + buf * (ABYTES + BBYTES);
+            const char* bb = ab + ABYTES;
+            bf16x8 a[MF], bfr[4];
+#pragma unroll
+            for (int mf = 0; mf < MF; ++mf) {
+                const int row = mf * 16 + lr;
+                a[mf] = *(const bf16x8*)(ab + row * 64 + ((lk ^ (row & 3)) * 16));
+            }
+#pragma unroll
+            for (int g = 0; g < 4; ++g) {
+                const int gc = g * 64 + wave * 16 + lr;
+                bfr[g] = *(const bf16x8*)(bb + gc * 64 + ((lk ^ (gc & 3)) * 16));
+            }
+#pragma unroll
+            for (int g = 0; g < 4; ++g)
+#pragma unroll
+                for (int mf = 0; mf < MF; ++mf)
+                    acc[g][mf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        a[mf], bfr[g], acc[g][mf], 0, 0, 0);
+        }
     } else if constexpr (STAGE == 1) {
         constexpr int LDA = 72;  // 64 + 8 pad: conflict-free b128 reads
         __shared__ __hip_bfloat16 a_lds[2 * BM][LDA];  // double-buffered
@@ -683,7 +758,9 @@ void launch_lstm_rec_fwd_v(
     const int stage = variant & 3;
     dim3 grid((B + BM - 1) / BM, H / 64);
     dim3 block(256);
-    const size_t smem = (stage == 2) ? (size_t)(2 * BM * 128 + 2 * 32768) : 0;
+    size_t smem = 0;
+    if (stage == 2) smem = (size_t)(2 * BM * 128 + 2 * 32768);
+    if (stage == 3) smem = (size_t)(3 * (BM * 64 + 16384));
 #define ARGS                                                                  \
     (const __hip_bfloat16*)h_prev, hp_stride, (const __hip_bfloat16*)w_t,     \
     (const __hip_bfloat16*)pre, pre_stride, c_prev, cp_stride,                \
@@ -708,15 +785,19 @@ void launch_lstm_rec_fwd_v(
         case 0: LAUNCH(64, RAW_, 0, false); break;                            \
         case 1: LAUNCH(64, RAW_, 1, false); break;                            \
         case 2: LAUNCH(64, RAW_, 2, false); break;                            \
+        case 3: LAUNCH(64, RAW_, 3, false); break;                            \
         case 4: LAUNCH(32, RAW_, 0, false); break;                            \
         case 5: LAUNCH(32, RAW_, 1, false); break;                            \
         case 6: LAUNCH(32, RAW_, 2, false); break;                            \
+        case 7: LAUNCH(32, RAW_, 3, false); break;                            \
         case 8: LAUNCH(64, RAW_, 0, true); break;                             \
         case 9: LAUNCH(64, RAW_, 1, true); break;                             \
         case 10: LAUNCH(64, RAW_, 2, true); break;                            \
+        case 11: LAUNCH(64, RAW_, 3, true); break;                            \
         case 12: LAUNCH(32, RAW_, 0, true); break;                            \
         case 13: LAUNCH(32, RAW_, 1, true); break;                            \
         case 14: LAUNCH(32, RAW_, 2, true); break;                            \
+        case 15: LAUNCH(32, RAW_, 3, true); break;                            \
         default: LAUNCH(64, RAW_, 2, true); break;                            \
     }
     if (raw_debug) { DISPATCH(true); } else { DISPATCH(false); }

@@ -86,7 +86,8 @@ class LSTMLayerFn(torch.autograd.Function):
         hprev = _hprev_seq(h_seq, h0, ctx.reverse).reshape(B * T, H)
         dw_h = (hprev.t() @ dg_flat).to(w_h.dtype)
         dw_x = (x.reshape(B * T, X).t() @ dg_flat).to(w_x.dtype)
-        db = dg_flat.to(functional._acc_dtype(dg_flat.dtype)).sum(dim=0).to(w_x.dtype)
+        # sum with fp32 accumulation WITHOUT materializing an fp32 copy
+        db = dg_flat.sum(dim=0, dtype=functional._acc_dtype(dg_flat.dtype)).to(w_x.dtype)
         dx = (dg_flat @ w_x.t()).view(B, T, X)
         return dx, dh0, dc0, dw_x, dw_h, db, None
 
