@@ -74,9 +74,10 @@ def main():
     gen.manual_seed(1234 + ctx.rank)
     pool = []
     for _ in range(4):
-        x = torch.randn(B, T, cfg.model.input_size, generator=gen, device=dev,
+        # TIME-MAJOR (T, B, ·): the engine's native layout
+        x = torch.randn(T, B, cfg.model.input_size, generator=gen, device=dev,
                         dtype=torch.float32).to(dt)
-        y = torch.randint(0, cfg.model.num_classes, (B, T), generator=gen,
+        y = torch.randint(0, cfg.model.num_classes, (T, B), generator=gen,
                           device=dev)
         pool.append((x, y))
 

@@ -34,8 +34,8 @@ void launch_xent_fwd(const void*, const long*, float*, float*, long, int,
                      hipStream_t);
 void launch_xent_bwd(const void*, const long*, const float*, void*,
                      const float*, float, long, int, hipStream_t);
-int launch_lstm_seq_fwd_persistent(const void*, const void*, const void*,
-                                   const float*, void*, float*, void*,
+int launch_lstm_seq_fwd_persistent(void*, const void*, const void*,
+                                   const float*, float*, void*,
                                    unsigned*, int, int, int, int, hipStream_t);
 }
 
@@ -56,8 +56,12 @@ void check_bf16_cuda(const Tensor& t, const char* name) {
 }
 
 // ---------------------------------------------------------------------
-// Forward: whole-sequence driver.  pre (B,T,4H) bf16, w_h (H,4H) bf16,
-// h0 (B,H) bf16, c0 (B,H) fp32 -> (h_seq, c_seq, acts, hT, cT)
+// Forward: whole-sequence driver.  TIME-MAJOR: pre (T,B,4H) bf16,
+// w_h (H,4H) bf16, h0 (B,H) bf16, c0 (B,H) fp32
+// -> (h_all (T+1,B,H), c_seq (T,B,H), acts (T,B,4H), hT, cT).
+// h_all slot convention (ops/functional.py): fwd h_all[0]=h0,
+// h_all[t+1]=h_t; rev h_all[T]=h0, h_all[t]=h_t — so h_{t-1} per step AND
+// the layer output are both contiguous leading-dim slices (no cat/copies).
 // ---------------------------------------------------------------------
 std::vector<Tensor> lstm_seq_fwd(Tensor pre, Tensor w_h, Tensor h0, Tensor c0,
                                  bool reverse) {
@@ -65,8 +69,8 @@ std::vector<Tensor> lstm_seq_fwd(Tensor pre, Tensor w_h, Tensor h0, Tensor c0,
     check_bf16_cuda(w_h, "w_h");
     check_bf16_cuda(h0, "h0");
     TORCH_CHECK(c0.scalar_type() == at::kFloat && c0.is_cuda(), "c0 fp32 GPU");
-    TORCH_CHECK(pre.dim() == 3, "pre must be (B,T,4H)");
-    const long B = pre.size(0), T = pre.size(1), H4 = pre.size(2);
+    TORCH_CHECK(pre.dim() == 3, "pre must be (T,B,4H)");
+    const long T = pre.size(0), B = pre.size(1), H4 = pre.size(2);
     const long H = H4 / 4;
     TORCH_CHECK(w_h.size(0) == H && w_h.size(1) == H4, "w_h must be (H,4H)");
     TORCH_CHECK(H % 64 == 0, "fused kernel requires H % 64 == 0, got ", H);
@@ -79,12 +83,13 @@ std::vector<Tensor> lstm_seq_fwd(Tensor pre, Tensor w_h, Tensor h0, Tensor c0,
 
     auto opts_bf = pre.options();
     auto opts_f32 = pre.options().dtype(at::kFloat);
-    auto h_seq = at::empty({B, T, H}, opts_bf);
-    auto c_seq = at::empty({B, T, H}, opts_f32);
-    auto acts = at::empty({B, T, H4}, opts_bf);
+    auto h_all = at::empty({T + 1, B, H}, opts_bf);
+    auto c_seq = at::empty({T, B, H}, opts_f32);
+    auto acts = at::empty({T, B, H4}, opts_bf);
+    h_all.select(0, reverse ? T : 0).copy_(h0c, /*non_blocking=*/true);
 
     auto* pre_p = (const char*)pre.data_ptr();
-    auto* hs_p = (char*)h_seq.data_ptr();
+    auto* ha_p = (char*)h_all.data_ptr();
     auto* cs_p = (float*)c_seq.data_ptr();
     auto* act_p = (char*)acts.data_ptr();
     auto stream = cur_stream();
@@ -95,8 +100,8 @@ std::vector<Tensor> lstm_seq_fwd(Tensor pre, Tensor w_h, Tensor h0, Tensor c0,
     if (g_persistent_enabled) {
         auto cnt = at::zeros({T + 1}, pre.options().dtype(at::kInt));
         int rc = launch_lstm_seq_fwd_persistent(
-            h0c.data_ptr(), w_t.data_ptr(), pre.data_ptr(),
-            (const float*)c0c.data_ptr(), h_seq.data_ptr(),
+            h_all.data_ptr(), w_t.data_ptr(), pre.data_ptr(),
+            (const float*)c0c.data_ptr(),
             (float*)c_seq.data_ptr(), acts.data_ptr(),
             (unsigned*)cnt.data_ptr(), (int)B, (int)H, (int)T, reverse ? 1 : 0,
             stream);
@@ -109,38 +114,37 @@ std::vector<Tensor> lstm_seq_fwd(Tensor pre, Tensor w_h, Tensor h0, Tensor c0,
                             "resident?) B=", B, " H=", H, " T=", T);
                 g_validated_grids.insert(key);
             }
-            const long last = reverse ? 0 : T - 1;
-            return {h_seq, c_seq, acts, h_seq.select(1, last).clone(),
-                    c_seq.select(1, last).clone()};
+            auto hT = h_all.select(0, reverse ? 0 : T).clone();
+            auto cT = c_seq.select(0, reverse ? 0 : T - 1).clone();
+            return {h_all, c_seq, acts, hT, cT};
         }
     }
 
-    const void* hp = h0c.data_ptr();
-    long hp_stride = H;
-    const float* cp = (const float*)c0c.data_ptr();
-    long cp_stride = H;
-    const long hs_stride = T * H, cs_stride = T * H;
-    const long pre_stride = T * H4, act_stride = T * H4;
-
     for (long k = 0; k < T; ++k) {
         const long t = reverse ? (T - 1 - k) : k;
+        // prev slot holds h_{t-1} (h0 at the first iteration by the slot
+        // convention); out slot receives h_t.
+        const long slot_prev = reverse ? (t + 1) : t;
+        const long slot_out = reverse ? t : (t + 1);
+        const float* cp;
+        if (k == 0) {
+            cp = (const float*)c0c.data_ptr();
+        } else {
+            const long t_prev = reverse ? (t + 1) : (t - 1);
+            cp = cs_p + t_prev * B * H;
+        }
         launch_lstm_rec_fwd(
-            hp, hp_stride, w_t.data_ptr(),
-            pre_p + 2 * t * H4, pre_stride,  // bf16 = 2 bytes
-            cp, cp_stride,
-            hs_p + 2 * t * H, hs_stride,
-            cs_p + t * H, cs_stride,
-            act_p + 2 * t * H4, act_stride,
+            ha_p + 2 * slot_prev * B * H, H, w_t.data_ptr(),
+            pre_p + 2 * t * B * H4, H4,  // bf16 = 2 bytes
+            cp, H,
+            ha_p + 2 * slot_out * B * H, H,
+            cs_p + t * B * H, H,
+            act_p + 2 * t * B * H4, H4,
             (int)B, (int)H, /*raw_debug=*/0, stream);
-        hp = hs_p + 2 * t * H;
-        hp_stride = hs_stride;
-        cp = cs_p + t * H;
-        cp_stride = cs_stride;
     }
-    const long last = reverse ? 0 : T - 1;
-    auto hT = h_seq.select(1, last).clone();
-    auto cT = c_seq.select(1, last).clone();
-    return {h_seq, c_seq, acts, hT, cT};
+    auto hT = h_all.select(0, reverse ? 0 : T).clone();
+    auto cT = c_seq.select(0, reverse ? 0 : T - 1).clone();
+    return {h_all, c_seq, acts, hT, cT};
 }
 
 // Single fused step with optional raw-gate debug output (numerics tests).
@@ -171,8 +175,9 @@ std::vector<Tensor> lstm_seq_bwd(c10::optional<Tensor> dh_seq_opt,
                                  c10::optional<Tensor> d_cT_opt,
                                  Tensor w_h, Tensor acts, Tensor c_seq,
                                  Tensor c0, bool reverse) {
+    // TIME-MAJOR: acts (T,B,4H), c_seq (T,B,H), dh_seq (T,B,H).
     check_bf16_cuda(acts, "acts");
-    const long B = acts.size(0), T = acts.size(1), H4 = acts.size(2);
+    const long T = acts.size(0), B = acts.size(1), H4 = acts.size(2);
     const long H = H4 / 4;
     TORCH_CHECK(H % 4 == 0, "H % 4 != 0");
     acts = acts.contiguous();
@@ -188,7 +193,7 @@ std::vector<Tensor> lstm_seq_bwd(c10::optional<Tensor> dh_seq_opt,
     const bool has_cT = d_cT_opt.has_value();
     if (has_cT) d_cT = d_cT_opt->contiguous();
 
-    auto dgates = at::empty({B, T, H4}, acts.options());
+    auto dgates = at::empty({T, B, H4}, acts.options());
     auto dc = at::empty({B, H}, acts.options().dtype(at::kFloat));
     auto dh_buf = at::empty({B, H}, acts.options());
     // W_h^T view for the recurrent grad GEMM (rocBLAS handles the transpose)
@@ -199,20 +204,16 @@ std::vector<Tensor> lstm_seq_bwd(c10::optional<Tensor> dh_seq_opt,
     auto* dg_p = (char*)dgates.data_ptr();
     auto* up_p = has_up ? (const char*)dh_seq.data_ptr() : nullptr;
     auto stream = cur_stream();
-    const long act_stride = T * H4, cs_stride = T * H, up_stride = T * H;
 
     for (long k = T - 1; k >= 0; --k) {
         const long t = reverse ? (T - 1 - k) : k;
         const long t_prev = reverse ? (T - k) : (k - 1);  // index of c_{t-1}
         const bool first = (k == T - 1);
         const float* cprev_p;
-        long cprev_stride;
         if (k > 0) {
-            cprev_p = cs_p + t_prev * H;
-            cprev_stride = cs_stride;
+            cprev_p = cs_p + t_prev * B * H;
         } else {
             cprev_p = (const float*)c0c.data_ptr();
-            cprev_stride = H;
         }
         const void* carry_p = nullptr;
         long carry_stride = H;
@@ -222,17 +223,17 @@ std::vector<Tensor> lstm_seq_bwd(c10::optional<Tensor> dh_seq_opt,
             carry_p = dh_buf.data_ptr();
         }
         launch_lstm_pw_bwd(
-            has_up ? up_p + 2 * t * H : nullptr, up_stride,
+            has_up ? up_p + 2 * t * B * H : nullptr, H,
             carry_p, carry_stride,
-            act_p + 2 * t * H4, act_stride,
-            cs_p + t * H, cs_stride, cprev_p, cprev_stride,
-            dg_p + 2 * t * H4, act_stride,
+            act_p + 2 * t * B * H4, H4,
+            cs_p + t * B * H, H, cprev_p, H,
+            dg_p + 2 * t * B * H4, H4,
             (float*)dc.data_ptr(), H, (int)B, (int)H,
             first ? 1 : 0,
             (first && has_cT) ? (const float*)d_cT.data_ptr() : nullptr, H,
             stream);
         // recurrent hidden-grad GEMM [K5a]: dh_{t-1} = dgates_t @ W_h^T
-        at::mm_out(dh_buf, dgates.select(1, t), w_h_t);
+        at::mm_out(dh_buf, dgates.select(0, t), w_h_t);
     }
     return {dgates, dh_buf, dc};  // (dgates, dh0, dc0)
 }

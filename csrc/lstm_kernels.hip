@@ -422,13 +422,13 @@ __global__ __launch_bounds__(256) void k_lstm_rec_fwd(
 //              per-step kernel path.
 // ---------------------------------------------------------------------------
 __global__ __launch_bounds__(256) void k_lstm_seq_fwd_persistent(
-    const __hip_bfloat16* __restrict__ h0,      // (B,H)
+    __hip_bfloat16* __restrict__ h_all,         // (T+1,B,H); h0 pre-written
+                                                // at slot 0 (fwd) / T (rev)
     const __hip_bfloat16* __restrict__ w_t,     // (4H,H) = W_h^T
-    const __hip_bfloat16* __restrict__ pre,     // (B,T,4H)
+    const __hip_bfloat16* __restrict__ pre,     // (T,B,4H)
     const float* __restrict__ c0,               // (B,H)
-    __hip_bfloat16* __restrict__ h_seq,         // (B,T,H)
-    float* __restrict__ c_seq,                  // (B,T,H)
-    __hip_bfloat16* __restrict__ acts,          // (B,T,4H)
+    float* __restrict__ c_seq,                  // (T,B,H)
+    __hip_bfloat16* __restrict__ acts,          // (T,B,4H)
     unsigned* __restrict__ cnt,                 // [T+1] zeroed; [T] = timeout flag
     int B, int H, int T, int reverse)
 {
@@ -446,7 +446,8 @@ __global__ __launch_bounds__(256) void k_lstm_seq_fwd_persistent(
     const int n0 = n_blk * 16;                  // first h-column
     const int m_base = m_blk * 256 + wave * 64; // wave's first batch row
     const int nblocks = gridDim.x * gridDim.y;
-    const long HS = (long)T * H, AS = (long)T * 4 * H;  // row strides
+    // TIME-MAJOR: per-step blocks are dense; row stride H (h/c), 4H (pre/acts)
+    const long BH = (long)B * H, BH4 = (long)B * 4 * H;
 
     // ---- fill the resident W slice: local gate-col gc = g*16+c  <-
     //      global w_t row g*H + n0 + c, all K
@@ -472,14 +473,10 @@ __global__ __launch_bounds__(256) void k_lstm_seq_fwd_persistent(
     int tmo = 0;
     for (int step = 0; step < T; ++step) {
         const int t = reverse ? (T - 1 - step) : step;
-        const __hip_bfloat16* hsrc;
-        long hstride;
-        if (step == 0) { hsrc = h0; hstride = H; }
-        else {
-            const int tp = reverse ? (t + 1) : (t - 1);
-            hsrc = h_seq + (long)tp * H;   // + row*HS below
-            hstride = HS;
-        }
+        // slot convention: h_prev(t) at slot t (fwd) / t+1 (rev); h0 was
+        // pre-written at its slot by the driver, so no special-casing.
+        const __hip_bfloat16* hsrc = h_all + (long)(reverse ? t + 1 : t) * BH;
+        const long hstride = H;
 
         f32x4 acc[4][4];
 #pragma unroll
@@ -518,10 +515,10 @@ __global__ __launch_bounds__(256) void k_lstm_seq_fwd_persistent(
         }
 
         // ---- fused epilogue: pre + activations + in-register cell update
-        const __hip_bfloat16* pre_t = pre + (long)t * 4 * H;
-        __hip_bfloat16* h_t = h_seq + (long)t * H;
-        float* c_t = c_seq + (long)t * H;
-        __hip_bfloat16* act_t = acts + (long)t * 4 * H;
+        const __hip_bfloat16* pre_t = pre + (long)t * BH4;
+        __hip_bfloat16* h_t = h_all + (long)(reverse ? t : t + 1) * BH;
+        float* c_t = c_seq + (long)t * BH;
+        __hip_bfloat16* act_t = acts + (long)t * BH4;
 #pragma unroll
         for (int mf = 0; mf < 4; ++mf) {
 #pragma unroll
@@ -531,7 +528,7 @@ __global__ __launch_bounds__(256) void k_lstm_seq_fwd_persistent(
 #pragma unroll
                 for (int g = 0; g < 4; ++g)
                     raw[g] = acc[g][mf][r] +
-                             __bfloat162float(pre_t[row * AS + (long)g * H + hcol]);
+                             __bfloat162float(pre_t[row * 4 * H + (long)g * H + hcol]);
                 const float gi = sigmoidf_(raw[0]);
                 const float gf = sigmoidf_(raw[1]);
                 const float gg = tanhf_(raw[2]);
@@ -539,9 +536,9 @@ __global__ __launch_bounds__(256) void k_lstm_seq_fwd_persistent(
                 const float c = gf * c_reg[mf][r] + gi * gg;
                 c_reg[mf][r] = c;
                 const float h = go * tanhf_(c);
-                h_t[row * HS + hcol] = __float2bfloat16(h);
-                c_t[row * HS + hcol] = c;
-                __hip_bfloat16* ap = act_t + row * AS + hcol;
+                h_t[row * H + hcol] = __float2bfloat16(h);
+                c_t[row * H + hcol] = c;
+                __hip_bfloat16* ap = act_t + row * 4 * H + hcol;
                 ap[0] = __float2bfloat16(gi);
                 ap[(long)H] = __float2bfloat16(gf);
                 ap[(long)2 * H] = __float2bfloat16(gg);
@@ -815,8 +812,8 @@ int get_fwd_variant() { return g_fwd_variant; }
 // Persistent whole-sequence forward.  Returns 0 if launched, -1 if the
 // shape is ineligible (caller falls back to the per-step loop).
 int launch_lstm_seq_fwd_persistent(
-    const void* h0, const void* w_t, const void* pre, const float* c0,
-    void* h_seq, float* c_seq, void* acts, unsigned* cnt,
+    void* h_all, const void* w_t, const void* pre, const float* c0,
+    float* c_seq, void* acts, unsigned* cnt,
     int B, int H, int T, int reverse, hipStream_t stream)
 {
     if (B % 256 != 0 || H % 64 != 0) return -1;
@@ -835,8 +832,8 @@ int launch_lstm_seq_fwd_persistent(
     dim3 grid(nm, nn);  // m fastest: each XCD L2 caches one 256-row h panel
     dim3 block(256);
     hipLaunchKernelGGL(k_lstm_seq_fwd_persistent, grid, block, smem, stream,
-        (const __hip_bfloat16*)h0, (const __hip_bfloat16*)w_t,
-        (const __hip_bfloat16*)pre, c0, (__hip_bfloat16*)h_seq, c_seq,
+        (__hip_bfloat16*)h_all, (const __hip_bfloat16*)w_t,
+        (const __hip_bfloat16*)pre, c0, c_seq,
         (__hip_bfloat16*)acts, cnt, B, H, T, reverse);
     return 0;
 }

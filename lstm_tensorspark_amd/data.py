@@ -44,23 +44,24 @@ class SyntheticSeqData:
         return g
 
     def batch(self, step: int):
-        """Return (x, y) for a training step.
+        """Return (x, y) for a training step (TIME-MAJOR).
 
-        x: (B, T, input_size) float for feature tasks, (B, T) long for "lm".
-        y: per-head target -- see models/heads.
+        x: (T, B, input_size) float for feature tasks, (T, B) long for "lm".
+        y: per-head target -- (T, B) for per-step heads, (B,) for
+           "classifier" -- see models/heads.
         """
         g = self._gen(step)
         B, T = self.batch_size, self.seq_len
         task = self.task
         if task == "random":
-            x = torch.randn(B, T, self.input_size, generator=g)
+            x = torch.randn(T, B, self.input_size, generator=g)
             head = self.cfg.model.head
             if head == "classifier":
                 y = torch.randint(0, self.cfg.model.num_classes, (B,), generator=g)
             elif head == "seq_classifier":
-                y = torch.randint(0, self.cfg.model.num_classes, (B, T), generator=g)
+                y = torch.randint(0, self.cfg.model.num_classes, (T, B), generator=g)
             else:
-                y = torch.randn(B, T, self.cfg.model.hidden_size, generator=g)
+                y = torch.randn(T, B, self.cfg.model.hidden_size, generator=g)
             return (x.to(self.device, self.dtype), y.to(self.device))
         if task == "copy":
             # Learnable task: predict the input class seen `delay` steps ago.
@@ -68,18 +69,18 @@ class SyntheticSeqData:
             # sequence (first `delay` targets are class 0 = "blank").
             C = self.cfg.model.num_classes
             d = self.cfg.data.copy_delay
-            toks = torch.randint(1, min(C, self.input_size), (B, T), generator=g)
-            x = torch.zeros(B, T, self.input_size)
+            toks = torch.randint(1, min(C, self.input_size), (T, B), generator=g)
+            x = torch.zeros(T, B, self.input_size)
             x.scatter_(2, toks.unsqueeze(-1), 1.0)
-            y = torch.zeros(B, T, dtype=torch.long)
+            y = torch.zeros(T, B, dtype=torch.long)
             if d < T:
-                y[:, d:] = toks[:, : T - d]
+                y[d:] = toks[: T - d]
             return (x.to(self.device, self.dtype), y.to(self.device))
         if task == "lm":
             V = self.cfg.model.vocab_size
             assert V > 1, "lm task requires model.vocab_size"
-            stream = torch.randint(0, V, (B, T + 1), generator=g)
-            x, y = stream[:, :-1], stream[:, 1:]
+            stream = torch.randint(0, V, (T + 1, B), generator=g)
+            x, y = stream[:-1], stream[1:]
             return (x.to(self.device), y.to(self.device))
         raise ValueError(f"unknown data task {task!r}")
 

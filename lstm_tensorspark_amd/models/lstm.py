@@ -95,7 +95,7 @@ class LSTMModel(nn.Module):
 
     # -- forward -----------------------------------------------------------
     def _layer_io(self, x, layer: int, rev: bool, state=None):
-        B = x.shape[0]
+        B = x.shape[1]
         H = self.cfg.hidden_size
         w_x = self.weight(_pname(layer, rev, "W_x"))
         w_h = self.weight(_pname(layer, rev, "W_h"))
@@ -110,10 +110,11 @@ class LSTMModel(nn.Module):
     def backbone(self, x: torch.Tensor, state: Optional[list] = None):
         """Run embedding (if LM) + all LSTM layers.
 
-        x: (B,T,input_size) features, or (B,T) long tokens for LM.
+        x: (T,B,input_size) features, or (T,B) long tokens for LM
+           (TIME-MAJOR -- the engine's native layout, see ops/functional.py).
         state: per-(layer,dir) list of (h0, c0) carry (truncated BPTT), or
                None for zero state.
-        Returns (features (B,T,H*D), new_state list).
+        Returns (features (T,B,H*D), new_state list).
         """
         cfg = self.cfg
         if cfg.head == "lm":
@@ -136,8 +137,8 @@ class LSTMModel(nn.Module):
     def forward(self, x, state: Optional[list] = None):
         """Returns (output, new_state).
 
-        output: logits -- (B,C) for "classifier", (B,T,C) for
-        "seq_classifier", (B,T,V) for "lm", features (B,T,H*D) for "none".
+        output: logits -- (B,C) for "classifier", (T,B,C) for
+        "seq_classifier", (T,B,V) for "lm", features (T,B,H*D) for "none".
         """
         feats, new_state = self.backbone(x, state)
         cfg = self.cfg
@@ -145,9 +146,9 @@ class LSTMModel(nn.Module):
             return feats, new_state
         w, b = self.weight("head/W"), self.weight("head/b")
         if cfg.head == "classifier":
-            return torch.addmm(b, feats[:, -1], w), new_state
-        B, T, F_ = feats.shape
-        logits = torch.addmm(b, feats.reshape(B * T, F_), w).view(B, T, -1)
+            return torch.addmm(b, feats[-1], w), new_state
+        T, B, F_ = feats.shape
+        logits = torch.addmm(b, feats.reshape(T * B, F_), w).view(T, B, -1)
         return logits, new_state
 
     def loss(self, output: torch.Tensor, y: torch.Tensor) -> torch.Tensor:

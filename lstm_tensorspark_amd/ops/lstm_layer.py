@@ -5,7 +5,7 @@ This is the MI355X replacement for the reference's per-timestep TF graph ops
 timestep replayed by Session.run, the whole-sequence forward is
 
   1. ONE library GEMM (hipBLASLt via torch.matmul) for the input projection
-     of ALL timesteps:  pre = x.view(B*T, X) @ W_x + b          [K1a]
+     of ALL timesteps:  pre = x.view(T*B, X) @ W_x + b          [K1a]
   2. a C++-driven time loop of ONE fused hand-written CDNA4 kernel per step:
      MFMA recurrent GEMM h_{t-1} @ W_h + gate activations + cell-state
      update + h/c/act stores, all fused                          [K1b+K2]
@@ -16,6 +16,13 @@ and the backward is
   4. THREE library GEMMs for the weight/input gradients:
      dW_h = hprev^T @ dgates, dW_x = x^T @ dgates, dx = dgates @ W_x^T
      (batched over all timesteps -- no per-step dW round trips)  [K5b]
+
+Everything is TIME-MAJOR (T, B, ·): the whole-sequence hidden state lives
+in ONE (T+1, B, H) buffer holding h0 at slot 0 (forward) or slot T
+(reverse), so h_{t-1} for the dW_h GEMM is a contiguous leading-dim slice
+-- no torch.cat, no padding, zero copies (functional.py documents the slot
+convention).  The layer's (T,B,H) output is itself a contiguous view of
+that buffer, so the next layer's input GEMM reshapes it for free.
 
 On CPU the reference implementations in functional.py run instead (same
 signatures, same precision policy), so the identical Trainer code path is
@@ -28,6 +35,7 @@ from typing import Optional
 import torch
 
 from . import extension, functional
+from .functional import h_prev_view, h_seq_view
 
 
 def _seq_fwd(pre, w_h, h0, c0, reverse, use_hip):
@@ -44,34 +52,26 @@ def _seq_bwd(d_hseq, d_hT, d_cT, w_h, acts, c_seq, c0, reverse, use_hip):
     return functional.ref_seq_bwd(d_hseq, d_hT, d_cT, w_h, acts, c_seq, c0, reverse)
 
 
-def _hprev_seq(h_seq, h0, reverse):
-    """h_{t-1} per step in natural time order (h0 enters at the first step
-    of the iteration order)."""
-    if reverse:
-        return torch.cat([h_seq[:, 1:], h0.unsqueeze(1)], dim=1)
-    return torch.cat([h0.unsqueeze(1), h_seq[:, :-1]], dim=1)
-
-
 class LSTMLayerFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, h0, c0, w_x, w_h, b, reverse: bool):
-        # x: (B,T,X) compute dtype; h0: (B,H); c0: (B,H) fp32
-        B, T, X = x.shape
+        # x: (T,B,X) compute dtype; h0: (B,H); c0: (B,H) fp32
+        T, B, X = x.shape
         H = w_h.shape[0]
         ctx.set_materialize_grads(False)
         use_hip = extension.use_hip(x.device, x.dtype)
         # [K1a] input projection for all timesteps in one library GEMM
-        pre = torch.addmm(b.to(x.dtype), x.reshape(B * T, X), w_x).view(B, T, 4 * H)
-        h_seq, c_seq, acts, hT, cT = _seq_fwd(pre, w_h, h0, c0, reverse, use_hip)
-        ctx.save_for_backward(x, h0, c0, w_x, w_h, h_seq, c_seq, acts)
+        pre = torch.addmm(b.to(x.dtype), x.reshape(T * B, X), w_x).view(T, B, 4 * H)
+        h_all, c_seq, acts, hT, cT = _seq_fwd(pre, w_h, h0, c0, reverse, use_hip)
+        ctx.save_for_backward(x, c0, w_x, w_h, h_all, c_seq, acts)
         ctx.reverse = reverse
         ctx.use_hip = use_hip
-        return h_seq, hT, cT
+        return h_seq_view(h_all, reverse), hT, cT
 
     @staticmethod
     def backward(ctx, d_hseq, d_hT, d_cT):
-        x, h0, c0, w_x, w_h, h_seq, c_seq, acts = ctx.saved_tensors
-        B, T, X = x.shape
+        x, c0, w_x, w_h, h_all, c_seq, acts = ctx.saved_tensors
+        T, B, X = x.shape
         H = w_h.shape[0]
         # set_materialize_grads(False): unused outputs arrive as None and the
         # backward kernels treat null pointers as zero.
@@ -81,14 +81,14 @@ class LSTMLayerFn(torch.autograd.Function):
             d_cT.contiguous() if d_cT is not None else None,
             w_h, acts, c_seq, c0, ctx.reverse, ctx.use_hip,
         )
-        dg_flat = dgates.reshape(B * T, 4 * H)
-        # [K5b] batched weight/input gradients: three library GEMMs
-        hprev = _hprev_seq(h_seq, h0, ctx.reverse).reshape(B * T, H)
+        dg_flat = dgates.reshape(T * B, 4 * H)
+        # [K5b] weight/input gradients: three flat contiguous library GEMMs
+        hprev = h_prev_view(h_all, ctx.reverse).reshape(T * B, H)
         dw_h = (hprev.t() @ dg_flat).to(w_h.dtype)
-        dw_x = (x.reshape(B * T, X).t() @ dg_flat).to(w_x.dtype)
+        dw_x = (x.reshape(T * B, X).t() @ dg_flat).to(w_x.dtype)
         # sum with fp32 accumulation WITHOUT materializing an fp32 copy
         db = dg_flat.sum(dim=0, dtype=functional._acc_dtype(dg_flat.dtype)).to(w_x.dtype)
-        dx = (dg_flat @ w_x.t()).view(B, T, X)
+        dx = (dg_flat @ w_x.t()).view(T, B, X)
         return dx, dh0, dc0, dw_x, dw_h, db, None
 
 
@@ -101,7 +101,7 @@ def lstm_layer(
     b: torch.Tensor,
     reverse: bool = False,
 ):
-    """Run one LSTM layer (one direction) over (B,T,X) -> (B,T,H).
+    """Run one LSTM layer (one direction) over (T,B,X) -> (T,B,H).
 
     Returns (h_seq, hT, cT); cT is fp32.
     """

@@ -69,16 +69,17 @@ class Trainer:
 
     # ------------------------------------------------------------------
     def _bptt_windows(self, x, y):
-        """Split (x, y) along time into truncated-BPTT windows."""
-        T = x.shape[1]
+        """Split (x, y) along time (dim 0 -- time-major) into truncated-BPTT
+        windows."""
+        T = x.shape[0]
         k = self.cfg.train.bptt
         if k <= 0 or k >= T:
             return [(x, y)]
-        per_step_y = y.dim() >= 2 and y.shape[1] == T
+        per_step_y = y.dim() >= 2 and y.shape[0] == T
         wins = []
         for s in range(0, T, k):
             e = min(s + k, T)
-            wins.append((x[:, s:e], y[:, s:e] if per_step_y else y))
+            wins.append((x[s:e], y[s:e] if per_step_y else y))
         return wins
 
     def train_step(self, batch) -> dict:

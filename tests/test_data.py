@@ -41,16 +41,16 @@ def test_copy_task_learnable_structure():
     cfg.data.copy_delay = 2
     d = SyntheticSeqData(cfg, rank=0)
     x, y = d.batch(0)
-    assert x.shape == (4, 12, 16) and y.shape == (4, 12)
+    assert x.shape == (12, 4, 16) and y.shape == (12, 4)  # time-major
     # target at t equals input class at t-delay
     toks = x.argmax(-1)
-    assert torch.equal(y[:, 2:], toks[:, :-2])
-    assert (y[:, :2] == 0).all()
+    assert torch.equal(y[2:], toks[:-2])
+    assert (y[:2] == 0).all()
 
 
 def test_lm_task_shapes():
     d = SyntheticSeqData(_cfg("lm"))
     x, y = d.batch(0)
-    assert x.dtype == torch.long and x.shape == (4, 12)
+    assert x.dtype == torch.long and x.shape == (12, 4)  # time-major
     # next-token prediction: y is x shifted by one
-    assert torch.equal(x[:, 1:], y[:, :-1])
+    assert torch.equal(x[1:], y[:-1])

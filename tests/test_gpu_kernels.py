@@ -70,7 +70,7 @@ def seq_inputs(B, T, H, seed=0, dev="cuda"):
     def r(*s, scale=1.0):
         return (torch.randn(*s, generator=g) * scale).to(dev)
 
-    pre = r(B, T, 4 * H).bfloat16()
+    pre = r(T, B, 4 * H).bfloat16()  # time-major
     w_h = r(H, 4 * H, scale=1.0 / H ** 0.5).bfloat16()
     h0 = r(B, H, scale=0.5).bfloat16()
     c0 = r(B, H, scale=0.5)
@@ -83,11 +83,13 @@ def test_seq_fwd_vs_ref(ext, reverse, B, T, H):
     from lstm_tensorspark_amd.ops import functional
 
     pre, w_h, h0, c0 = seq_inputs(B, T, H)
-    h_seq, c_seq, acts, hT, cT = ext.lstm_seq_fwd(pre, w_h, h0, c0, reverse)
+    # both return h_all (T+1,B,H) with the h0 slot included
+    h_all, c_seq, acts, hT, cT = ext.lstm_seq_fwd(pre, w_h, h0, c0, reverse)
     h_r, c_r, a_r, hT_r, cT_r = functional.ref_seq_fwd(pre, w_h, h0, c0, reverse)
     # recurrence compounds bf16 rounding; tolerance grows mildly with T
     tol = 0.02 + 0.005 * T
-    assert (h_seq.float() - h_r.float()).abs().max().item() < tol
+    assert h_all.shape == (T + 1, B, H)
+    assert (h_all.float() - h_r.float()).abs().max().item() < tol
     assert (c_seq - c_r).abs().max().item() < 2 * tol
     assert (hT.float() - hT_r.float()).abs().max().item() < tol
     assert (cT - cT_r).abs().max().item() < 2 * tol
@@ -103,7 +105,7 @@ def test_seq_bwd_vs_ref(ext, reverse):
     # use the reference forward's saved tensors for BOTH backward paths
     h_r, c_r, a_r, hT_r, cT_r = functional.ref_seq_fwd(pre, w_h, h0, c0, reverse)
     g = torch.Generator(device="cpu").manual_seed(3)
-    d_hseq = (torch.randn(B, T, H, generator=g) * 0.1).to("cuda").bfloat16()
+    d_hseq = (torch.randn(T, B, H, generator=g) * 0.1).to("cuda").bfloat16()
     d_hT = (torch.randn(B, H, generator=g) * 0.1).to("cuda").bfloat16()
     d_cT = (torch.randn(B, H, generator=g) * 0.1).to("cuda")
 
@@ -136,13 +138,13 @@ def test_layer_autograd_gpu_vs_cpu_fp32():
 
     B, T, X, H = 32, 8, 128, 128
     g = torch.Generator(device="cpu").manual_seed(5)
-    x32 = torch.randn(B, T, X, generator=g) * 0.5
+    x32 = torch.randn(T, B, X, generator=g) * 0.5  # time-major
     h032 = torch.randn(B, H, generator=g) * 0.5
     c032 = torch.randn(B, H, generator=g) * 0.5
     wx32 = torch.randn(X, 4 * H, generator=g) / X ** 0.5
     wh32 = torch.randn(H, 4 * H, generator=g) / H ** 0.5
     b32 = torch.randn(4 * H, generator=g) * 0.1
-    gout = torch.randn(B, T, H, generator=g) * 0.1
+    gout = torch.randn(T, B, H, generator=g) * 0.1
 
     # CPU fp32 reference
     cpu = [t.clone().requires_grad_(True) for t in (x32, h032, c032, wx32, wh32, b32)]

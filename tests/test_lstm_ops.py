@@ -11,14 +11,15 @@ from lstm_tensorspark_amd.ops.lstm_layer import lstm_layer
 
 
 def autograd_ref(x, h0, c0, w_x, w_h, b, reverse):
-    B, T, X = x.shape
+    # time-major: x (T,B,X)
+    T, B, X = x.shape
     H = w_h.shape[0]
     pre = x @ w_x + b
     h, c = h0, c0
     hs = [None] * T
     ts = range(T - 1, -1, -1) if reverse else range(T)
     for t in ts:
-        raw = pre[:, t] + h @ w_h
+        raw = pre[t] + h @ w_h
         i = raw[:, :H].sigmoid()
         f = raw[:, H:2 * H].sigmoid()
         g = raw[:, 2 * H:3 * H].tanh()
@@ -26,7 +27,7 @@ def autograd_ref(x, h0, c0, w_x, w_h, b, reverse):
         c = f * c + i * g
         h = o * torch.tanh(c)
         hs[t] = h
-    return torch.stack(hs, 1), h, c
+    return torch.stack(hs, 0), h, c
 
 
 def make_inputs(B, T, X, H, dtype=torch.float64, seed=0):
@@ -35,7 +36,7 @@ def make_inputs(B, T, X, H, dtype=torch.float64, seed=0):
     def r(*s, scale=1.0):
         return (torch.randn(*s, generator=g, dtype=torch.float32) * scale).to(dtype)
 
-    return (r(B, T, X), r(B, H), r(B, H), r(X, 4 * H, scale=0.3),
+    return (r(T, B, X), r(B, H), r(B, H), r(X, 4 * H, scale=0.3),
             r(H, 4 * H, scale=0.3), r(4 * H, scale=0.1))
 
 
@@ -46,7 +47,7 @@ def test_manual_backward_matches_autograd_fp64(reverse, shape):
     tensors = [t.requires_grad_(True) for t in make_inputs(B, T, X, H)]
     x, h0, c0, w_x, w_h, b = tensors
     g = torch.Generator().manual_seed(1)
-    gouts = [torch.randn(B, T, H, generator=g, dtype=torch.float32).double(),
+    gouts = [torch.randn(T, B, H, generator=g, dtype=torch.float32).double(),
              torch.randn(B, H, generator=g, dtype=torch.float32).double(),
              torch.randn(B, H, generator=g, dtype=torch.float32).double()]
 
@@ -87,7 +88,7 @@ def test_parity_vs_torch_nn_lstm(bidi):
                       bidirectional=bidi, head="none", dropout=0.0)
     model = LSTMModel(cfg).double()
     ref = torch.nn.LSTM(X, H, num_layers=L, bidirectional=bidi,
-                        batch_first=True).double()
+                        batch_first=False).double()  # time-major, like us
     # copy weights: torch LSTM w_ih (4H, in) -> our W_x (in, 4H); same gate
     # order (i, f, g, o); our b = b_ih + b_hh
     with torch.no_grad():
@@ -102,7 +103,7 @@ def test_parity_vs_torch_nn_lstm(bidi):
                 model.weight(f"{name}/b").copy_(
                     getattr(ref, f"bias_ih_l{l}{sfx}")
                     + getattr(ref, f"bias_hh_l{l}{sfx}"))
-    x = torch.randn(B, T, X, dtype=torch.float64)
+    x = torch.randn(T, B, X, dtype=torch.float64)
     out_ref, (hn, cn) = ref(x)
     out, state = model.backbone(x)
     assert (out - out_ref).abs().max() < 1e-12
@@ -125,7 +126,7 @@ def test_grad_parity_vs_torch_nn_lstm():
                       bidirectional=True, head="none")
     model = LSTMModel(cfg).double()
     ref = torch.nn.LSTM(X, H, num_layers=L, bidirectional=True,
-                        batch_first=True).double()
+                        batch_first=False).double()  # time-major, like us
     with torch.no_grad():
         for l in range(L):
             for d in range(2):
@@ -135,8 +136,8 @@ def test_grad_parity_vs_torch_nn_lstm():
                 model.weight(f"{name}/W_h").copy_(getattr(ref, f"weight_hh_l{l}{sfx}").t())
                 model.weight(f"{name}/b").copy_(
                     getattr(ref, f"bias_ih_l{l}{sfx}") + getattr(ref, f"bias_hh_l{l}{sfx}"))
-    x = torch.randn(B, T, X, dtype=torch.float64)
-    gout = torch.randn(B, T, 2 * H, dtype=torch.float64)
+    x = torch.randn(T, B, X, dtype=torch.float64)
+    gout = torch.randn(T, B, 2 * H, dtype=torch.float64)
 
     out_ref, _ = ref(x)
     out_ref.backward(gout)

@@ -91,7 +91,7 @@ def main():
         args.iters), rec_flops, B=B, H=H)
 
     # sequence forward (transpose amortized over T steps)
-    pre = torch.randn(B, T, 4 * H, generator=g, device=dev).bfloat16()
+    pre = torch.randn(T, B, 4 * H, generator=g, device=dev).bfloat16()  # time-major
     h0 = torch.randn(B, H, generator=g, device=dev).bfloat16()
     c0 = torch.randn(B, H, generator=g, device=dev)
     # correctness: persistent vs per-step path on identical inputs
@@ -112,8 +112,8 @@ def main():
     ms = ms
 
     # sequence backward
-    h_seq, c_seq, acts, hT, cT = ext.lstm_seq_fwd(pre, w_h, h0, c0, False)
-    d_hseq = torch.randn(B, T, H, generator=g, device=dev).bfloat16()
+    h_all, c_seq, acts, hT, cT = ext.lstm_seq_fwd(pre, w_h, h0, c0, False)
+    d_hseq = torch.randn(T, B, H, generator=g, device=dev).bfloat16()
     ms = time_op(lambda: ext.lstm_seq_bwd(d_hseq, None, None, w_h, acts,
                                           c_seq, c0, False),
                  max(args.iters // 10, 3))
@@ -134,13 +134,13 @@ def main():
 
     # full layer autograd fwd+bwd
     from lstm_tensorspark_amd.ops.lstm_layer import lstm_layer
-    x = torch.randn(B, T, H, generator=g, device=dev).bfloat16()
+    x = torch.randn(T, B, H, generator=g, device=dev).bfloat16()
     wx_p = w_x.clone().requires_grad_(True)
     wh_p = w_h.clone().requires_grad_(True)
     b_p = torch.zeros(4 * H, device=dev).bfloat16().requires_grad_(True)
     h0z = torch.zeros(B, H, device=dev).bfloat16()
     c0z = torch.zeros(B, H, device=dev)
-    gout = torch.randn(B, T, H, generator=g, device=dev).bfloat16()
+    gout = torch.randn(T, B, H, generator=g, device=dev).bfloat16()
 
     def layer_fb():
         hs, _, _ = lstm_layer(x, h0z, c0z, wx_p, wh_p, b_p)
@@ -155,8 +155,8 @@ def main():
     dc = torch.zeros(B, H, device=dev)
     dg_out = torch.empty(B, 4 * H, device=dev).bfloat16()
     report("pw_bwd isolated (via seq T=1)", time_op(
-        lambda: ext.lstm_seq_bwd(None, d_hseq[:, 0].contiguous(), None, w_h,
-                                 acts[:, :1].contiguous(), c_seq[:, :1].contiguous(),
+        lambda: ext.lstm_seq_bwd(None, d_hseq[0].contiguous(), None, w_h,
+                                 acts[:1].contiguous(), c_seq[:1].contiguous(),
                                  c0, False), args.iters))
 
 
