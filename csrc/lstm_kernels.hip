@@ -742,8 +742,9 @@ __global__ __launch_bounds__(256) void k_adam(
 // ---------------------------------------------------------------------------
 extern "C" {
 
-// variant encoding: (v & 3) = STAGE {0 direct, 1 LDS-A, 2 all-glds},
-// bit2 = BM32 (else BM64), bit3 = XCD_SWZ
+// variant encoding: (v & 3) = STAGE {0 direct, 1 LDS-A, 2 all-glds, 3
+// counted-vmcnt}, bit2 = BM32 (else BM64), bit3 = XCD_SWZ, bit4 = BM128
+// (overrides bit2; halves per-output W_h panel L2 traffic vs BM64)
 void launch_lstm_rec_fwd_v(
     const void* h_prev, long hp_stride, const void* w_t,
     const void* pre, long pre_stride, const float* c_prev, long cp_stride,
@@ -751,7 +752,10 @@ void launch_lstm_rec_fwd_v(
     void* acts, long act_stride, int B, int H, int raw_debug, int variant,
     hipStream_t stream)
 {
-    const int BM = (variant & 4) ? 32 : 64;
+    // STAGE3 requires BM >= 64: its staging writes full 4 KiB glds slices,
+    // which would overrun the 2 KiB A image at BM=32 (corrupts B tile).
+    if ((variant & 7) == 7) variant &= ~1;  // BM32+STAGE3 -> BM32+STAGE2
+    const int BM = (variant & 16) ? 128 : ((variant & 4) ? 32 : 64);
     const int stage = variant & 3;
     dim3 grid((B + BM - 1) / BM, H / 64);
     dim3 block(256);
@@ -765,7 +769,7 @@ void launch_lstm_rec_fwd_v(
     (__hip_bfloat16*)acts, act_stride, B, H
 #define LAUNCH(BM_, RAW_, ST_, SW_)                                           \
     do {                                                                      \
-        if (ST_ == 2) {                                                       \
+        if (ST_ == 2 || ST_ == 3) {                                           \
             static bool attr_done_##BM_##RAW_##SW_ = false;                   \
             if (!attr_done_##BM_##RAW_##SW_ && smem > 64 * 1024) {            \
                 hipFuncSetAttribute(                                          \
@@ -778,7 +782,11 @@ void launch_lstm_rec_fwd_v(
                            block, smem, stream, ARGS);                        \
     } while (0)
 #define DISPATCH(RAW_)                                                        \
-    switch (variant & 15) {                                                   \
+    switch (variant & 31) {                                                   \
+        case 18: LAUNCH(128, RAW_, 2, false); break;                          \
+        case 19: LAUNCH(128, RAW_, 3, false); break;                          \
+        case 26: LAUNCH(128, RAW_, 2, true); break;                           \
+        case 27: LAUNCH(128, RAW_, 3, true); break;                           \
         case 0: LAUNCH(64, RAW_, 0, false); break;                            \
         case 1: LAUNCH(64, RAW_, 1, false); break;                            \
         case 2: LAUNCH(64, RAW_, 2, false); break;                            \

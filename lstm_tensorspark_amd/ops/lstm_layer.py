@@ -82,13 +82,21 @@ class LSTMLayerFn(torch.autograd.Function):
             w_h, acts, c_seq, c0, ctx.reverse, ctx.use_hip,
         )
         dg_flat = dgates.reshape(T * B, 4 * H)
-        # [K5b] weight/input gradients: three flat contiguous library GEMMs
-        hprev = h_prev_view(h_all, ctx.reverse).reshape(T * B, H)
-        dw_h = (hprev.t() @ dg_flat).to(w_h.dtype)
-        dw_x = (x.reshape(T * B, X).t() @ dg_flat).to(w_x.dtype)
-        # sum with fp32 accumulation WITHOUT materializing an fp32 copy
-        db = dg_flat.sum(dim=0, dtype=functional._acc_dtype(dg_flat.dtype)).to(w_x.dtype)
-        dx = (dg_flat @ w_x.t()).view(T, B, X)
+        # [K5b] weight/input gradients: flat contiguous library GEMMs, each
+        # gated on needs_input_grad (dx for the FIRST layer is a whole
+        # (T*B,4H)x(4H,X) GEMM that nothing consumes -- skip it).
+        ng = ctx.needs_input_grad
+        dw_h = dw_x = db = dx = None
+        if ng[4]:
+            hprev = h_prev_view(h_all, ctx.reverse).reshape(T * B, H)
+            dw_h = (hprev.t() @ dg_flat).to(w_h.dtype)
+        if ng[3]:
+            dw_x = (x.reshape(T * B, X).t() @ dg_flat).to(w_x.dtype)
+        if ng[5]:
+            # sum with fp32 accumulation WITHOUT materializing an fp32 copy
+            db = dg_flat.sum(dim=0, dtype=functional._acc_dtype(dg_flat.dtype)).to(w_x.dtype)
+        if ng[0]:
+            dx = (dg_flat @ w_x.t()).view(T, B, X)
         return dx, dh0, dc0, dw_x, dw_h, db, None
 
 
