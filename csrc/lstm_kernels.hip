@@ -27,6 +27,17 @@ typedef float f32x4 __attribute__((ext_vector_type(4)));
 
 #define DEVINL __device__ __forceinline__
 
+// Chunk swizzle for 64-B LDS rows (4x 16-B chunks) that is bank-conflict-
+// free for ds_read_b128's fixed lane groups ({0-3,12-15,20-27} etc.): the
+// physical slot of k-chunk lk at row r is lk ^ M64[(r>>2)&3] with
+// M64 = {0,3,2,1}.  Derivation: a lane group spans rows {c,c+4,c+8,c+12}
+// x two k-groups; this permutation gives the 16 lanes 16 distinct
+// (row%4, slot) bank positions.  XOR is involutive, so staging uses the
+// same formula on the SOURCE chunk index.  (The row&3 / row&7 XOR used for
+// 128-B rows does NOT transfer to 64-B rows: rows 12 apart collide and
+// k-groups 0/2 collapse -- measured 157M bank-conflict cycles.)
+DEVINL int swz64(int row) { return (0x1230 >> (((row >> 2) & 3) * 4)) & 3; }
+
 // Fast activations: v_exp + v_rcp (1-ulp) -- bf16 outputs cannot see the
 // difference vs libcalls, and __ocml_tanh_f32 is a branchy libcall.
 DEVINL float rcp_(float x) { return __builtin_amdgcn_rcpf(x); }
@@ -202,7 +213,7 @@ __global__ __launch_bounds__(256) void k_lstm_rec_fwd(
                 int grow = m_base + row;
                 grow = grow < B ? grow : B - 1;
                 const void* src = h_prev + (long)grow * hp_stride + k0
-                                  + ((ch ^ (row & 3)) * 8);
+                                  + ((ch ^ swz64(row)) * 8);
                 __builtin_amdgcn_global_load_lds(
                     (const __attribute__((address_space(1))) void*)src,
                     (__attribute__((address_space(3))) void*)(
@@ -213,7 +224,7 @@ __global__ __launch_bounds__(256) void k_lstm_rec_fwd(
                 const int ci = s2 * 256 + tid;
                 const int gc = ci >> 2, ch = ci & 3;
                 const long grow = (long)(gc >> 6) * H + n_col0 + (gc & 63);
-                const void* src = w_t + grow * H + k0 + ((ch ^ (gc & 3)) * 8);
+                const void* src = w_t + grow * H + k0 + ((ch ^ swz64(gc)) * 8);
                 __builtin_amdgcn_global_load_lds(
                     (const __attribute__((address_space(1))) void*)src,
                     (__attribute__((address_space(3))) void*)(
@@ -239,12 +250,12 @@ __global__ __launch_bounds__(256) void k_lstm_rec_fwd(
 #pragma unroll
             for (int mf = 0; mf < MF; ++mf) {
                 const int row = mf * 16 + lr;
-                a[mf] = *(const bf16x8*)(ab + row * 64 + ((lk ^ (row & 3)) * 16));
+                a[mf] = *(const bf16x8*)(ab + row * 64 + ((lk ^ swz64(row)) * 16));
             }
 #pragma unroll
             for (int g = 0; g < 4; ++g) {
                 const int gc = g * 64 + wave * 16 + lr;
-                bfr[g] = *(const bf16x8*)(bb + gc * 64 + ((lk ^ (gc & 3)) * 16));
+                bfr[g] = *(const bf16x8*)(bb + gc * 64 + ((lk ^ swz64(gc)) * 16));
             }
 #pragma unroll
             for (int g = 0; g < 4; ++g)
@@ -390,6 +401,317 @@ __global__ __launch_bounds__(256) void k_lstm_rec_fwd(
             ap[(long)2 * H] = __float2bfloat16(gg);
             ap[(long)3 * H] = __float2bfloat16(go);
         }
+    }
+}
+
+// ---------------------------------------------------------------------------
+// MT256 fused recurrent GEMM + cell update: the 256x256-tile structure of
+// the verified 8-phase GEMM template (guide §5 "256² 8-phase"), adapted to
+// the LSTM epilogue:
+//   - block = 512 threads = 8 waves laid out 2M x 4N.  Wave (wm, wn)
+//     computes rows [wm*128, +128) x h-columns [n_blk*64 + wn*16, +16) for
+//     ALL FOUR gates (gate-interleaved N): its 64 output gate-columns are
+//     {g*H + hcol} for g=0..3, so the cell update stays lane-local exactly
+//     as in k_lstm_rec_fwd.
+//   - BK=32 K-tiles (one mfma_16x16x32 depth), THREE 32 KiB LDS buffers
+//     (A 256x32 + B 256x32 bf16), glds-staged two tiles ahead with counted
+//     vmcnt and raw s_barrier (never vmcnt(0) in the main loop) -- the
+//     STAGE=3 discipline at 4 glds/tile/wave.
+//   - XOR-chunk source swizzle (ch ^ (row & 3) on 64-B rows) keeps the
+//     ds_read_b128 fragment reads bank-conflict-free.
+//   - 96 KiB LDS -> 1 block/CU, 2 waves/SIMD; acc = 4 gates x 8 m-frags
+//     (128 VGPRs).  grid (B/256, H/64) = 256 blocks at the headline shape:
+//     exactly one block per CU.
+// Per-output operand traffic is (256+256)/(256*256) = 2.5x less than the
+// BM64 variants; measured MFMA:LDS cycle ratio leaves LDS at ~37% load.
+// Requires B % 256 == 0 (launcher falls back to the BM64 path otherwise).
+// ---------------------------------------------------------------------------
+// ABL (ablation, µbench only): 0 normal; 1 = no staging/vmcnt (reads one
+// pre-staged buffer every tile -- WRONG results, isolates compute+barrier
+// cost); 2 = staging but no MFMA (isolates the glds/vmcnt pipeline cost).
+// STYLE 0 = 3-buffer BK=32 two-phase counted-vmcnt; STYLE 1 = 2-buffer
+// BK=64, one vmcnt(0)-draining __syncthreads per K-tile (guide: ties the
+// best register pipeline on the 256^2 tile at 1/4 the LDS instructions).
+template <bool RAW_DEBUG, bool XCD_SWZ, int ABL = 0, int STYLE = 0>
+__global__ __launch_bounds__(512) void k_lstm_rec_fwd_mt256(
+    const __hip_bfloat16* __restrict__ h_prev, long hp_stride,
+    const __hip_bfloat16* __restrict__ w_t,     // (4H, H) row-major = W_h^T
+    const __hip_bfloat16* __restrict__ pre, long pre_stride,   // (B, 4H)
+    const float* __restrict__ c_prev, long cp_stride,
+    __hip_bfloat16* __restrict__ h_out, long ho_stride,
+    float* __restrict__ c_out, long co_stride,
+    __hip_bfloat16* __restrict__ acts, long act_stride,
+    int B, int H)
+{
+    constexpr int BK = 32;
+    constexpr int ABYTES = 256 * BK * 2;   // 16 KiB
+    constexpr int TILE_BYTES = 2 * ABYTES; // A + B = 32 KiB
+    constexpr int TILE_GLDS = 4;           // 2 A + 2 B glds per tile per wave
+    extern __shared__ __attribute__((aligned(16))) char smem[];
+
+    const int tid = threadIdx.x;
+    const int wave = tid >> 6;
+    const int lane = tid & 63;
+    const int lr = lane & 15;   // fragment row (A) / col (B, C/D)
+    const int lk = lane >> 4;   // k-group 0..3
+    const int wm = wave & 1;    // M half: rows [wm*128, +128)
+    const int wn = wave >> 1;   // N group: h-cols [wn*16, +16)
+
+    int m_blk = blockIdx.x, n_blk = blockIdx.y;
+    if (XCD_SWZ) {
+        const int nm = gridDim.x, nwg = gridDim.x * gridDim.y;
+        const int bid = blockIdx.x + blockIdx.y * nm;
+        const int q = nwg >> 3, r = nwg & 7, xcd = bid & 7, loc = bid >> 3;
+        const int swz = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q)
+                        + loc;
+        m_blk = swz % nm;
+        n_blk = swz / nm;
+    }
+    const int m_base = m_blk * 256;
+    const long n_col0 = (long)n_blk * 64;
+
+    // stage K-tile k0 into buffer buf: A rows 256 x BK, B gate-cols 256 x BK,
+    // 64-B rows, XOR-chunk source swizzle so fragment reads are conflict-free
+    auto stage_a = [&](int buf, int k0) {
+        char* const ab = smem + buf * TILE_BYTES;
+#pragma unroll
+        for (int s2 = 0; s2 < 2; ++s2) {
+            const int ci = s2 * 512 + tid;
+            const int row = ci >> 2, ch = ci & 3;
+            const void* src = h_prev + (long)(m_base + row) * hp_stride + k0
+                              + ((ch ^ swz64(row)) * 8);
+            __builtin_amdgcn_global_load_lds(
+                (const __attribute__((address_space(1))) void*)src,
+                (__attribute__((address_space(3))) void*)(
+                    ab + s2 * 8192 + wave * 1024), 16, 0, 0);
+        }
+    };
+    auto stage_b = [&](int buf, int k0) {
+        char* const bb = smem + buf * TILE_BYTES + ABYTES;
+#pragma unroll
+        for (int s2 = 0; s2 < 2; ++s2) {
+            const int ci = s2 * 512 + tid;
+            const int gc = ci >> 2, ch = ci & 3;
+            const long grow = (long)(gc >> 6) * H + n_col0 + (gc & 63);
+            const void* src = w_t + grow * H + k0 + ((ch ^ swz64(gc)) * 8);
+            __builtin_amdgcn_global_load_lds(
+                (const __attribute__((address_space(1))) void*)src,
+                (__attribute__((address_space(3))) void*)(
+                    bb + s2 * 8192 + wave * 1024), 16, 0, 0);
+        }
+    };
+    auto stage = [&](int buf, int k0) { stage_a(buf, k0); stage_b(buf, k0); };
+
+    f32x4 acc[4][8];
+#pragma unroll
+    for (int g = 0; g < 4; ++g)
+#pragma unroll
+        for (int mf = 0; mf < 8; ++mf) acc[g][mf] = f32x4{0, 0, 0, 0};
+
+    if (STYLE == 1) {
+        // 2-buffer BK=64: images are [256][128] bytes with the proven
+        // ch^(row&7) swizzle; ONE __syncthreads per K-tile carries the
+        // vmcnt(0) glds drain (STAGE2 discipline at MT256 scale).
+        constexpr int BK64 = 64;
+        constexpr int A64 = 256 * BK64 * 2;        // 32 KiB
+        constexpr int T64 = 2 * A64;               // A+B = 64 KiB
+        auto stage64 = [&](int buf, int k0) {
+            char* const ab = smem + buf * T64;
+            char* const bb = ab + A64;
+#pragma unroll
+            for (int s2 = 0; s2 < 4; ++s2) {
+                const int ci = s2 * 512 + tid;
+                const int row = ci >> 3, ch = ci & 7;
+                const void* src = h_prev + (long)(m_base + row) * hp_stride
+                                  + k0 + ((ch ^ (row & 7)) * 8);
+                __builtin_amdgcn_global_load_lds(
+                    (const __attribute__((address_space(1))) void*)src,
+                    (__attribute__((address_space(3))) void*)(
+                        ab + s2 * 8192 + wave * 1024), 16, 0, 0);
+            }
+#pragma unroll
+            for (int s2 = 0; s2 < 4; ++s2) {
+                const int ci = s2 * 512 + tid;
+                const int gc = ci >> 3, ch = ci & 7;
+                const long grow = (long)(gc >> 6) * H + n_col0 + (gc & 63);
+                const void* src = w_t + grow * H + k0 + ((ch ^ (gc & 7)) * 8);
+                __builtin_amdgcn_global_load_lds(
+                    (const __attribute__((address_space(1))) void*)src,
+                    (__attribute__((address_space(3))) void*)(
+                        bb + s2 * 8192 + wave * 1024), 16, 0, 0);
+            }
+        };
+        const int nst = H / BK64;
+        stage64(0, 0);
+        __syncthreads();   // drains the glds (vmcnt0) + releases reads
+        for (int j = 0; j < nst; ++j) {
+            const int buf = j & 1;
+            if (j + 1 < nst) stage64(buf ^ 1, (j + 1) * BK64);
+            const char* ab = smem + buf * T64;
+            const char* bb = ab + A64;
+#pragma unroll
+            for (int kc = 0; kc < 2; ++kc) {
+                bf16x8 bfr[4], a[8];
+#pragma unroll
+                for (int g = 0; g < 4; ++g) {
+                    const int gc = g * 64 + wn * 16 + lr;
+                    bfr[g] = *(const bf16x8*)(
+                        bb + gc * 128 + (((kc * 4 + lk) ^ (gc & 7)) * 16));
+                }
+#pragma unroll
+                for (int mi = 0; mi < 8; ++mi) {
+                    const int row = wm * 128 + mi * 16 + lr;
+                    a[mi] = *(const bf16x8*)(
+                        ab + row * 128 + (((kc * 4 + lk) ^ (row & 7)) * 16));
+                }
+                __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+                for (int g = 0; g < 4; ++g)
+#pragma unroll
+                    for (int mi = 0; mi < 8; ++mi)
+                        acc[g][mi] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                            bfr[g], a[mi], acc[g][mi], 0, 0, 0);
+                __builtin_amdgcn_s_setprio(0);
+            }
+            __syncthreads();  // next tile's glds drained + buf reuse safe
+        }
+        goto epilogue;
+    }
+    {
+    const int nsteps = H / BK;
+    stage(0, 0);
+    if (ABL != 1 && nsteps > 1) stage(1, BK);
+    // Phase discipline (guide §5 8-phase template, T3+T4): per phase
+    //   {ds_read subtile; 2 glds; s_barrier; lgkmcnt(0); setprio(1);
+    //    16 MFMA; setprio(0); s_barrier}
+    // with vmcnt counted ONCE per K-tile (never 0 in the main loop).
+    // CRITICAL ORDER: vmcnt is PER-WAVE, but fragments are staged by ALL
+    // waves' glds -- so every tile-landing wait must be followed by an
+    // s_barrier BEFORE any wave reads that tile's LDS.  The waits therefore
+    // sit at the TAIL of the previous tile (just before its trailing
+    // barrier), counting only the tile staged in that same iteration.
+    if (nsteps > 1 && ABL != 1)
+        asm volatile("s_waitcnt vmcnt(%0)" :: "i"(TILE_GLDS) : "memory");
+    else
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    for (int j = 0; j < nsteps; ++j) {
+        const int buf = (ABL == 1) ? 0 : (j % 3);
+        const char* ab = smem + buf * TILE_BYTES;
+        const char* bb = ab + ABYTES;
+        const bool pf = (ABL != 1) && (j + 2 < nsteps);
+        const int pbuf = (j + 2) % 3, pk0 = (j + 2) * BK;
+        // ---- phase A: B frags + A frags 0-3; stage tile j+2's A half
+        bf16x8 bfr[4], a[4];
+#pragma unroll
+        for (int g = 0; g < 4; ++g) {
+            const int gc = g * 64 + wn * 16 + lr;
+            bfr[g] = *(const bf16x8*)(bb + gc * 64 + ((lk ^ swz64(gc)) * 16));
+        }
+#pragma unroll
+        for (int mi = 0; mi < 4; ++mi) {
+            const int row = wm * 128 + mi * 16 + lr;
+            a[mi] = *(const bf16x8*)(ab + row * 64 + ((lk ^ swz64(row)) * 16));
+        }
+        if (pf) stage_a(pbuf, pk0);
+        __builtin_amdgcn_s_barrier();
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        __builtin_amdgcn_s_setprio(1);
+        // OPERAND ORDER: A = weight fragment, B = hidden fragment, so the
+        // D layout is (gate-col, batch): lane (lk,lr) holds 4 CONSECUTIVE
+        // h-columns (lk*4+j) of one batch row (lr) -> the epilogue's
+        // pre/acts/h/c accesses are bf16x4/f32x4, not scalar bf16 (the
+        // scalar-epilogue version measured 50 of its 71 us in the epilogue).
+        if (ABL != 2)
+#pragma unroll
+            for (int g = 0; g < 4; ++g)
+#pragma unroll
+                for (int mi = 0; mi < 4; ++mi)
+                    acc[g][mi] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        bfr[g], a[mi], acc[g][mi], 0, 0, 0);
+        __builtin_amdgcn_s_setprio(0);
+        __builtin_amdgcn_s_barrier();
+        // ---- phase B: A frags 4-7; stage tile j+2's B half
+#pragma unroll
+        for (int mi = 0; mi < 4; ++mi) {
+            const int row = wm * 128 + (4 + mi) * 16 + lr;
+            a[mi] = *(const bf16x8*)(ab + row * 64 + ((lk ^ swz64(row)) * 16));
+        }
+        if (pf) stage_b(pbuf, pk0);
+        __builtin_amdgcn_s_barrier();
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        __builtin_amdgcn_s_setprio(1);
+        if (ABL != 2)
+#pragma unroll
+            for (int g = 0; g < 4; ++g)
+#pragma unroll
+                for (int mi = 0; mi < 4; ++mi)
+                    acc[g][4 + mi] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        bfr[g], a[mi], acc[g][4 + mi], 0, 0, 0);
+        __builtin_amdgcn_s_setprio(0);
+        // tail wait: tile j+1 must be landed before the next iteration's
+        // ds_reads; only tile j+2's glds (issued THIS iteration) may fly.
+        if (ABL != 1 && j + 1 < nsteps) {
+            if (pf)
+                asm volatile("s_waitcnt vmcnt(%0)" :: "i"(TILE_GLDS) : "memory");
+            else
+                asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        }
+        __builtin_amdgcn_s_barrier();
+    }
+    }
+epilogue:
+
+    // Fused epilogue, fully vectorized: lane owns h-columns
+    // [hcol0, hcol0+4) of batch rows (wm*128 + nf*16 + lr) -- every global
+    // access is bf16x4 (8 B) or f32x4 (16 B).
+    const int hcol0 = (int)n_col0 + wn * 16 + lk * 4;
+#pragma unroll
+    for (int nf = 0; nf < 8; ++nf) {
+        const long row = m_base + wm * 128 + nf * 16 + lr;
+        f32x4 raw[4];
+#pragma unroll
+        for (int g = 0; g < 4; ++g) {
+            const bf16x4 pv = *(const bf16x4*)(
+                pre + row * pre_stride + (long)g * H + hcol0);
+#pragma unroll
+            for (int j = 0; j < 4; ++j)
+                raw[g][j] = acc[g][nf][j] + (float)pv[j];
+        }
+        __hip_bfloat16* ap = acts + row * act_stride + hcol0;
+        if (RAW_DEBUG) {
+#pragma unroll
+            for (int g = 0; g < 4; ++g) {
+                bf16x4 o;
+#pragma unroll
+                for (int j = 0; j < 4; ++j) o[j] = (__bf16)raw[g][j];
+                *(bf16x4*)(ap + (long)g * H) = o;
+            }
+            continue;
+        }
+        const f32x4 cp = *(const f32x4*)(c_prev + row * cp_stride + hcol0);
+        bf16x4 oi, of, og, oo, oh;
+        f32x4 oc;
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+            const float gi = sigmoidf_(raw[0][j]);
+            const float gf = sigmoidf_(raw[1][j]);
+            const float gg = tanhf_(raw[2][j]);
+            const float go = sigmoidf_(raw[3][j]);
+            const float c  = gf * cp[j] + gi * gg;
+            oc[j] = c;
+            oh[j] = (__bf16)(go * tanhf_(c));
+            oi[j] = (__bf16)gi;
+            of[j] = (__bf16)gf;
+            og[j] = (__bf16)gg;
+            oo[j] = (__bf16)go;
+        }
+        *(bf16x4*)(h_out + row * ho_stride + hcol0) = oh;
+        *(f32x4*)(c_out + row * co_stride + hcol0) = oc;
+        *(bf16x4*)(ap)               = oi;
+        *(bf16x4*)(ap + (long)H)     = of;
+        *(bf16x4*)(ap + (long)2 * H) = og;
+        *(bf16x4*)(ap + (long)3 * H) = oo;
     }
 }
 
@@ -752,6 +1074,78 @@ void launch_lstm_rec_fwd_v(
     void* acts, long act_stride, int B, int H, int raw_debug, int variant,
     hipStream_t stream)
 {
+    // MT256 path (bit5): 256x256 tile, 512 threads, requires B % 256 == 0.
+    if ((variant & 32) && (B % 256 == 0)) {
+        const size_t smem = (variant & 1)
+            ? (size_t)(2 * 2 * 256 * 64 * 2)                 // 128 KiB (STYLE1)
+            : 3 * 2 * (size_t)(256 * 32 * 2);                // 96 KiB
+        static bool attr_done[4] = {false, false, false, false};
+        const int ai = (variant & 8 ? 1 : 0) | (raw_debug ? 2 : 0);
+        dim3 grid(B / 256, H / 64), block(512);
+#define MTLAUNCH4(RAW_, SW_, ABL_, ST_)                                       \
+    do {                                                                      \
+        hipError_t e_ = hipFuncSetAttribute(                                  \
+            (const void*)(k_lstm_rec_fwd_mt256<RAW_, SW_, ABL_, ST_>),        \
+            hipFuncAttributeMaxDynamicSharedMemorySize, (int)smem);           \
+        (void)e_;                                                             \
+        hipLaunchKernelGGL((k_lstm_rec_fwd_mt256<RAW_, SW_, ABL_, ST_>),      \
+                           grid, block, smem, stream,                         \
+            (const __hip_bfloat16*)h_prev, hp_stride,                         \
+            (const __hip_bfloat16*)w_t, (const __hip_bfloat16*)pre,           \
+            pre_stride, c_prev, cp_stride, (__hip_bfloat16*)h_out, ho_stride, \
+            c_out, co_stride, (__hip_bfloat16*)acts, act_stride, B, H);       \
+    } while (0)
+#define MTLAUNCH3(RAW_, SW_, ABL_)                                            \
+    do {                                                                      \
+        hipError_t e_ = hipFuncSetAttribute(                                  \
+            (const void*)(k_lstm_rec_fwd_mt256<RAW_, SW_, ABL_>),             \
+            hipFuncAttributeMaxDynamicSharedMemorySize, (int)smem);           \
+        (void)e_;                                                             \
+        hipLaunchKernelGGL((k_lstm_rec_fwd_mt256<RAW_, SW_, ABL_>), grid,     \
+                           block, smem, stream,                               \
+            (const __hip_bfloat16*)h_prev, hp_stride,                         \
+            (const __hip_bfloat16*)w_t, (const __hip_bfloat16*)pre,           \
+            pre_stride, c_prev, cp_stride, (__hip_bfloat16*)h_out, ho_stride, \
+            c_out, co_stride, (__hip_bfloat16*)acts, act_stride, B, H);       \
+    } while (0)
+#define MTLAUNCH(RAW_, SW_)                                                   \
+    do {                                                                      \
+        if (!attr_done[ai]) {                                                 \
+            hipError_t e2_ = hipFuncSetAttribute(                             \
+                (const void*)(k_lstm_rec_fwd_mt256<RAW_, SW_>),               \
+                hipFuncAttributeMaxDynamicSharedMemorySize, (int)smem);       \
+            (void)e2_;                                                        \
+            attr_done[ai] = true;                                             \
+        }                                                                     \
+        hipLaunchKernelGGL((k_lstm_rec_fwd_mt256<RAW_, SW_>), grid, block,    \
+                           smem, stream,                                      \
+            (const __hip_bfloat16*)h_prev, hp_stride,                         \
+            (const __hip_bfloat16*)w_t, (const __hip_bfloat16*)pre,           \
+            pre_stride, c_prev, cp_stride, (__hip_bfloat16*)h_out, ho_stride, \
+            c_out, co_stride, (__hip_bfloat16*)acts, act_stride, B, H);       \
+    } while (0)
+        if (variant & 64) { MTLAUNCH3(true, false, 1); return; }    // no-stage
+        if (variant & 128) { MTLAUNCH3(true, false, 2); return; }   // no-MFMA
+        if (variant & 1) {  // STYLE1: 2-buffer BK=64 (needs 128 KiB LDS)
+            if (raw_debug) {
+                if (variant & 8) MTLAUNCH4(true, true, 0, 1);
+                else MTLAUNCH4(true, false, 0, 1);
+            } else {
+                if (variant & 8) MTLAUNCH4(false, true, 0, 1);
+                else MTLAUNCH4(false, false, 0, 1);
+            }
+            return;
+        }
+        if (raw_debug) {
+            if (variant & 8) MTLAUNCH(true, true); else MTLAUNCH(true, false);
+        } else {
+            if (variant & 8) MTLAUNCH(false, true); else MTLAUNCH(false, false);
+        }
+#undef MTLAUNCH
+#undef MTLAUNCH3
+#undef MTLAUNCH4
+        return;
+    }
     // STAGE3 requires BM >= 64: its staging writes full 4 KiB glds slices,
     // which would overrun the 2 KiB A image at BM=32 (corrupts B tile).
     if ((variant & 7) == 7) variant &= ~1;  // BM32+STAGE3 -> BM32+STAGE2
@@ -854,7 +1248,8 @@ void launch_lstm_rec_fwd(
     hipStream_t stream)
 {
     int v = g_fwd_variant;
-    if (v < 0) v = (B >= 2048) ? 10 : 14;
+    if (v < 0) v = (B % 256 == 0 && B >= 2048) ? 40   // MT256 + XCD swizzle
+                 : (B >= 2048) ? 10 : 14;
     launch_lstm_rec_fwd_v(h_prev, hp_stride, w_t, pre, pre_stride, c_prev,
                           cp_stride, h_out, ho_stride, c_out, co_stride, acts,
                           act_stride, B, H, raw_debug, v, stream);
