@@ -127,6 +127,12 @@ def main():
     dg1 = torch.randn(B, 4 * H, generator=g, device=dev).bfloat16()
     report("mm dh=dgates@WhT (per step)", time_op(lambda: dg1 @ w_h.t(), args.iters),
            rec_flops)
+    # the fwd recurrent shape on hipBLASLt (split-path candidate):
+    raw_buf = torch.empty(B, 4 * H, device=dev).bfloat16()
+    report("mm fwd h@Wh (per step)", time_op(
+        lambda: torch.mm(h_prev, w_h, out=raw_buf), args.iters), rec_flops)
+    report("addmm fwd pre+h@Wh (per step)", time_op(
+        lambda: torch.addmm(pre1, h_prev, w_h, out=raw_buf), args.iters), rec_flops)
     x_flat = torch.randn(B * T, H, generator=g, device=dev).bfloat16()
     w_x = (torch.randn(H, 4 * H, generator=g, device=dev) / H ** 0.5).bfloat16()
     report("pre-GEMM (BT,X)@(X,4H)", time_op(lambda: x_flat @ w_x, 10),
