@@ -37,6 +37,50 @@ import torch
 from . import extension, functional
 from .functional import h_prev_view, h_seq_view
 
+# ---------------------------------------------------------------------------
+# Side-stream weight-gradient GEMMs.
+#
+# The recurrent backward is a serial per-timestep chain (pw_bwd -> dh GEMM)
+# that underutilizes the chip; the big dW_h/dW_x/db GEMMs have no consumer
+# until the optimizer (or the DDP all-reduce).  Launching them on a side
+# HIP stream lets them overlap the NEXT layer's sequential backward loop,
+# taking ~all of the dW time off the critical path.
+#
+# Correctness protocol: the side stream waits on an event recorded after
+# dgates is ready; every consumer of .grad (optimizer step, grad-clip,
+# bucketed all-reduce) must first call wait_side_stream() -- Trainer and
+# GradBucketer do.  Producer tensors read by the side stream are
+# record_stream()-ed so the caching allocator cannot reuse them early.
+# ---------------------------------------------------------------------------
+_side_stream = None
+_side_pending = False
+# Side-stream dW is only safe when autograd STORES the returned grads
+# (fresh .grad) rather than accumulating in place on the main stream --
+# i.e. no pre-assigned bucket views (DDP GradBucketer) and no multi-window
+# truncated-BPTT accumulation.  Trainer enables it when those hold.
+_side_enabled = False
+
+
+def enable_side_dw(on: bool):
+    global _side_enabled
+    _side_enabled = bool(on)
+
+
+def _get_side_stream():
+    global _side_stream
+    if _side_stream is None:
+        _side_stream = torch.cuda.Stream()
+    return _side_stream
+
+
+def wait_side_stream():
+    """Make the CURRENT stream wait for all side-stream dW GEMMs.  Cheap
+    no-op when nothing is pending or CUDA is unavailable."""
+    global _side_pending
+    if _side_pending and _side_stream is not None:
+        torch.cuda.current_stream().wait_stream(_side_stream)
+        _side_pending = False
+
 
 def _seq_fwd(pre, w_h, h0, c0, reverse, use_hip):
     if use_hip:
@@ -87,14 +131,35 @@ class LSTMLayerFn(torch.autograd.Function):
         # (T*B,4H)x(4H,X) GEMM that nothing consumes -- skip it).
         ng = ctx.needs_input_grad
         dw_h = dw_x = db = dx = None
-        if ng[4]:
-            hprev = h_prev_view(h_all, ctx.reverse).reshape(T * B, H)
-            dw_h = (hprev.t() @ dg_flat).to(w_h.dtype)
-        if ng[3]:
-            dw_x = (x.reshape(T * B, X).t() @ dg_flat).to(w_x.dtype)
-        if ng[5]:
-            # sum with fp32 accumulation WITHOUT materializing an fp32 copy
-            db = dg_flat.sum(dim=0, dtype=functional._acc_dtype(dg_flat.dtype)).to(w_x.dtype)
+        x_flat = x.reshape(T * B, X)
+        use_side = _side_enabled and ctx.use_hip and dgates.is_cuda
+        if use_side:
+            global _side_pending
+            side = _get_side_stream()
+            side.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(side):
+                if ng[4]:
+                    hprev = h_prev_view(h_all, ctx.reverse).reshape(T * B, H)
+                    dw_h = (hprev.t() @ dg_flat).to(w_h.dtype)
+                    h_all.record_stream(side)
+                if ng[3]:
+                    dw_x = (x_flat.t() @ dg_flat).to(w_x.dtype)
+                    x.record_stream(side)
+                if ng[5]:
+                    db = dg_flat.sum(
+                        dim=0, dtype=torch.float32).to(w_x.dtype)
+                dgates.record_stream(side)
+            _side_pending = True
+        else:
+            if ng[4]:
+                hprev = h_prev_view(h_all, ctx.reverse).reshape(T * B, H)
+                dw_h = (hprev.t() @ dg_flat).to(w_h.dtype)
+            if ng[3]:
+                dw_x = (x_flat.t() @ dg_flat).to(w_x.dtype)
+            if ng[5]:
+                # fp32 accumulation WITHOUT materializing an fp32 copy
+                db = dg_flat.sum(dim=0, dtype=functional._acc_dtype(
+                    dg_flat.dtype)).to(w_x.dtype)
         if ng[0]:
             dx = (dg_flat @ w_x.t()).view(T, B, X)
         return dx, dh0, dc0, dw_x, dw_h, db, None

@@ -28,6 +28,7 @@ from .checkpoint import load_checkpoint, save_checkpoint
 from .config import Config
 from .data import SyntheticSeqData
 from .models.lstm import LSTMModel
+from .ops.lstm_layer import enable_side_dw, wait_side_stream
 from .ops.optim import build_optimizer
 from .parallel.ddp import GradBucketer, broadcast_parameters
 from .parallel.launch import DistContext
@@ -64,6 +65,12 @@ class Trainer:
         self.data = SyntheticSeqData(cfg, rank=self.ctx.rank,
                                      world_size=self.ctx.world_size,
                                      device=self.device, dtype=self.dtype)
+        # Side-stream dW overlap (ops/lstm_layer.py): only when autograd
+        # stores (not accumulates) weight grads -- no DDP bucket views, no
+        # multi-window BPTT accumulation.
+        single_window = cfg.train.bptt <= 0 or cfg.train.bptt >= cfg.train.seq_len
+        enable_side_dw(self.bucketer is None and single_window
+                       and self.device.type == "cuda")
         self.step = 0
         self.history: list[dict] = []
 
@@ -110,6 +117,8 @@ class Trainer:
             total_loss += float(loss.detach())
         if self.bucketer is not None:
             self.bucketer.finalize()
+        # grads may still be in flight on the dW side stream
+        wait_side_stream()
 
         norm = None
         if cfg.train.grad_clip > 0:

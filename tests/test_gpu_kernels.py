@@ -280,3 +280,37 @@ def test_fused_xent_grad_scale(ext):
     l2 = logits.clone().requires_grad_(True)
     FusedCrossEntropyFn.apply(l2, targets).backward()
     assert (l1.grad.float() - 3.0 * l2.grad.float()).abs().max().item() < 1e-2
+
+
+# ------------------------------------------------- side-stream dW overlap
+def test_side_stream_dw_matches_sync_path():
+    """dW/db produced on the side stream (after wait_side_stream) must be
+    identical to the synchronous path."""
+    from lstm_tensorspark_amd.ops.lstm_layer import (
+        enable_side_dw, lstm_layer, wait_side_stream)
+
+    T, B, X, H = 8, 256, 128, 128
+    g = torch.Generator(device="cpu").manual_seed(7)
+    x32 = torch.randn(T, B, X, generator=g) * 0.5
+    wx32 = torch.randn(X, 4 * H, generator=g) / X ** 0.5
+    wh32 = torch.randn(H, 4 * H, generator=g) / H ** 0.5
+    b32 = torch.randn(4 * H, generator=g) * 0.1
+    gout = (torch.randn(T, B, H, generator=g) * 0.1).to("cuda").bfloat16()
+    h0 = torch.zeros(B, H, device="cuda").bfloat16()
+    c0 = torch.zeros(B, H, device="cuda")
+
+    grads = {}
+    for side in (False, True):
+        enable_side_dw(side)
+        xs = x32.to("cuda").bfloat16().requires_grad_(True)
+        wx = wx32.to("cuda").bfloat16().requires_grad_(True)
+        wh = wh32.to("cuda").bfloat16().requires_grad_(True)
+        bb = b32.to("cuda").bfloat16().requires_grad_(True)
+        hs, _, _ = lstm_layer(xs, h0, c0, wx, wh, bb)
+        hs.backward(gout)
+        wait_side_stream()
+        torch.cuda.synchronize()
+        grads[side] = (wx.grad.clone(), wh.grad.clone(), bb.grad.clone())
+    enable_side_dw(False)
+    for a, b in zip(grads[False], grads[True]):
+        assert torch.equal(a, b)
