@@ -905,6 +905,7 @@ __global__ __launch_bounds__(256) void k_lstm_seq_fwd_persistent(
 // the best fused hand kernel -- the split's extra raw round-trip costs less
 // than the GEMM gap).  Fully vectorized: 4 h-columns per lane.
 // ---------------------------------------------------------------------------
+template <int W>  // columns per lane: 8 (dwordx4 stores) when H%8==0
 __global__ __launch_bounds__(256) void k_lstm_pw_fwd(
     const __hip_bfloat16* __restrict__ raw_rec, long rr_stride,  // (B,4H)
     const __hip_bfloat16* __restrict__ pre, long pre_stride,     // (B,4H)
@@ -914,25 +915,27 @@ __global__ __launch_bounds__(256) void k_lstm_pw_fwd(
     __hip_bfloat16* __restrict__ acts, long act_stride,
     int B, int H)
 {
+    typedef __bf16 bfv __attribute__((ext_vector_type(W)));
+    typedef float fv __attribute__((ext_vector_type(W)));
     const int row = blockIdx.y;
     if (row >= B) return;
-    const int col0 = (blockIdx.x * blockDim.x + threadIdx.x) * 4;
+    const int col0 = (blockIdx.x * blockDim.x + threadIdx.x) * W;
     if (col0 >= H) return;
     const __hip_bfloat16* rp = raw_rec + (long)row * rr_stride + col0;
     const __hip_bfloat16* pp = pre + (long)row * pre_stride + col0;
-    f32x4 raw[4];
+    fv raw[4];
 #pragma unroll
     for (int g = 0; g < 4; ++g) {
-        const bf16x4 rv = *(const bf16x4*)(rp + (long)g * H);
-        const bf16x4 pv = *(const bf16x4*)(pp + (long)g * H);
+        const bfv rv = *(const bfv*)(rp + (long)g * H);
+        const bfv pv = *(const bfv*)(pp + (long)g * H);
 #pragma unroll
-        for (int j = 0; j < 4; ++j) raw[g][j] = (float)rv[j] + (float)pv[j];
+        for (int j = 0; j < W; ++j) raw[g][j] = (float)rv[j] + (float)pv[j];
     }
-    const f32x4 cp = *(const f32x4*)(c_prev + (long)row * cp_stride + col0);
-    bf16x4 oi, of, og, oo, oh;
-    f32x4 oc;
+    const fv cp = *(const fv*)(c_prev + (long)row * cp_stride + col0);
+    bfv oi, of, og, oo, oh;
+    fv oc;
 #pragma unroll
-    for (int j = 0; j < 4; ++j) {
+    for (int j = 0; j < W; ++j) {
         const float gi = sigmoidf_(raw[0][j]);
         const float gf = sigmoidf_(raw[1][j]);
         const float gg = tanhf_(raw[2][j]);
@@ -945,13 +948,13 @@ __global__ __launch_bounds__(256) void k_lstm_pw_fwd(
         og[j] = (__bf16)gg;
         oo[j] = (__bf16)go;
     }
-    *(bf16x4*)(h_out + (long)row * ho_stride + col0) = oh;
-    *(f32x4*)(c_out + (long)row * co_stride + col0) = oc;
+    *(bfv*)(h_out + (long)row * ho_stride + col0) = oh;
+    *(fv*)(c_out + (long)row * co_stride + col0) = oc;
     __hip_bfloat16* ap = acts + (long)row * act_stride + col0;
-    *(bf16x4*)(ap)               = oi;
-    *(bf16x4*)(ap + (long)H)     = of;
-    *(bf16x4*)(ap + (long)2 * H) = og;
-    *(bf16x4*)(ap + (long)3 * H) = oo;
+    *(bfv*)(ap)               = oi;
+    *(bfv*)(ap + (long)H)     = of;
+    *(bfv*)(ap + (long)2 * H) = og;
+    *(bfv*)(ap + (long)3 * H) = oo;
 }
 
 extern "C" void launch_lstm_pw_fwd(
@@ -961,12 +964,21 @@ extern "C" void launch_lstm_pw_fwd(
     int B, int H, hipStream_t stream)
 {
     dim3 block(256);
-    dim3 grid((H / 4 + 255) / 256, B);
-    hipLaunchKernelGGL(k_lstm_pw_fwd, grid, block, 0, stream,
-        (const __hip_bfloat16*)raw_rec, rr_stride,
-        (const __hip_bfloat16*)pre, pre_stride, c_prev, cp_stride,
-        (__hip_bfloat16*)h_out, ho_stride, c_out, co_stride,
-        (__hip_bfloat16*)acts, act_stride, B, H);
+    if (H % 8 == 0) {
+        dim3 grid((H / 8 + 255) / 256, B);
+        hipLaunchKernelGGL(k_lstm_pw_fwd<8>, grid, block, 0, stream,
+            (const __hip_bfloat16*)raw_rec, rr_stride,
+            (const __hip_bfloat16*)pre, pre_stride, c_prev, cp_stride,
+            (__hip_bfloat16*)h_out, ho_stride, c_out, co_stride,
+            (__hip_bfloat16*)acts, act_stride, B, H);
+    } else {
+        dim3 grid((H / 4 + 255) / 256, B);
+        hipLaunchKernelGGL(k_lstm_pw_fwd<4>, grid, block, 0, stream,
+            (const __hip_bfloat16*)raw_rec, rr_stride,
+            (const __hip_bfloat16*)pre, pre_stride, c_prev, cp_stride,
+            (__hip_bfloat16*)h_out, ho_stride, c_out, co_stride,
+            (__hip_bfloat16*)acts, act_stride, B, H);
+    }
 }
 
 // ---------------------------------------------------------------------------
@@ -974,8 +986,9 @@ extern "C" void launch_lstm_pw_fwd(
 // Produces raw-gate grads dgates_t (bf16) and the fp32 cell-grad carry
 // IN PLACE (dc reads+writes the same (row,col)).
 // dh_total = dh_up[t] + dh_carry (either may be null = zero).
-// Vectorized x4 per lane (G13); H % 4 == 0 required.
+// Vectorized W=8 per lane (dwordx4 stores) when H % 8 == 0, else W=4.
 // ---------------------------------------------------------------------------
+template <int W>
 __global__ __launch_bounds__(256) void k_lstm_pw_bwd(
     const __hip_bfloat16* __restrict__ dh_up, long up_stride,   // nullable
     const __hip_bfloat16* __restrict__ dh_carry, long hc_stride, // nullable
@@ -987,41 +1000,47 @@ __global__ __launch_bounds__(256) void k_lstm_pw_bwd(
     int B, int H, int first_step /* dc uninitialised: treat as dcT or 0 */,
     const float* __restrict__ dcT, long dcT_stride)             // nullable
 {
+    typedef __bf16 bfv __attribute__((ext_vector_type(W)));
+    typedef float fv __attribute__((ext_vector_type(W)));
     const int row = blockIdx.y;
     if (row >= B) return;
-    const int col0 = (blockIdx.x * blockDim.x + threadIdx.x) * 4;
+    const int col0 = (blockIdx.x * blockDim.x + threadIdx.x) * W;
     if (col0 >= H) return;
 
-    float dh[4] = {0, 0, 0, 0};
-    if (dh_up) {
-        bf16x4 v = *(const bf16x4*)(dh_up + (long)row * up_stride + col0);
+    fv dh;
 #pragma unroll
-        for (int j = 0; j < 4; ++j) dh[j] += (float)v[j];
+    for (int j = 0; j < W; ++j) dh[j] = 0.f;
+    if (dh_up) {
+        bfv v = *(const bfv*)(dh_up + (long)row * up_stride + col0);
+#pragma unroll
+        for (int j = 0; j < W; ++j) dh[j] += (float)v[j];
     }
     if (dh_carry) {
-        bf16x4 v = *(const bf16x4*)(dh_carry + (long)row * hc_stride + col0);
+        bfv v = *(const bfv*)(dh_carry + (long)row * hc_stride + col0);
 #pragma unroll
-        for (int j = 0; j < 4; ++j) dh[j] += (float)v[j];
+        for (int j = 0; j < W; ++j) dh[j] += (float)v[j];
     }
-    f32x4 dcv;
+    fv dcv;
     if (first_step) {
-        if (dcT) dcv = *(const f32x4*)(dcT + (long)row * dcT_stride + col0);
-        else dcv = f32x4{0, 0, 0, 0};
-    } else {
-        dcv = *(const f32x4*)(dc + (long)row * dc_stride + col0);
-    }
-    const f32x4 cc = *(const f32x4*)(c_cur + (long)row * cc_stride + col0);
-    const f32x4 cp = *(const f32x4*)(c_prev + (long)row * cp_stride + col0);
-    const __hip_bfloat16* ap = acts + (long)row * act_stride + col0;
-    bf16x4 vi = *(const bf16x4*)(ap);
-    bf16x4 vf = *(const bf16x4*)(ap + (long)H);
-    bf16x4 vg = *(const bf16x4*)(ap + (long)2 * H);
-    bf16x4 vo = *(const bf16x4*)(ap + (long)3 * H);
-
-    bf16x4 out_i, out_f, out_g, out_o;
-    f32x4 dc_out;
+        if (dcT) dcv = *(const fv*)(dcT + (long)row * dcT_stride + col0);
+        else
 #pragma unroll
-    for (int j = 0; j < 4; ++j) {
+            for (int j = 0; j < W; ++j) dcv[j] = 0.f;
+    } else {
+        dcv = *(const fv*)(dc + (long)row * dc_stride + col0);
+    }
+    const fv cc = *(const fv*)(c_cur + (long)row * cc_stride + col0);
+    const fv cp = *(const fv*)(c_prev + (long)row * cp_stride + col0);
+    const __hip_bfloat16* ap = acts + (long)row * act_stride + col0;
+    bfv vi = *(const bfv*)(ap);
+    bfv vf = *(const bfv*)(ap + (long)H);
+    bfv vg = *(const bfv*)(ap + (long)2 * H);
+    bfv vo = *(const bfv*)(ap + (long)3 * H);
+
+    bfv out_i, out_f, out_g, out_o;
+    fv dc_out;
+#pragma unroll
+    for (int j = 0; j < W; ++j) {
         const float i = (float)vi[j], f = (float)vf[j];
         const float g = (float)vg[j], o = (float)vo[j];
         const float tc = tanhf_(cc[j]);
@@ -1037,11 +1056,11 @@ __global__ __launch_bounds__(256) void k_lstm_pw_bwd(
         out_o[j] = (__bf16)(d_o * o * (1.0f - o));
     }
     __hip_bfloat16* dgp = dgates + (long)row * dg_stride + col0;
-    *(bf16x4*)(dgp)               = out_i;
-    *(bf16x4*)(dgp + (long)H)     = out_f;
-    *(bf16x4*)(dgp + (long)2 * H) = out_g;
-    *(bf16x4*)(dgp + (long)3 * H) = out_o;
-    *(f32x4*)(dc + (long)row * dc_stride + col0) = dc_out;
+    *(bfv*)(dgp)               = out_i;
+    *(bfv*)(dgp + (long)H)     = out_f;
+    *(bfv*)(dgp + (long)2 * H) = out_g;
+    *(bfv*)(dgp + (long)3 * H) = out_o;
+    *(fv*)(dc + (long)row * dc_stride + col0) = dc_out;
 }
 
 // ---------------------------------------------------------------------------
@@ -1334,13 +1353,23 @@ void launch_lstm_pw_bwd(
     const float* dcT, long dcT_stride, hipStream_t stream)
 {
     dim3 block(256);
-    dim3 grid((H / 4 + 255) / 256, B);
-    hipLaunchKernelGGL(k_lstm_pw_bwd, grid, block, 0, stream,
-        (const __hip_bfloat16*)dh_up, up_stride,
-        (const __hip_bfloat16*)dh_carry, hc_stride,
-        (const __hip_bfloat16*)acts, act_stride, c_cur, cc_stride,
-        c_prev, cp_stride, (__hip_bfloat16*)dgates, dg_stride,
-        dc, dc_stride, B, H, first_step, dcT, dcT_stride);
+    if (H % 8 == 0) {
+        dim3 grid((H / 8 + 255) / 256, B);
+        hipLaunchKernelGGL(k_lstm_pw_bwd<8>, grid, block, 0, stream,
+            (const __hip_bfloat16*)dh_up, up_stride,
+            (const __hip_bfloat16*)dh_carry, hc_stride,
+            (const __hip_bfloat16*)acts, act_stride, c_cur, cc_stride,
+            c_prev, cp_stride, (__hip_bfloat16*)dgates, dg_stride,
+            dc, dc_stride, B, H, first_step, dcT, dcT_stride);
+    } else {
+        dim3 grid((H / 4 + 255) / 256, B);
+        hipLaunchKernelGGL(k_lstm_pw_bwd<4>, grid, block, 0, stream,
+            (const __hip_bfloat16*)dh_up, up_stride,
+            (const __hip_bfloat16*)dh_carry, hc_stride,
+            (const __hip_bfloat16*)acts, act_stride, c_cur, cc_stride,
+            c_prev, cp_stride, (__hip_bfloat16*)dgates, dg_stride,
+            dc, dc_stride, B, H, first_step, dcT, dcT_stride);
+    }
 }
 
 static inline int norm_grid(long n) {
