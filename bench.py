@@ -75,10 +75,15 @@ def main():
     pool = []
     for _ in range(4):
         # TIME-MAJOR (T, B, ·): the engine's native layout
-        x = torch.randn(T, B, cfg.model.input_size, generator=gen, device=dev,
-                        dtype=torch.float32).to(dt)
-        y = torch.randint(0, cfg.model.num_classes, (T, B), generator=gen,
-                          device=dev)
+        if cfg.model.head == "lm":
+            stream = torch.randint(0, cfg.model.vocab_size, (T + 1, B),
+                                   generator=gen, device=dev)
+            x, y = stream[:-1], stream[1:]
+        else:
+            x = torch.randn(T, B, cfg.model.input_size, generator=gen,
+                            device=dev, dtype=torch.float32).to(dt)
+            y = torch.randint(0, cfg.model.num_classes, (T, B), generator=gen,
+                              device=dev)
         pool.append((x, y))
 
     def sync():
