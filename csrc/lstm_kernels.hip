@@ -917,10 +917,14 @@ __global__ __launch_bounds__(256) void k_lstm_pw_fwd(
 {
     typedef __bf16 bfv __attribute__((ext_vector_type(W)));
     typedef float fv __attribute__((ext_vector_type(W)));
-    const int row = blockIdx.y;
+    // flattened (row, col) indexing: every thread active regardless of H/W
+    // vs blockDim (a (H/W)-per-block geometry left half the block idle at
+    // H=1024, W=8 -- measured 3.0 TB/s instead of ~6).
+    const long idx = (long)blockIdx.x * blockDim.x + threadIdx.x;
+    const int cpr = H / W;            // vec-chunks per row
+    const int row = (int)(idx / cpr);
     if (row >= B) return;
-    const int col0 = (blockIdx.x * blockDim.x + threadIdx.x) * W;
-    if (col0 >= H) return;
+    const int col0 = (int)(idx % cpr) * W;
     const __hip_bfloat16* rp = raw_rec + (long)row * rr_stride + col0;
     const __hip_bfloat16* pp = pre + (long)row * pre_stride + col0;
     fv raw[4];
@@ -965,14 +969,14 @@ extern "C" void launch_lstm_pw_fwd(
 {
     dim3 block(256);
     if (H % 8 == 0) {
-        dim3 grid((H / 8 + 255) / 256, B);
+        dim3 grid((unsigned)(((long)B * (H / 8) + 255) / 256));
         hipLaunchKernelGGL(k_lstm_pw_fwd<8>, grid, block, 0, stream,
             (const __hip_bfloat16*)raw_rec, rr_stride,
             (const __hip_bfloat16*)pre, pre_stride, c_prev, cp_stride,
             (__hip_bfloat16*)h_out, ho_stride, c_out, co_stride,
             (__hip_bfloat16*)acts, act_stride, B, H);
     } else {
-        dim3 grid((H / 4 + 255) / 256, B);
+        dim3 grid((unsigned)(((long)B * (H / 4) + 255) / 256));
         hipLaunchKernelGGL(k_lstm_pw_fwd<4>, grid, block, 0, stream,
             (const __hip_bfloat16*)raw_rec, rr_stride,
             (const __hip_bfloat16*)pre, pre_stride, c_prev, cp_stride,
@@ -1002,10 +1006,11 @@ __global__ __launch_bounds__(256) void k_lstm_pw_bwd(
 {
     typedef __bf16 bfv __attribute__((ext_vector_type(W)));
     typedef float fv __attribute__((ext_vector_type(W)));
-    const int row = blockIdx.y;
+    const long idx = (long)blockIdx.x * blockDim.x + threadIdx.x;
+    const int cpr = H / W;
+    const int row = (int)(idx / cpr);
     if (row >= B) return;
-    const int col0 = (blockIdx.x * blockDim.x + threadIdx.x) * W;
-    if (col0 >= H) return;
+    const int col0 = (int)(idx % cpr) * W;
 
     fv dh;
 #pragma unroll
@@ -1354,7 +1359,7 @@ void launch_lstm_pw_bwd(
 {
     dim3 block(256);
     if (H % 8 == 0) {
-        dim3 grid((H / 8 + 255) / 256, B);
+        dim3 grid((unsigned)(((long)B * (H / 8) + 255) / 256));
         hipLaunchKernelGGL(k_lstm_pw_bwd<8>, grid, block, 0, stream,
             (const __hip_bfloat16*)dh_up, up_stride,
             (const __hip_bfloat16*)dh_carry, hc_stride,
@@ -1362,7 +1367,7 @@ void launch_lstm_pw_bwd(
             c_prev, cp_stride, (__hip_bfloat16*)dgates, dg_stride,
             dc, dc_stride, B, H, first_step, dcT, dcT_stride);
     } else {
-        dim3 grid((H / 4 + 255) / 256, B);
+        dim3 grid((unsigned)(((long)B * (H / 4) + 255) / 256));
         hipLaunchKernelGGL(k_lstm_pw_bwd<4>, grid, block, 0, stream,
             (const __hip_bfloat16*)dh_up, up_stride,
             (const __hip_bfloat16*)dh_carry, hc_stride,
