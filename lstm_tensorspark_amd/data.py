@@ -38,7 +38,11 @@ class SyntheticSeqData:
         self.base_seed = t.seed
 
     def _gen(self, step: int) -> torch.Generator:
-        g = torch.Generator(device="cpu")
+        # On GPU, generate device-side: an HBM-sized batch (e.g. 16384x128
+        # sequences) takes ~10 s to randn on the host but <10 ms on-device.
+        # Determinism contract holds per (seed, rank, step, device-kind).
+        dev = self.device if self.device.type == "cuda" else torch.device("cpu")
+        g = torch.Generator(device=dev)
         # distinct stream per (seed, rank, step)
         g.manual_seed(self.base_seed * 1000003 + self.rank * 7919 + step)
         return g
@@ -51,17 +55,21 @@ class SyntheticSeqData:
            "classifier" -- see models/heads.
         """
         g = self._gen(step)
+        dev = g.device
         B, T = self.batch_size, self.seq_len
         task = self.task
         if task == "random":
-            x = torch.randn(T, B, self.input_size, generator=g)
+            x = torch.randn(T, B, self.input_size, generator=g, device=dev)
             head = self.cfg.model.head
             if head == "classifier":
-                y = torch.randint(0, self.cfg.model.num_classes, (B,), generator=g)
+                y = torch.randint(0, self.cfg.model.num_classes, (B,),
+                                  generator=g, device=dev)
             elif head == "seq_classifier":
-                y = torch.randint(0, self.cfg.model.num_classes, (T, B), generator=g)
+                y = torch.randint(0, self.cfg.model.num_classes, (T, B),
+                                  generator=g, device=dev)
             else:
-                y = torch.randn(T, B, self.cfg.model.hidden_size, generator=g)
+                y = torch.randn(T, B, self.cfg.model.hidden_size, generator=g,
+                                device=dev)
             return (x.to(self.device, self.dtype), y.to(self.device))
         if task == "copy":
             # Learnable task: predict the input class seen `delay` steps ago.
@@ -69,18 +77,19 @@ class SyntheticSeqData:
             # sequence (first `delay` targets are class 0 = "blank").
             C = self.cfg.model.num_classes
             d = self.cfg.data.copy_delay
-            toks = torch.randint(1, min(C, self.input_size), (T, B), generator=g)
-            x = torch.zeros(T, B, self.input_size)
+            toks = torch.randint(1, min(C, self.input_size), (T, B),
+                                 generator=g, device=dev)
+            x = torch.zeros(T, B, self.input_size, device=dev)
             x.scatter_(2, toks.unsqueeze(-1), 1.0)
-            y = torch.zeros(T, B, dtype=torch.long)
+            y = torch.zeros(T, B, dtype=torch.long, device=dev)
             if d < T:
                 y[d:] = toks[: T - d]
             return (x.to(self.device, self.dtype), y.to(self.device))
         if task == "lm":
             V = self.cfg.model.vocab_size
             assert V > 1, "lm task requires model.vocab_size"
-            stream = torch.randint(0, V, (T + 1, B), generator=g)
-            x, y = stream[:-1], stream[1:]
+            stream = torch.randint(0, V, (T + 1, B), generator=g, device=dev)
+            x, y = stream[:-1].contiguous(), stream[1:].contiguous()
             return (x.to(self.device), y.to(self.device))
         raise ValueError(f"unknown data task {task!r}")
 

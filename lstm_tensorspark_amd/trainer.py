@@ -197,6 +197,31 @@ class Trainer:
     # alias matching the reference's API surface
     train = fit
 
+    @torch.no_grad()
+    def evaluate(self, steps: int = 10, data=None) -> dict:
+        """Average loss over `steps` held-out synthetic batches (the eval
+        counterpart of the reference's Session.run(logits) path, SURVEY.md
+        §3.3).  Uses a disjoint RNG stream (step offset 1e6) so eval batches
+        never coincide with training batches.  All-reduces the mean loss
+        across ranks."""
+        self.model.eval()
+        src = data if data is not None else self.data
+        total, n = 0.0, 0
+        for i in range(steps):
+            x, y = src.batch(1_000_000 + i)
+            out, _ = self.model(x)
+            total += float(self.model.loss(out, y).detach())
+            n += 1
+        self.model.train()
+        mean = total / max(n, 1)
+        if dist.is_initialized() and self.ctx.world_size > 1:
+            t = torch.tensor([mean], dtype=torch.float64)
+            if self.ctx.backend == "nccl":
+                t = t.to(self.device)
+            dist.all_reduce(t, op=dist.ReduceOp.AVG)
+            mean = float(t.item())
+        return {"eval_loss": mean, "steps": n}
+
     # ------------------------------------------------------------------
     def save(self, path: str):
         if not self.ctx.is_main:

@@ -85,3 +85,20 @@ def test_lm_head_train():
     t = Trainer(cfg)
     t.fit(2)
     assert t.step == 2
+
+
+def test_evaluate_api():
+    """evaluate(): disjoint eval stream, mean loss finite, model back in
+    train mode afterwards."""
+    from lstm_tensorspark_amd.config import Config, ModelConfig, TrainConfig
+    from lstm_tensorspark_amd.trainer import Trainer
+
+    cfg = Config(model=ModelConfig(input_size=16, hidden_size=16, num_classes=8),
+                 train=TrainConfig(batch_size=4, seq_len=8, log_interval=0))
+    t = Trainer(cfg, device="cpu")
+    r = t.evaluate(steps=3)
+    assert r["steps"] == 3 and r["eval_loss"] == r["eval_loss"]
+    assert t.model.training
+    # deterministic given the same trainer state
+    r2 = t.evaluate(steps=3)
+    assert abs(r2["eval_loss"] - r["eval_loss"]) < 1e-6
