@@ -29,6 +29,17 @@ def _pname(layer: int, rev: bool, which: str) -> str:
     return f"layer{layer}{'_rev' if rev else ''}/{which}"
 
 
+_REV_STREAM = None
+
+
+def _rev_stream():
+    """Second HIP stream for the reverse direction of bidirectional layers."""
+    global _REV_STREAM
+    if _REV_STREAM is None:
+        _REV_STREAM = torch.cuda.Stream()
+    return _REV_STREAM
+
+
 class LSTMModel(nn.Module):
     """Multi-layer (bi)LSTM + head. Compute dtype follows parameter dtype;
     cell state is always fp32."""
@@ -122,14 +133,35 @@ class LSTMModel(nn.Module):
         new_state = []
         si = 0
         for l in range(cfg.num_layers):
-            outs = []
-            for rev in ([False, True] if cfg.bidirectional else [False]):
+            if cfg.bidirectional and x.is_cuda:
+                # The two directions are independent: run the REVERSE pass
+                # on a second HIP stream so its sequential recurrence
+                # overlaps the forward direction's (SURVEY.md §7 hard part
+                # 5).  Autograd records each op's stream, so the two
+                # backward recurrences overlap the same way; the engine
+                # inserts the cross-stream grad syncs itself.
+                cur = torch.cuda.current_stream()
+                s2 = _rev_stream()
+                s2.wait_stream(cur)            # x (and carries) are ready
+                st = state[si + 1] if state is not None else None
+                with torch.cuda.stream(s2):
+                    x.record_stream(s2)
+                    h_rev, hT_r, cT_r = self._layer_io(x, l, True, st)
                 st = state[si] if state is not None else None
-                h_seq, hT, cT = self._layer_io(x, l, rev, st)
-                outs.append(h_seq)
-                new_state.append((hT, cT))
-                si += 1
-            x = outs[0] if len(outs) == 1 else torch.cat(outs, dim=2)
+                h_fwd, hT_f, cT_f = self._layer_io(x, l, False, st)
+                cur.wait_stream(s2)            # before the concat reads
+                new_state += [(hT_f, cT_f), (hT_r, cT_r)]
+                si += 2
+                x = torch.cat([h_fwd, h_rev], dim=2)
+            else:
+                outs = []
+                for rev in ([False, True] if cfg.bidirectional else [False]):
+                    st = state[si] if state is not None else None
+                    h_seq, hT, cT = self._layer_io(x, l, rev, st)
+                    outs.append(h_seq)
+                    new_state.append((hT, cT))
+                    si += 1
+                x = outs[0] if len(outs) == 1 else torch.cat(outs, dim=2)
             if cfg.dropout > 0 and self.training and l < cfg.num_layers - 1:
                 x = F.dropout(x, p=cfg.dropout)
         return x, new_state

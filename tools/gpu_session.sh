@@ -1,8 +1,21 @@
 set -x
-mkdir -p gpurun_out
-R=$GRAFT_REPO_ROOT
-cd /tmp && export TMPDIR=/tmp
-timeout 300 rocprofv3 --pmc MfmaUtil VALUBusy OccupancyPercent -d $R/gpurun_out/pmcu1 --output-format csv -- python $R/tools/pmc_rec_fwd.py --variant 2 --iters 30 > $R/gpurun_out/pmcu1.log 2>&1 || tail -3 $R/gpurun_out/pmcu1.log
-timeout 300 rocprofv3 --pmc MfmaUtil VALUBusy OccupancyPercent -d $R/gpurun_out/pmcu2 --output-format csv -- python $R/tools/pmc_rec_fwd.py --variant 32 --iters 30 > $R/gpurun_out/pmcu2.log 2>&1 || tail -3 $R/gpurun_out/pmcu2.log
-timeout 400 rocprofv3 --pmc MfmaUtil VALUBusy OccupancyPercent -d $R/gpurun_out/pmcu3 --output-format csv -- python $R/bench.py --steps 3 --warmup 1 --batch 4096 > $R/gpurun_out/pmcu3.log 2>&1 || tail -3 $R/gpurun_out/pmcu3.log
-find $R/gpurun_out/pmcu* -name "*counter*" | head
+python -m pytest tests -m gpu -q 2>&1 | tail -1
+timeout 400 python bench.py --gpus 1 --steps 6 --warmup 2 --preset bidir-4l 2>&1 | tail -1
+python - <<'PY'
+import sys, torch, time; sys.path.insert(0, '.')
+from lstm_tensorspark_amd.config import get_preset
+from lstm_tensorspark_amd.trainer import Trainer
+from lstm_tensorspark_amd.ops import extension
+ext = extension.get_extension()
+cfg = get_preset("bidir-4l"); cfg.train.log_interval = 0; cfg.train.ckpt_dir = ""
+t = Trainer(cfg, device="cuda:0")
+b = t.data.batch(0)
+for mode, name in [(-1, "auto(fused@B1024)"), (1, "split")]:
+    ext.set_split_fwd(mode)
+    for _ in range(2): t.train_step(b)
+    torch.cuda.synchronize(); t0 = time.perf_counter()
+    for _ in range(4): t.train_step(b)
+    torch.cuda.synchronize()
+    print(f"bidir split={name}: {(time.perf_counter()-t0)/4*1e3:.0f} ms/step")
+ext.set_split_fwd(-1)
+PY
