@@ -50,8 +50,10 @@ bool g_persistent_enabled = false;  // per-step glds path is faster (kbench6)
 // Split forward: hipBLASLt recurrent GEMM + k_lstm_pw_fwd pointwise.  The
 // library GEMM runs the (B,H)x(H,4H) step shape ~1.9x faster than the best
 // fused hand kernel at B=4096 (908 vs ~475 TF/s), which more than pays for
-// the raw-gates round trip.  -1 = auto (split for B >= 512; measured split
-// wins down to B=512 at h>=1024: 1107 vs 1209 ms/step on bidir-4l B=1024).
+// the raw-gates round trip.  -1 = auto: split when the per-step GEMM is
+// big enough for the library to win, B*H >= 2M (measured at H=1024:
+// fused wins to B=512 (18.7 vs 22.0 us), ~tie at B=1024, split wins from
+// B=2048; at H=2048 split wins from B=1024: 1107 vs 1209 ms/step).
 int g_split_fwd = -1;
 std::set<long> g_validated_grids;
 
@@ -89,7 +91,7 @@ std::vector<Tensor> lstm_seq_fwd(Tensor pre, Tensor w_h, Tensor h0, Tensor c0,
     // W_h^T (4H, H): one transposed copy per layer call so each lane's MFMA
     // B-fragment is a contiguous 16-byte load (fused path only).
     const bool need_wt = g_persistent_enabled
-        || !((g_split_fwd < 0) ? (B >= 512) : (g_split_fwd != 0));
+        || !((g_split_fwd < 0) ? (B * H >= 2097152) : (g_split_fwd != 0));
     auto w_t = need_wt ? w_h.t().contiguous() : w_h;
 
     auto opts_bf = pre.options();
@@ -131,7 +133,7 @@ std::vector<Tensor> lstm_seq_fwd(Tensor pre, Tensor w_h, Tensor h0, Tensor c0,
         }
     }
 
-    const bool split = (g_split_fwd < 0) ? (B >= 512) : (g_split_fwd != 0);
+    const bool split = (g_split_fwd < 0) ? (B * H >= 2097152) : (g_split_fwd != 0);
     at::Tensor raw_buf;
     if (split) raw_buf = at::empty({B, H4}, opts_bf);
     for (long k = 0; k < T; ++k) {
