@@ -1,0 +1,39 @@
+"""The driver depends on bench.py's exact output contract: one JSON line
+from rank 0 with the required keys.  Run on CPU at plumbing scale."""
+import json
+import os
+import subprocess
+import sys
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def test_bench_cpu_contract():
+    out = subprocess.run(
+        [sys.executable, os.path.join(REPO, "bench.py"),
+         "--cpu", "--steps", "2", "--warmup", "1"],
+        capture_output=True, text=True, timeout=300, cwd=REPO)
+    assert out.returncode == 0, out.stderr[-2000:]
+    lines = [l for l in out.stdout.strip().splitlines() if l.startswith("{")]
+    assert len(lines) == 1, out.stdout
+    rec = json.loads(lines[0])
+    for key in ("metric", "value", "unit", "n_gpus", "steps", "warmup",
+                "ms_per_step", "higher_is_better", "scaling", "vs_baseline",
+                "dtype", "data", "config"):
+        assert key in rec, key
+    assert rec["data"] == "synthetic"
+    assert rec["scaling"] == "weak"
+    assert rec["value"] > 0
+    for key in ("model", "global_batch", "seq_len", "parallelism"):
+        assert key in rec["config"], key
+
+
+def test_train_cli_smoke(tmp_path):
+    out = subprocess.run(
+        [sys.executable, os.path.join(REPO, "train.py"),
+         "--preset", "plumbing-cpu", "--set", "train.steps=2",
+         "--set", "train.batch_size=4", "--set", "train.seq_len=8",
+         "--set", f"train.ckpt_dir={tmp_path}", "--set", "train.log_interval=1"],
+        capture_output=True, text=True, timeout=300, cwd=REPO)
+    assert out.returncode == 0, out.stderr[-2000:]
+    assert os.path.exists(os.path.join(str(tmp_path), "ckpt_final.pt"))
