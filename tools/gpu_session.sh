@@ -1,27 +1,21 @@
 set -x
-python -m pytest tests -m gpu -q 2>&1 | tail -1
-timeout 400 python bench.py --gpus 1 --steps 6 --warmup 2 --preset bidir-4l 2>&1 | tail -1
-timeout 400 python bench.py --gpus 1 --steps 6 --warmup 2 --preset lm-50k 2>&1 | tail -1
-timeout 400 python bench.py --gpus 1 --steps 15 --warmup 3 2>&1 | tail -1
 python - <<'PY'
-import sys, torch; sys.path.insert(0, '.')
-from lstm_tensorspark_amd.ops import extension
-ext = extension.get_extension()
-g = torch.Generator(device="cuda").manual_seed(0)
-for B in (256, 512, 1024):
-    H, T = 1024, 16
-    pre = torch.randn(T, B, 4*H, generator=g, device="cuda").bfloat16()
-    w_h = (torch.randn(H, 4*H, generator=g, device="cuda") / H**0.5).bfloat16()
-    h0 = torch.randn(B, H, generator=g, device="cuda").bfloat16()
-    c0 = torch.randn(B, H, generator=g, device="cuda")
-    for mode in (0, 1):
-        ext.set_split_fwd(mode)
-        for _ in range(3): ext.lstm_seq_fwd(pre, w_h, h0, c0, False)
-        torch.cuda.synchronize()
-        s, e = torch.cuda.Event(enable_timing=True), torch.cuda.Event(enable_timing=True)
-        s.record()
-        for _ in range(10): ext.lstm_seq_fwd(pre, w_h, h0, c0, False)
-        e.record(); torch.cuda.synchronize()
-        print(f"B={B} split={mode}: {1e3*s.elapsed_time(e)/10/T:.1f} us/step")
-ext.set_split_fwd(-1)
+import sys, torch, time; sys.path.insert(0, '.')
+from lstm_tensorspark_amd.config import get_preset
+from lstm_tensorspark_amd.trainer import Trainer
+from lstm_tensorspark_amd.ops import lstm_layer as _m
+import lstm_tensorspark_amd.ops.lstm_layer as m
+cfg = get_preset("mi355x-1gpu"); cfg.train.log_interval = 0; cfg.train.ckpt_dir = ""
+t = Trainer(cfg, device="cuda:0")
+g = torch.Generator(device="cuda"); g.manual_seed(1)
+B, T = cfg.train.batch_size, cfg.train.seq_len
+x = torch.randn(T, B, 1024, generator=g, device="cuda").bfloat16()
+y = torch.randint(0, 128, (T, B), generator=g, device="cuda")
+for side in (True, False):
+    m.enable_side_dw(side)
+    for _ in range(2): t.train_step((x, y))
+    torch.cuda.synchronize(); t0 = time.perf_counter()
+    for _ in range(6): t.train_step((x, y))
+    torch.cuda.synchronize()
+    print(f"side_dw={side}: {(time.perf_counter()-t0)/6*1e3:.1f} ms/step")
 PY
