@@ -78,6 +78,15 @@ def build(force: bool = False) -> str:
     includes, defines, link = torch_flags()
     common = [f"--offload-arch={ARCH}", "-O3", "-std=c++17", "-fPIC",
               "-Wno-unused-result"]
+    # Debug / sanitizer build (SURVEY.md §5 race-detection row):
+    #   LSTM_TS_BUILD_MODE=debug -> -O1 -g + device asserts
+    #   LSTM_TS_ASAN=1           -> host AddressSanitizer on the bindings
+    if os.environ.get("LSTM_TS_BUILD_MODE") == "debug":
+        common = [f"--offload-arch={ARCH}", "-O1", "-g", "-std=c++17",
+                  "-fPIC", "-Wno-unused-result", "-DLSTM_TS_DEBUG"]
+    if os.environ.get("LSTM_TS_ASAN") == "1":
+        common += ["-fsanitize=address", "-shared-libasan"]
+        link += ["-fsanitize=address", "-shared-libasan"]
     objs = []
     for src in SOURCES:
         obj = os.path.join(BUILD, os.path.splitext(src)[0] + ".o")

@@ -47,6 +47,12 @@ class TrainConfig:
     ckpt_interval: int = 0        # 0 => only at end if ckpt_dir set
     ckpt_dir: str = ""
     seed: int = 1234
+    # capture the whole train step in a hipGraph (torch.cuda.CUDAGraph) and
+    # replay it: removes ALL per-step launch/dispatch overhead.  Eligible
+    # when: GPU, world_size==1, optimizer=="sgd" (no host-side step-count
+    # scalars), single BPTT window, dropout==0.  Ineligible configs fall
+    # back to eager silently.
+    cuda_graph: bool = False
     # fault injection (failure-detection tests): raise on this rank at this
     # step; -1 = disabled
     fault_step: int = -1
@@ -57,6 +63,7 @@ class TrainConfig:
 class DistConfig:
     backend: str = "auto"         # "auto" | "gloo" | "nccl" (nccl IS RCCL on ROCm)
     bucket_mb: int = 16           # gradient all-reduce bucket size, sized for xGMI links
+    side_dw: bool = False         # dW GEMMs on a side stream (neutral at HBM-sized B)
 
 
 @dataclass

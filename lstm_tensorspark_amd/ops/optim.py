@@ -58,9 +58,14 @@ class MasterOptimizer:
                     g.mul_(coef.to(g.dtype))
         return norm
 
-    def zero_grad(self):
+    def zero_grad(self, set_to_none: bool = True):
+        # set_to_none=False keeps grad storage stable (required inside a
+        # captured hipGraph, where tensors must not be re-allocated)
         for p in self.params:
-            p.grad = None
+            if set_to_none:
+                p.grad = None
+            elif p.grad is not None:
+                p.grad.zero_()
 
     # -- checkpoint --------------------------------------------------------
     def state_dict(self) -> dict:

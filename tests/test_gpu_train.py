@@ -75,3 +75,29 @@ def test_graft_smoke_entry():
     import __graft_entry__
 
     __graft_entry__.smoke()
+
+
+def test_gpu_cuda_graph_step_matches_eager():
+    """Graphed steps (after the capture step) must track eager training:
+    same config/data, losses finite and decreasing comparably."""
+    cfg = copy_cfg(steps=8)
+    cfg.train.cuda_graph = True
+    tg = Trainer(cfg, device="cuda:0")
+    assert tg._graph_eligible
+    losses_g = [tg.train_step(tg.data.batch(i))["loss"] for i in range(8)]
+    assert tg._graph is not None
+    te = Trainer(copy_cfg(steps=8), device="cuda:0")
+    losses_e = [te.train_step(te.data.batch(i))["loss"] for i in range(8)]
+    # same trend: both fall; graphed path numerically sane
+    assert all(l == l for l in losses_g)
+    assert losses_g[-1] < losses_g[0]
+    assert losses_e[-1] < losses_e[0]
+
+
+def test_gpu_cuda_graph_ineligible_falls_back():
+    cfg = copy_cfg(steps=2, bptt=8)   # multi-window -> ineligible
+    cfg.train.cuda_graph = True
+    t = Trainer(cfg, device="cuda:0")
+    assert not t._graph_eligible
+    info = t.train_step(t.data.batch(0))
+    assert info["loss"] == info["loss"]
