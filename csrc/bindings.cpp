@@ -224,8 +224,10 @@ std::vector<Tensor> lstm_seq_bwd(c10::optional<Tensor> dh_seq_opt,
     auto dgates = at::empty({T, B, H4}, acts.options());
     auto dc = at::empty({B, H}, acts.options().dtype(at::kFloat));
     auto dh_buf = at::empty({B, H}, acts.options());
-    // W_h^T view for the recurrent grad GEMM (rocBLAS handles the transpose)
-    auto w_h_t = w_h.t();
+    // W_h^T as a CONTIGUOUS copy (one 8 MB transpose per layer-backward):
+    // hipBLASLt runs the per-step (B,4H)x(4H,H) grad GEMM ~1.4x faster on
+    // the plain-NN layout than on the transposed view (980 vs 670 TF/s).
+    auto w_h_t = w_h.t().contiguous();
 
     auto* act_p = (const char*)acts.data_ptr();
     auto* cs_p = (const float*)c_seq.data_ptr();
