@@ -68,7 +68,7 @@ class DistConfig:
 
 @dataclass
 class DataConfig:
-    task: str = "random"          # "random" | "copy" | "lm"
+    task: str = "random"          # "random" | "copy" | "lm" | "lm_assoc"
     copy_delay: int = 4           # for the learnable copy task
 
 

@@ -91,6 +91,23 @@ class SyntheticSeqData:
             stream = torch.randint(0, V, (T + 1, B), generator=g, device=dev)
             x, y = stream[:-1].contiguous(), stream[1:].contiguous()
             return (x.to(self.device), y.to(self.device))
+        if task == "lm_assoc":
+            # LEARNABLE next-token task: the sequence follows a fixed affine
+            # map  tok_{t+1} = (a*tok_t + c) mod V  (a,c from the config
+            # seed), so a capable model drives the loss from ln(V) toward 0.
+            # The plain "lm" task is uniform random (loss pinned at ln V) --
+            # right for throughput, useless for convergence checks.
+            V = self.cfg.model.vocab_size
+            assert V > 1, "lm_assoc requires model.vocab_size"
+            a = 2 * (self.base_seed % 1000) + 1          # odd -> bijective
+            c = (self.base_seed // 7) % V
+            t0 = torch.randint(0, V, (1, B), generator=g, device=dev)
+            toks = [t0]
+            for _ in range(T):
+                toks.append((toks[-1] * a + c) % V)
+            stream = torch.cat(toks, dim=0)
+            x, y = stream[:-1].contiguous(), stream[1:].contiguous()
+            return (x.to(self.device), y.to(self.device))
         raise ValueError(f"unknown data task {task!r}")
 
     def __iter__(self):

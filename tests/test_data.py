@@ -54,3 +54,15 @@ def test_lm_task_shapes():
     assert x.dtype == torch.long and x.shape == (12, 4)  # time-major
     # next-token prediction: y is x shifted by one
     assert torch.equal(x[1:], y[:-1])
+
+
+def test_lm_assoc_learnable_structure():
+    cfg = _cfg("lm")
+    cfg.data.task = "lm_assoc"
+    d = SyntheticSeqData(cfg, rank=0)
+    x, y = d.batch(0)
+    assert x.shape == (12, 4) and y.shape == (12, 4)
+    # y follows the affine map of x (deterministic given the seed)
+    a = 2 * (cfg.train.seed % 1000) + 1
+    c = (cfg.train.seed // 7) % cfg.model.vocab_size
+    assert torch.equal(y, (x * a + c) % cfg.model.vocab_size)
