@@ -200,3 +200,40 @@ class LSTMModel(nn.Module):
         if state is None:
             return None
         return [(h.detach(), c.detach()) for h, c in state]
+
+    @torch.no_grad()
+    def generate(self, prompt: torch.Tensor, steps: int,
+                 temperature: float = 0.0,
+                 generator: "Optional[torch.Generator]" = None) -> torch.Tensor:
+        """Autoregressive decode for the LM head (the serving path).
+
+        prompt: (Tp, B) long tokens (time-major).  Returns (Tp+steps, B):
+        the prompt followed by `steps` generated tokens.  temperature 0 =
+        greedy argmax; >0 = softmax sampling.  The recurrent (h, c) state
+        is carried across steps, so each decode step is one T=1 pass.
+        """
+        assert self.cfg.head == "lm", "generate() requires the LM head"
+        assert not self.cfg.bidirectional, "decode is causal: fwd-only"
+        assert prompt.dim() == 2 and prompt.dtype == torch.long
+        was_training = self.training
+        self.eval()
+        try:
+            out, state = self.forward(prompt)          # (Tp, B, V)
+            toks = [prompt]
+            cur = self._sample(out[-1], temperature, generator)
+            for _ in range(steps - 1):
+                toks.append(cur.unsqueeze(0))
+                out, state = self.forward(cur.unsqueeze(0), state)
+                cur = self._sample(out[-1], temperature, generator)
+            toks.append(cur.unsqueeze(0))
+            return torch.cat(toks, dim=0)
+        finally:
+            if was_training:
+                self.train()
+
+    @staticmethod
+    def _sample(logits, temperature, generator):
+        if temperature <= 0:
+            return logits.argmax(dim=-1)
+        probs = torch.softmax(logits.float() / temperature, dim=-1)
+        return torch.multinomial(probs, 1, generator=generator).squeeze(-1)

@@ -101,3 +101,18 @@ def test_gpu_cuda_graph_ineligible_falls_back():
     assert not t._graph_eligible
     info = t.train_step(t.data.batch(0))
     assert info["loss"] == info["loss"]
+
+
+def test_gpu_lm_generate():
+    from lstm_tensorspark_amd.config import ModelConfig
+    from lstm_tensorspark_amd.models.lstm import LSTMModel
+
+    torch.manual_seed(5)
+    cfg = ModelConfig(input_size=128, hidden_size=128, num_layers=2,
+                      head="lm", vocab_size=512)
+    m = LSTMModel(cfg).to("cuda").bfloat16()
+    prompt = torch.randint(0, 512, (8, 4), device="cuda")
+    out = m.generate(prompt, steps=16)
+    assert out.shape == (24, 4)
+    assert torch.equal(out[:8], prompt)
+    assert (out < 512).all() and (out >= 0).all()
