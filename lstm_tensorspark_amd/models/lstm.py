@@ -4,6 +4,12 @@ Replaces the reference's hand-rolled TF-graph LSTM cell/model module
 (SURVEY.md §2.1 "LSTM model / cell": weights 4 gates x {W_x, W_h, b},
 stepped over the sequence, truncated BPTT, loss, optimizer update).
 
+Inter-layer dropout is applied with torch's native dropout on the layer
+OUTPUT (never on the recurrent path -- standard LSTM practice, and the
+reason it is NOT fused into the cell kernels: the recurrent chain consumes
+the un-dropped h_t, so fusing would require writing a second, dropped copy
+of h -- MORE HBM traffic than the separate elementwise kernel, not less).
+
 Weight naming follows the checkpoint layout contract (SURVEY.md §5
 Checkpoint row): a flat named-weight dict with keys
     layer{i}/W_x, layer{i}/W_h, layer{i}/b           (forward direction)
