@@ -64,6 +64,7 @@ class DistConfig:
     backend: str = "auto"         # "auto" | "gloo" | "nccl" (nccl IS RCCL on ROCm)
     bucket_mb: int = 16           # gradient all-reduce bucket size, sized for xGMI links
     side_dw: bool = False         # dW GEMMs on a side stream (neutral at HBM-sized B)
+    reduce_dtype: str = ""        # "" native (bf16) | "fp32" wire precision for all-reduce
 
 
 @dataclass

@@ -51,8 +51,13 @@ class GradBucketer:
     """
 
     def __init__(self, params, bucket_mb: int = 16,
-                 process_group: Optional[dist.ProcessGroup] = None):
+                 process_group: Optional[dist.ProcessGroup] = None,
+                 reduce_dtype: Optional[torch.dtype] = None):
+        """reduce_dtype: cast each bucket to this dtype for the wire
+        (e.g. torch.float32 to avoid bf16 summation error across 8 ranks);
+        None reduces in the native grad dtype."""
         self.pg = process_group
+        self.reduce_dtype = reduce_dtype
         self.world = dist.get_world_size(process_group) if dist.is_initialized() else 1
         self.params = [p for p in params if p.requires_grad]
         if not self.params:
@@ -125,8 +130,7 @@ class GradBucketer:
         b = self._param_bucket[p]
         b.pending -= 1
         if b.pending == 0:
-            b.work = dist.all_reduce(b.flat, op=dist.ReduceOp.SUM,
-                                     group=self.pg, async_op=True)
+            b.work = self._launch_reduce(b)
 
     def finalize(self):
         """Wait for all in-flight all-reduces and average."""
@@ -138,13 +142,25 @@ class GradBucketer:
             if b.work is None and b.pending > 0:
                 # grads never produced this step (e.g. frozen path): reduce
                 # anyway so ranks stay collective-aligned
-                b.work = dist.all_reduce(b.flat, op=dist.ReduceOp.SUM,
-                                         group=self.pg, async_op=True)
+                b.work = self._launch_reduce(b)
         for b in self.buckets:
             if b.work is not None:
                 b.work.wait()
+            if getattr(b, "wire", None) is not None:
+                b.flat.copy_(b.wire)   # cast back from the fp32 wire buffer
+                b.wire = None
             b.flat.mul_(inv)
         self._reset_pending()
+
+    def _launch_reduce(self, b: "_Bucket"):
+        if (self.reduce_dtype is not None
+                and self.reduce_dtype != b.flat.dtype):
+            b.wire = b.flat.to(self.reduce_dtype)
+            return dist.all_reduce(b.wire, op=dist.ReduceOp.SUM,
+                                   group=self.pg, async_op=True)
+        b.wire = None
+        return dist.all_reduce(b.flat, op=dist.ReduceOp.SUM,
+                               group=self.pg, async_op=True)
 
     def detach(self):
         for h in self._hooks:

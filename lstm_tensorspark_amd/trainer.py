@@ -60,8 +60,11 @@ class Trainer:
         self.optimizer = build_optimizer(cfg.train, self.model.parameters())
         self.bucketer = None
         if self.ctx.world_size > 1:
+            rd = {"fp32": torch.float32, "": None}.get(
+                getattr(cfg.dist, "reduce_dtype", ""), None)
             self.bucketer = GradBucketer(self.model.parameters(),
-                                         bucket_mb=cfg.dist.bucket_mb)
+                                         bucket_mb=cfg.dist.bucket_mb,
+                                         reduce_dtype=rd)
         self.data = SyntheticSeqData(cfg, rank=self.ctx.rank,
                                      world_size=self.ctx.world_size,
                                      device=self.device, dtype=self.dtype)

@@ -101,3 +101,32 @@ def test_bucket_grouping_respects_size():
     params = list(t.model.parameters())
     b = GradBucketer(params, bucket_mb=0)  # degenerate: 1 param per bucket
     assert len(b.buckets) == len(params)
+
+
+def _grads_fp32wire_worker(ctx, tmpdir):
+    cfg = small_cfg()
+    cfg.dist.reduce_dtype = "fp32"
+    t = Trainer(cfg, dist_ctx=ctx)
+    t.train_step(t.data.batch(0))
+    grads = {n: p.grad.detach().clone()
+             for n, p in zip(t.model.weight_dict(), t.model.parameters())}
+    torch.save(grads, os.path.join(tmpdir, f"g32_rank{ctx.rank}.pt"))
+
+
+def test_2rank_fp32_wire_reduction(tmp_path):
+    """fp32 wire precision: averaged grads match the per-rank average."""
+    run_spawned(_grads_fp32wire_worker, world_size=2, backend="gloo",
+                args=(str(tmp_path),), master_port=29541)
+    per_rank = []
+    for rank in range(2):
+        t = Trainer(small_cfg())
+        t.data.rank = rank
+        t.train_step(t.data.batch(0))
+        per_rank.append({n: p.grad.detach().clone()
+                         for n, p in zip(t.model.weight_dict(),
+                                         t.model.parameters())})
+    got = torch.load(os.path.join(str(tmp_path), "g32_rank0.pt"),
+                     weights_only=False)
+    for name in got:
+        want = (per_rank[0][name] + per_rank[1][name]) / 2
+        assert (got[name] - want).abs().max() < 1e-6, name
