@@ -33,6 +33,7 @@ int get_fwd_variant();
 void launch_lstm_pw_fwd(const void*, long, const void*, long, const float*,
                         long, void*, long, float*, long, void*, long, int,
                         int, hipStream_t);
+void launch_quant_fp8_fixed(const void*, void*, long, float, hipStream_t);
 void launch_xent_fwd(const void*, const long*, float*, float*, long, int,
                      hipStream_t);
 void launch_xent_bwd(const void*, const long*, const float*, void*,
@@ -55,6 +56,12 @@ bool g_persistent_enabled = false;  // per-step glds path is faster (kbench6)
 // fused wins to B=512 (18.7 vs 22.0 us), ~tie at B=1024, split wins from
 // B=2048; at H=2048 split wins from B=1024: 1107 vs 1209 ms/step).
 int g_split_fwd = -1;
+// FP8 recurrent GEMM (experimental, MI355X 2x-bf16 MFMA rate): quantize
+// h (tanh-bounded, FIXED scale 448) and W_h (per-call amax scale) to OCP
+// e4m3 and run the per-step GEMM through hipBLASLt _scaled_mm (measured
+// 1939 vs 963 TF/s on the (16384,1024)x(1024,4096) step shape).  Only the
+// recurrent GEMM runs fp8; pre/activations/backward stay bf16.
+bool g_fp8_fwd = false;
 std::set<long> g_validated_grids;
 
 hipStream_t cur_stream() {
@@ -134,8 +141,20 @@ std::vector<Tensor> lstm_seq_fwd(Tensor pre, Tensor w_h, Tensor h0, Tensor c0,
     }
 
     const bool split = (g_split_fwd < 0) ? (B * H >= 2097152) : (g_split_fwd != 0);
-    at::Tensor raw_buf;
+    const bool fp8 = g_fp8_fwd && split;
+    at::Tensor raw_buf, h8, w8_cm, sc_a, sc_b;
     if (split) raw_buf = at::empty({B, H4}, opts_bf);
+    if (fp8) {
+        auto opts_f8 = pre.options().dtype(at::kFloat8_e4m3fn);
+        h8 = at::empty({B, H}, opts_f8);
+        // W_h: per-call amax scale, quantized once; _scaled_mm wants the
+        // second operand COLUMN-major
+        auto w_amax = w_h.abs().amax().to(at::kFloat).clamp_min(1e-8);
+        sc_b = w_amax / 448.0;
+        w8_cm = at::mul(w_h, 448.0 / w_amax).to(at::kFloat8_e4m3fn)
+                    .t().contiguous().t();
+        sc_a = at::full({}, 1.0 / 448.0, pre.options().dtype(at::kFloat));
+    }
     for (long k = 0; k < T; ++k) {
         const long t = reverse ? (T - 1 - k) : k;
         // prev slot holds h_{t-1} (h0 at the first iteration by the slot
@@ -149,7 +168,20 @@ std::vector<Tensor> lstm_seq_fwd(Tensor pre, Tensor w_h, Tensor h0, Tensor c0,
             const long t_prev = reverse ? (t + 1) : (t - 1);
             cp = cs_p + t_prev * B * H;
         }
-        if (split) {
+        if (fp8) {
+            launch_quant_fp8_fixed(ha_p + 2 * slot_prev * B * H,
+                                   h8.data_ptr(), B * H, 448.0f, stream);
+            at::_scaled_mm_out(raw_buf, h8, w8_cm, sc_a, sc_b, {}, {},
+                               at::kBFloat16, true);
+            launch_lstm_pw_fwd(
+                raw_buf.data_ptr(), H4,
+                pre_p + 2 * t * B * H4, H4,
+                cp, H,
+                ha_p + 2 * slot_out * B * H, H,
+                cs_p + t * B * H, H,
+                act_p + 2 * t * B * H4, H4,
+                (int)B, (int)H, stream);
+        } else if (split) {
             at::mm_out(raw_buf, h_all.select(0, slot_prev), w_h);
             launch_lstm_pw_fwd(
                 raw_buf.data_ptr(), H4,
@@ -380,6 +412,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("lstm_rec_fwd_step", &lstm_rec_fwd_step,
           "single fused recurrent step (raw_debug for GEMM-mapping tests)");
     m.def("set_persistent_fwd", [](bool v) { g_persistent_enabled = v; });
+    m.def("set_fp8_fwd", [](bool v) { g_fp8_fwd = v; },
+          "experimental: fp8(e4m3) recurrent GEMM in the split forward");
+    m.def("get_fp8_fwd", []() { return g_fp8_fwd; });
     m.def("set_split_fwd", [](int v) { g_split_fwd = v; },
           "-1 auto (B>=2048), 0 fused hand kernel, 1 hipBLASLt+pointwise");
     m.def("get_split_fwd", []() { return g_split_fwd; });

@@ -20,6 +20,7 @@
 
 #include <hip/hip_runtime.h>
 #include <hip/hip_bf16.h>
+#include <hip/hip_fp8.h>
 
 typedef __bf16 bf16x8 __attribute__((ext_vector_type(8)));
 typedef __bf16 bf16x4 __attribute__((ext_vector_type(4)));
@@ -983,6 +984,42 @@ extern "C" void launch_lstm_pw_fwd(
             (__hip_bfloat16*)h_out, ho_stride, c_out, co_stride,
             (__hip_bfloat16*)acts, act_stride, B, H);
     }
+}
+
+// ---------------------------------------------------------------------------
+// FP8 quantization for the split-path recurrent GEMM: h is tanh-bounded to
+// (-1,1), so a FIXED scale of 448 (e4m3 max) needs no amax pass; the GEMM
+// dequantizes with scale_a = 1/448.  OCP e4m3 (gfx950), not fnuz.
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(256) void k_quant_fp8_fixed(
+    const __hip_bfloat16* __restrict__ in, unsigned char* __restrict__ out,
+    long n, float scale)
+{
+    const long i0 = ((long)blockIdx.x * blockDim.x + threadIdx.x) * 8;
+    if (i0 + 8 <= n) {
+        const bf16x8 v = *(const bf16x8*)(in + i0);
+        unsigned long long pk = 0;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+            const __hip_fp8_e4m3 q((float)v[j] * scale);
+            pk |= (unsigned long long)q.__x << (8 * j);
+        }
+        *(unsigned long long*)(out + i0) = pk;
+    } else {
+        for (long i = i0; i < n; ++i) {
+            const __hip_fp8_e4m3 q(__bfloat162float(in[i]) * scale);
+            out[i] = q.__x;
+        }
+    }
+}
+
+extern "C" void launch_quant_fp8_fixed(const void* in, void* out, long n,
+                                       float scale, hipStream_t stream)
+{
+    dim3 block(256), grid((unsigned)((n / 8 + 255) / 256));
+    hipLaunchKernelGGL(k_quant_fp8_fixed, grid, block, 0, stream,
+                       (const __hip_bfloat16*)in, (unsigned char*)out, n,
+                       scale);
 }
 
 // ---------------------------------------------------------------------------

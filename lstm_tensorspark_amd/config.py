@@ -41,7 +41,7 @@ class TrainConfig:
     adam_betas: tuple = (0.9, 0.999)
     adam_eps: float = 1e-8
     grad_clip: float = 0.0        # global-norm clip; 0 => off
-    dtype: str = "fp32"           # "fp32" | "bf16" (bf16 compute + fp32 master weights)
+    dtype: str = "fp32"           # "fp32" | "bf16" | "fp8" (bf16 + e4m3 recurrent GEMM)
     steps: int = 100
     log_interval: int = 10
     ckpt_interval: int = 0        # 0 => only at end if ckpt_dir set

@@ -36,7 +36,9 @@ from .utils.logging import get_logger
 
 
 def compute_dtype(cfg: Config, device: torch.device) -> torch.dtype:
-    if cfg.train.dtype == "bf16" and device.type == "cuda":
+    # "fp8" = bf16 weights/activations with the recurrent GEMM quantized to
+    # OCP e4m3 inside the C++ driver (experimental; 2x MFMA rate)
+    if cfg.train.dtype in ("bf16", "fp8") and device.type == "cuda":
         return torch.bfloat16
     return torch.float32
 
@@ -74,6 +76,10 @@ class Trainer:
         # Keep it off by default for a simpler stream graph; opt in via
         # dist.side_dw for small-model regimes.  Only safe when autograd
         # STORES (not accumulates) weight grads.
+        if self.device.type == "cuda":
+            from .ops.extension import extension_available, get_extension
+            if extension_available():
+                get_extension().set_fp8_fwd(cfg.train.dtype == "fp8")
         single_window = cfg.train.bptt <= 0 or cfg.train.bptt >= cfg.train.seq_len
         enable_side_dw(getattr(cfg.dist, "side_dw", False)
                        and self.bucketer is None and single_window

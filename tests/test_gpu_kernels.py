@@ -355,3 +355,36 @@ def test_shape_fuzz_seq_fwd_bwd_vs_ref():
             d_hseq, None, None, w_h, a_r, c_r, c0, i % 2 == 1)
         err = (dg.float() - dg_r.float()).abs().max().item()
         assert err < 0.06, f"bwd shape {(B,T,H)} rev={i%2} err={err}"
+
+
+# --------------------------------------------------------------- fp8 mode
+def test_fp8_recurrent_gemm_vs_bf16():
+    """Experimental fp8(e4m3) recurrent GEMM: output tracks the bf16 path
+    within fp8 quantization error (h is tanh-bounded, fixed scale 448)."""
+    from lstm_tensorspark_amd.ops import functional
+    from lstm_tensorspark_amd.ops.extension import get_extension
+
+    ext = get_extension()
+    B, T, H = 2048, 8, 1024  # B*H >= 2M -> split path (fp8 requires split)
+    g = torch.Generator(device="cpu").manual_seed(11)
+
+    def r(*s, scale=1.0):
+        return (torch.randn(*s, generator=g) * scale).to("cuda")
+
+    pre = r(T, B, 4 * H, scale=0.5).bfloat16()
+    w_h = r(H, 4 * H, scale=1.0 / H ** 0.5).bfloat16()
+    h0 = (r(B, H).tanh()).bfloat16()
+    c0 = r(B, H, scale=0.5)
+    out_bf = ext.lstm_seq_fwd(pre, w_h, h0, c0, False)
+    ext.set_fp8_fwd(True)
+    try:
+        out_f8 = ext.lstm_seq_fwd(pre, w_h, h0, c0, False)
+    finally:
+        ext.set_fp8_fwd(False)
+    # h compounds quantization error over T steps; generous but bounded
+    err = (out_bf[0].float() - out_f8[0].float()).abs().max().item()
+    assert err < 0.05 + 0.02 * T, err
+    # and the fp8 path must still be close to the fp32 reference
+    h_r, *_ = functional.ref_seq_fwd(pre, w_h, h0, c0, False)
+    err_ref = (out_f8[0].float() - h_r.float()).abs().max().item()
+    assert err_ref < 0.05 + 0.02 * T, err_ref

@@ -1,25 +1,22 @@
 set -x
-python - <<'PY'
-import sys, torch; sys.path.insert(0, '.')
-from lstm_tensorspark_amd.ops import extension
-ext = extension.get_extension()
-g = torch.Generator(device="cuda").manual_seed(0)
-B, H = 16384, 1024
-h_prev = torch.randn(B, H, generator=g, device="cuda").bfloat16()
-w_h = (torch.randn(H, 4*H, generator=g, device="cuda") / H**0.5).bfloat16()
-pre = torch.randn(B, 4*H, generator=g, device="cuda").bfloat16()
-c_prev = torch.randn(B, H, generator=g, device="cuda")
-for v in (2, 32, 33, 40):
-    ext.set_fwd_variant(v)
-    for _ in range(5): ext.lstm_rec_fwd_step(h_prev, w_h, pre, c_prev, False)
-    torch.cuda.synchronize()
-    s, e = torch.cuda.Event(enable_timing=True), torch.cuda.Event(enable_timing=True)
-    s.record()
-    for _ in range(20): ext.lstm_rec_fwd_step(h_prev, w_h, pre, c_prev, False)
-    e.record(); torch.cuda.synchronize()
-    us = 1e3*s.elapsed_time(e)/20
-    print(f"B=16384 variant={v}: {us:.0f} us ({2*B*4*H*H/us/1e6:.0f} TF/s)")
-ext.set_fwd_variant(-1)
+python -m pytest tests/test_gpu_kernels.py -m gpu -q 2>&1 | tail -2
+timeout 400 python bench.py --gpus 1 --steps 10 --warmup 3 2>&1 | tail -1
+PRESET=mi355x-1gpu timeout 400 python - <<'PY'
+import sys, torch, time; sys.path.insert(0, '.')
+from lstm_tensorspark_amd.config import get_preset
+from lstm_tensorspark_amd.trainer import Trainer
+cfg = get_preset("mi355x-1gpu"); cfg.train.log_interval = 0; cfg.train.ckpt_dir = ""
+cfg.train.dtype = "fp8"
+t = Trainer(cfg, device="cuda:0")
+b = t.data.batch(0)
+for _ in range(3): t.train_step(b)
+torch.cuda.synchronize(); t0 = time.perf_counter()
+for _ in range(6): t.train_step(b)
+torch.cuda.synchronize()
+ms = (time.perf_counter()-t0)/6*1e3
+print(f"fp8 mode: {ms:.1f} ms/step ({cfg.train.batch_size/ms*1000:.0f} seq/s)")
 PY
-timeout 420 python train.py --preset mi355x-1gpu --set train.steps=300 --set train.log_interval=100 --set train.batch_size=8192 2>&1 | grep '"step"\|Error\|nan' | tail -4
-rocm-smi --showmeminfo vram 2>/dev/null | grep Used
+# fp8 convergence: copy task
+timeout 400 python train.py --preset mi355x-1gpu --set train.batch_size=4096 \
+  --set train.steps=100 --set train.log_interval=25 --set data.task=copy \
+  --set model.input_size=1024 --set train.lr=0.4 --set train.dtype=fp8 2>&1 | grep '"step"'
