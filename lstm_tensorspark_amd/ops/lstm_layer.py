@@ -96,6 +96,18 @@ def _seq_bwd(d_hseq, d_hT, d_cT, w_h, acts, c_seq, c0, reverse, use_hip):
     return functional.ref_seq_bwd(d_hseq, d_hT, d_cT, w_h, acts, c_seq, c0, reverse)
 
 
+def _fp8_addmm(bias, a, w):
+    """pre = bias + a @ w with both GEMM operands amax-quantized to OCP
+    e4m3 (experimental fp8 mode; ~2x the bf16 MFMA rate via _scaled_mm)."""
+    sa = a.abs().amax().float().clamp_min(1e-8) / 448.0
+    sw = w.abs().amax().float().clamp_min(1e-8) / 448.0
+    a8 = (a * (1.0 / sa)).to(torch.float8_e4m3fn)
+    w8 = (w * (1.0 / sw)).to(torch.float8_e4m3fn).t().contiguous().t()
+    return torch._scaled_mm(a8, w8, scale_a=sa, scale_b=sw,
+                            bias=bias.to(torch.bfloat16),
+                            out_dtype=torch.bfloat16)
+
+
 class LSTMLayerFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, h0, c0, w_x, w_h, b, reverse: bool):
@@ -104,8 +116,15 @@ class LSTMLayerFn(torch.autograd.Function):
         H = w_h.shape[0]
         ctx.set_materialize_grads(False)
         use_hip = extension.use_hip(x.device, x.dtype)
-        # [K1a] input projection for all timesteps in one library GEMM
-        pre = torch.addmm(b.to(x.dtype), x.reshape(T * B, X), w_x).view(T, B, 4 * H)
+        # [K1a] input projection for all timesteps in one library GEMM.
+        # In fp8 mode the operands are amax-quantized to e4m3 (x may be
+        # unbounded, so dynamic scales); the manual backward keeps bf16
+        # gradients regardless, so only this forward GEMM changes.
+        if (use_hip and extension.get_extension().get_fp8_fwd()
+                and x.dtype == torch.bfloat16):
+            pre = _fp8_addmm(b, x.reshape(T * B, X), w_x).view(T, B, 4 * H)
+        else:
+            pre = torch.addmm(b.to(x.dtype), x.reshape(T * B, X), w_x).view(T, B, 4 * H)
         h_all, c_seq, acts, hT, cT = _seq_fwd(pre, w_h, h0, c0, reverse, use_hip)
         ctx.save_for_backward(x, c0, w_x, w_h, h_all, c_seq, acts)
         ctx.reverse = reverse
