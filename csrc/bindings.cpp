@@ -34,6 +34,9 @@ void launch_lstm_pw_fwd(const void*, long, const void*, long, const float*,
                         long, void*, long, float*, long, void*, long, int,
                         int, hipStream_t);
 void launch_quant_fp8_fixed(const void*, void*, long, float, hipStream_t);
+void launch_quant_fp8_dyn(const void*, void*, long, const float*, hipStream_t);
+void launch_quant_fp8_dyn_T(const void*, void*, long, long, const float*,
+                            hipStream_t);
 void launch_xent_fwd(const void*, const long*, float*, float*, long, int,
                      hipStream_t);
 void launch_xent_bwd(const void*, const long*, const float*, void*,
@@ -301,6 +304,30 @@ std::vector<Tensor> lstm_seq_bwd(c10::optional<Tensor> dh_seq_opt,
 }
 
 // ---------------------------------------------------------------------
+// FP8 quantization helpers (experimental fp8 mode, python side: the big
+// dW/dx GEMMs in ops/lstm_layer.py).  mult = 448/amax, a DEVICE scalar.
+// ---------------------------------------------------------------------
+Tensor quant_fp8(Tensor t, Tensor mult) {
+    check_bf16_cuda(t, "t");
+    t = t.contiguous();
+    auto out = at::empty_like(t, t.options().dtype(at::kFloat8_e4m3fn));
+    launch_quant_fp8_dyn(t.data_ptr(), out.data_ptr(), t.numel(),
+                         (const float*)mult.data_ptr(), cur_stream());
+    return out;
+}
+
+Tensor quant_fp8_T(Tensor t, Tensor mult) {
+    check_bf16_cuda(t, "t");
+    TORCH_CHECK(t.dim() == 2, "quant_fp8_T wants (R,C)");
+    t = t.contiguous();
+    const long R = t.size(0), C = t.size(1);
+    auto out = at::empty({C, R}, t.options().dtype(at::kFloat8_e4m3fn));
+    launch_quant_fp8_dyn_T(t.data_ptr(), out.data_ptr(), R, C,
+                           (const float*)mult.data_ptr(), cur_stream());
+    return out;
+}
+
+// ---------------------------------------------------------------------
 // K3: fused log-softmax cross-entropy (LM head, vocab up to 50k+)
 // ---------------------------------------------------------------------
 std::vector<Tensor> xent_fwd(Tensor logits, Tensor targets) {
@@ -412,6 +439,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("lstm_rec_fwd_step", &lstm_rec_fwd_step,
           "single fused recurrent step (raw_debug for GEMM-mapping tests)");
     m.def("set_persistent_fwd", [](bool v) { g_persistent_enabled = v; });
+    m.def("quant_fp8", &quant_fp8, "bf16 -> e4m3, device-scalar multiplier");
+    m.def("quant_fp8_T", &quant_fp8_T, "bf16 (R,C) -> e4m3 (C,R) transposed");
     m.def("set_fp8_fwd", [](bool v) { g_fp8_fwd = v; },
           "experimental: fp8(e4m3) recurrent GEMM in the split forward");
     m.def("get_fp8_fwd", []() { return g_fp8_fwd; });

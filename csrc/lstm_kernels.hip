@@ -1022,6 +1022,98 @@ extern "C" void launch_quant_fp8_fixed(const void* in, void* out, long n,
                        scale);
 }
 
+// Dynamic-scale variant: the multiplier (448/amax) is a DEVICE scalar so
+// the amax -> quantize chain never syncs the host.
+__global__ __launch_bounds__(256) void k_quant_fp8_dyn(
+    const __hip_bfloat16* __restrict__ in, unsigned char* __restrict__ out,
+    long n, const float* __restrict__ mult)
+{
+    const float scale = *mult;
+    const long i0 = ((long)blockIdx.x * blockDim.x + threadIdx.x) * 8;
+    if (i0 + 8 <= n) {
+        const bf16x8 v = *(const bf16x8*)(in + i0);
+        unsigned long long pk = 0;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+            const __hip_fp8_e4m3 q((float)v[j] * scale);
+            pk |= (unsigned long long)q.__x << (8 * j);
+        }
+        *(unsigned long long*)(out + i0) = pk;
+    } else {
+        for (long i = i0; i < n; ++i) {
+            const __hip_fp8_e4m3 q(__bfloat162float(in[i]) * scale);
+            out[i] = q.__x;
+        }
+    }
+}
+
+extern "C" void launch_quant_fp8_dyn(const void* in, void* out, long n,
+                                     const float* mult, hipStream_t stream)
+{
+    dim3 block(256), grid((unsigned)((n / 8 + 255) / 256));
+    hipLaunchKernelGGL(k_quant_fp8_dyn, grid, block, 0, stream,
+                       (const __hip_bfloat16*)in, (unsigned char*)out, n,
+                       mult);
+}
+
+// Transposing quantizer: in (R, C) bf16 row-major -> out (C, R) fp8
+// row-major, via a 64x64 LDS tile (fp8 in LDS, +8 pad kills write/read
+// bank conflicts).  Feeds _scaled_mm's layout rules for the dW GEMMs
+// (mat1 row-major (C, R), built from a row-major (R, C) gradient/input).
+__global__ __launch_bounds__(256) void k_quant_fp8_dyn_T(
+    const __hip_bfloat16* __restrict__ in, unsigned char* __restrict__ out,
+    long R, long C, const float* __restrict__ mult)
+{
+    const float scale = *mult;
+    __shared__ unsigned char tile[64][72];
+    const long r0 = (long)blockIdx.y * 64;
+    const long c0 = (long)blockIdx.x * 64;
+    const int tid = threadIdx.x;
+    // load+quantize: 8 rows per iteration, 8 cols per thread
+    const int lr = tid >> 3, lc = (tid & 7) * 8;   // 32 rows x 64 cols / iter
+#pragma unroll
+    for (int it = 0; it < 2; ++it) {
+        const long r = r0 + it * 32 + lr;
+        if (r < R && c0 + lc < C) {
+            const bf16x8 v = *(const bf16x8*)(in + r * C + c0 + lc);
+#pragma unroll
+            for (int j = 0; j < 8; ++j) {
+                const __hip_fp8_e4m3 q((float)v[j] * scale);
+                tile[lc + j][it * 32 + lr] = q.__x;
+            }
+        }
+    }
+    __syncthreads();
+    // store transposed: out row = column c, 8 consecutive R values/thread
+    const int oc = tid >> 3, orr = (tid & 7) * 8;  // 32 out-rows x 64 / iter
+#pragma unroll
+    for (int it = 0; it < 2; ++it) {
+        const long c = c0 + it * 32 + oc;
+        const long r = r0 + orr;
+        if (c < C && r + 8 <= R + (R & 7)) {
+            unsigned long long pk = 0;
+#pragma unroll
+            for (int j = 0; j < 8; ++j)
+                pk |= (unsigned long long)tile[it * 32 + oc][orr + j] << (8 * j);
+            if (r + 8 <= R)
+                *(unsigned long long*)(out + c * R + r) = pk;
+            else
+                for (int j = 0; j < 8 && r + j < R; ++j)
+                    out[c * R + r + j] = tile[it * 32 + oc][orr + j];
+        }
+    }
+}
+
+extern "C" void launch_quant_fp8_dyn_T(const void* in, void* out, long R,
+                                       long C, const float* mult,
+                                       hipStream_t stream)
+{
+    dim3 block(256), grid((unsigned)((C + 63) / 64), (unsigned)((R + 63) / 64));
+    hipLaunchKernelGGL(k_quant_fp8_dyn_T, grid, block, 0, stream,
+                       (const __hip_bfloat16*)in, (unsigned char*)out, R, C,
+                       mult);
+}
+
 // ---------------------------------------------------------------------------
 // K4: fused pointwise backward for one timestep (elementwise over B x H).
 // Produces raw-gate grads dgates_t (bf16) and the fp32 cell-grad carry

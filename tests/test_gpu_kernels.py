@@ -388,3 +388,45 @@ def test_fp8_recurrent_gemm_vs_bf16():
     h_r, *_ = functional.ref_seq_fwd(pre, w_h, h0, c0, False)
     err_ref = (out_f8[0].float() - h_r.float()).abs().max().item()
     assert err_ref < 0.05 + 0.02 * T, err_ref
+
+
+def test_fp8_layer_grads_track_bf16():
+    """fp8 dW/dx GEMMs: gradients track the bf16 path within quantization
+    error (relative to grad magnitude)."""
+    from lstm_tensorspark_amd.ops.extension import get_extension
+    from lstm_tensorspark_amd.ops.lstm_layer import enable_fp8_wgrad, lstm_layer
+
+    ext = get_extension()
+    T, B, X, H = 4, 2048, 1024, 1024   # B*H >= 2M -> split+fp8 eligible
+    g = torch.Generator(device="cpu").manual_seed(21)
+
+    def r(*s, scale=1.0):
+        return (torch.randn(*s, generator=g) * scale).to("cuda")
+
+    x32 = r(T, B, X, scale=0.5)
+    wx32 = r(X, 4 * H, scale=1.0 / X ** 0.5)
+    wh32 = r(H, 4 * H, scale=1.0 / H ** 0.5)
+    b32 = r(4 * H, scale=0.1)
+    gout = r(T, B, H, scale=0.1).bfloat16()
+    h0 = torch.zeros(B, H, device="cuda").bfloat16()
+    c0 = torch.zeros(B, H, device="cuda")
+
+    grads = {}
+    for fp8 in (False, True):
+        ext.set_fp8_fwd(fp8)
+        enable_fp8_wgrad(fp8)
+        try:
+            xs = x32.bfloat16().requires_grad_(True)
+            wx = wx32.bfloat16().requires_grad_(True)
+            wh = wh32.bfloat16().requires_grad_(True)
+            bb = b32.bfloat16().requires_grad_(True)
+            hs, _, _ = lstm_layer(xs, h0, c0, wx, wh, bb)
+            hs.backward(gout)
+            grads[fp8] = [t.grad.float() for t in (xs, wx, wh, bb)]
+        finally:
+            ext.set_fp8_fwd(False)
+            enable_fp8_wgrad(False)
+    for name, a, b in zip("dx dwx dwh db".split(), grads[False], grads[True]):
+        denom = a.abs().max().item() + 1e-6
+        rel = (a - b).abs().max().item() / denom
+        assert rel < 0.25, f"{name} rel err {rel}"

@@ -17,6 +17,6 @@ for dt in ("bf16", "fp8"):
     print(f"{dt}: {ms:.1f} ms/step ({cfg.train.batch_size/ms*1000:.0f} seq/s)")
     del t; torch.cuda.empty_cache()
 PY
-timeout 400 python train.py --preset mi355x-1gpu --set train.batch_size=4096 \
+timeout 300 python train.py --preset mi355x-1gpu --set train.batch_size=4096 \
   --set train.steps=100 --set train.log_interval=50 --set data.task=copy \
   --set model.input_size=1024 --set train.lr=0.4 --set train.dtype=fp8 2>&1 | grep '"step"'
