@@ -46,6 +46,34 @@ def _rev_stream():
     return _REV_STREAM
 
 
+class _Fp8HeadFn(torch.autograd.Function):
+    """LM-head projection with the forward GEMM in e4m3 (feats are
+    tanh-bounded -> FIXED scale 448; W amax-scaled).  Backward stays bf16:
+    dW = feats^T·dlogits and dfeats = dlogits·W^T at full precision, so
+    only the forward logits carry fp8 quantization error."""
+
+    @staticmethod
+    def forward(ctx, feats, w, b):
+        sw = w.abs().amax().float().clamp_min(1e-8) / 448.0
+        f8 = (feats * 448.0).to(torch.float8_e4m3fn)
+        w8 = (w * (1.0 / sw)).to(torch.float8_e4m3fn).t().contiguous().t()
+        sa = torch.full((), 1.0 / 448.0, device=feats.device)
+        out = torch._scaled_mm(f8, w8, scale_a=sa, scale_b=sw,
+                               bias=b.to(torch.bfloat16),
+                               out_dtype=torch.bfloat16)
+        ctx.save_for_backward(feats, w)
+        return out
+
+    @staticmethod
+    def backward(ctx, dlogits):
+        feats, w = ctx.saved_tensors
+        dl = dlogits.contiguous()
+        dfeats = dl @ w.t()
+        dw = feats.t() @ dl
+        db = dl.sum(dim=0, dtype=torch.float32).to(w.dtype)
+        return dfeats, dw, db
+
+
 class LSTMModel(nn.Module):
     """Multi-layer (bi)LSTM + head. Compute dtype follows parameter dtype;
     cell state is always fp32."""
@@ -186,8 +214,21 @@ class LSTMModel(nn.Module):
         if cfg.head == "classifier":
             return torch.addmm(b, feats[-1], w), new_state
         T, B, F_ = feats.shape
-        logits = torch.addmm(b, feats.reshape(T * B, F_), w).view(T, B, -1)
+        flat = feats.reshape(T * B, F_)
+        if (cfg.head == "lm" and feats.is_cuda
+                and feats.dtype == torch.bfloat16 and self._fp8_head()
+                and (T * B) % 16 == 0 and F_ % 16 == 0
+                and w.shape[1] % 16 == 0):
+            logits = _Fp8HeadFn.apply(flat, w, b).view(T, B, -1)
+        else:
+            logits = torch.addmm(b, flat, w).view(T, B, -1)
         return logits, new_state
+
+    @staticmethod
+    def _fp8_head() -> bool:
+        from ..ops import extension
+        return (extension.extension_available()
+                and extension.get_extension().get_fp8_fwd())
 
     def loss(self, output: torch.Tensor, y: torch.Tensor) -> torch.Tensor:
         head = self.cfg.head
