@@ -4,11 +4,14 @@ This is the MI355X replacement for the reference's per-timestep TF graph ops
 (SURVEY.md §2.3, rows F1-F5 / B1-B2): instead of ~10 graph nodes per
 timestep replayed by Session.run, the whole-sequence forward is
 
-  1. ONE library GEMM (hipBLASLt via torch.matmul) for the input projection
-     of ALL timesteps:  pre = x.view(T*B, X) @ W_x + b          [K1a]
-  2. a C++-driven time loop of ONE fused hand-written CDNA4 kernel per step:
-     MFMA recurrent GEMM h_{t-1} @ W_h + gate activations + cell-state
-     update + h/c/act stores, all fused                          [K1b+K2]
+  1. ONE library GEMM (hipBLASLt via torch.addmm, or _scaled_mm in fp8
+     mode) for the input projection of ALL timesteps:
+     pre = x.view(T*B, X) @ W_x + b                              [K1a]
+  2. a C++-driven time loop, one of two per-step paths chosen by shape
+     (csrc/bindings.cpp): the SPLIT path (hipBLASLt recurrent GEMM +
+     k_lstm_pw_fwd vectorized cell update; default at B*H >= 2M) or the
+     FUSED hand-written CDNA4 MFMA kernel (recurrent GEMM + gate
+     activations + cell update in one launch; default below)     [K1b+K2]
 
 and the backward is
   3. a C++ time loop of a fused pointwise-backward kernel + recurrent
@@ -40,11 +43,11 @@ from .functional import h_prev_view, h_seq_view
 # ---------------------------------------------------------------------------
 # Side-stream weight-gradient GEMMs.
 #
-# The recurrent backward is a serial per-timestep chain (pw_bwd -> dh GEMM)
-# that underutilizes the chip; the big dW_h/dW_x/db GEMMs have no consumer
-# until the optimizer (or the DDP all-reduce).  Launching them on a side
-# HIP stream lets them overlap the NEXT layer's sequential backward loop,
-# taking ~all of the dW time off the critical path.
+# The big dW_h/dW_x/db GEMMs have no consumer until the optimizer (or the
+# DDP all-reduce), so they CAN run on a side HIP stream overlapping the
+# next layer's backward.  Measured NEUTRAL at HBM-sized batches (the step
+# is bound by the combined HBM+MFMA resource envelope, not dependencies) --
+# default OFF (dist.side_dw), kept for small-model regimes.
 #
 # Correctness protocol: the side stream waits on an event recorded after
 # dgates is ready; every consumer of .grad (optimizer step, grad-clip,
