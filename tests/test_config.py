@@ -58,3 +58,33 @@ def test_yaml_load(tmp_path):
 
 def test_json_serializable():
     json.loads(Config().to_json())
+
+
+def test_example_yamls_load():
+    import glob
+    import os
+
+    from lstm_tensorspark_amd.config import Config
+
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    paths = sorted(glob.glob(os.path.join(root, "examples", "*.yaml")))
+    assert len(paths) >= 2
+    for p in paths:
+        cfg = Config.from_yaml(p)
+        assert cfg.train.batch_size > 0 and cfg.model.hidden_size > 0
+
+
+def test_fp8_dtype_plumbing_cpu_fallback():
+    import torch
+
+    from lstm_tensorspark_amd.config import Config, ModelConfig, TrainConfig
+    from lstm_tensorspark_amd.trainer import Trainer, compute_dtype
+
+    cfg = Config(model=ModelConfig(input_size=16, hidden_size=16),
+                 train=TrainConfig(batch_size=2, seq_len=4, dtype="fp8",
+                                   log_interval=0))
+    # on CPU, fp8 mode degrades to the fp32 reference path cleanly
+    assert compute_dtype(cfg, torch.device("cpu")) == torch.float32
+    t = Trainer(cfg, device="cpu")
+    info = t.train_step(t.data.batch(0))
+    assert info["loss"] == info["loss"]
