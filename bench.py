@@ -36,6 +36,8 @@ def parse_args():
     p.add_argument("--preset", default="mi355x-1gpu")
     p.add_argument("--batch", type=int, default=0,
                    help="override per-rank batch size")
+    p.add_argument("--dtype", default="",
+                   help="override compute dtype (bf16 | fp8)")
     p.add_argument("--cpu", action="store_true",
                    help="allow CPU run (small shapes; for plumbing checks)")
     return p.parse_args()
@@ -51,6 +53,8 @@ def main():
     cfg.train.ckpt_dir = ""
     if args.batch:
         cfg.train.batch_size = args.batch
+    if args.dtype:
+        cfg.train.dtype = args.dtype
     on_gpu = ctx.device.type == "cuda"
     if not on_gpu:
         if not args.cpu:

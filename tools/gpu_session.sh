@@ -1,7 +1,14 @@
 set -x
 mkdir -p gpurun_out
-for dt in bf16 fp8; do
-  timeout 400 python train.py --preset lm-50k --set train.batch_size=128 \
-    --set train.steps=150 --set train.log_interval=30 --set data.task=lm_assoc \
-    --set train.lr=0.0003 --set train.dtype=$dt 2>&1 | grep '"step"' | sed "s/^/[$dt] /"
-done | tee gpurun_out/curve_lm_fp8_ab.log
+R=$GRAFT_REPO_ROOT
+python - <<'PY'
+import __graft_entry__
+__graft_entry__.smoke()
+PY
+python -m pytest tests -m gpu -q 2>&1 | tail -1 | tee gpurun_out/final_pytest.log
+timeout 400 python bench.py --gpus 1 --steps 15 --warmup 3 2>&1 | tail -1 | tee gpurun_out/final_bench.json
+timeout 400 python bench.py --gpus 1 --steps 6 --warmup 2 --preset bidir-4l 2>&1 | tail -1 | tee -a gpurun_out/final_bench.json
+timeout 400 python bench.py --gpus 1 --steps 6 --warmup 2 --preset lm-50k 2>&1 | tail -1 | tee -a gpurun_out/final_bench.json
+cd /tmp && export TMPDIR=/tmp
+timeout 420 rocprofv3 --kernel-trace --stats -d $R/gpurun_out/proff -- python $R/bench.py --steps 5 --warmup 2 > $R/gpurun_out/final_prof.log 2>&1
+grep -o '"value": [0-9.]*' $R/gpurun_out/final_prof.log | head -1
