@@ -132,7 +132,7 @@ def main():
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,  # reference has no published numbers
-            "dtype": "bf16" if (on_gpu and cfg.train.dtype == "bf16") else "fp32",
+            "dtype": cfg.train.dtype if on_gpu else "fp32",
             "data": "synthetic",
             "config": {
                 "model": f"lstm-{cfg.model.num_layers}l-h{cfg.model.hidden_size}",
