@@ -266,23 +266,35 @@ class LSTMModel(nn.Module):
     @torch.no_grad()
     def generate(self, prompt: torch.Tensor, steps: int,
                  temperature: float = 0.0,
-                 generator: "Optional[torch.Generator]" = None) -> torch.Tensor:
+                 generator: "Optional[torch.Generator]" = None,
+                 use_graph: Optional[bool] = None) -> torch.Tensor:
         """Autoregressive decode for the LM head (the serving path).
 
         prompt: (Tp, B) long tokens (time-major).  Returns (Tp+steps, B):
         the prompt followed by `steps` generated tokens.  temperature 0 =
         greedy argmax; >0 = softmax sampling.  The recurrent (h, c) state
         is carried across steps, so each decode step is one T=1 pass.
+
+        use_graph (default: auto = greedy + CUDA + steps >= 16): capture
+        ONE decode step as a hipGraph whose body copies its outputs back
+        into its own static inputs and appends the token to a device
+        buffer, then replay it per token -- removes ALL per-token launch
+        and Python overhead (small-B decode is launch-latency bound).
         """
         assert self.cfg.head == "lm", "generate() requires the LM head"
         assert not self.cfg.bidirectional, "decode is causal: fwd-only"
         assert prompt.dim() == 2 and prompt.dtype == torch.long
         was_training = self.training
         self.eval()
+        if use_graph is None:
+            use_graph = (temperature <= 0 and prompt.is_cuda and steps >= 16)
         try:
             out, state = self.forward(prompt)          # (Tp, B, V)
-            toks = [prompt]
             cur = self._sample(out[-1], temperature, generator)
+            if use_graph:
+                return torch.cat(
+                    [prompt, self._graph_decode(cur, state, steps)], dim=0)
+            toks = [prompt]
             for _ in range(steps - 1):
                 toks.append(cur.unsqueeze(0))
                 out, state = self.forward(cur.unsqueeze(0), state)
@@ -292,6 +304,57 @@ class LSTMModel(nn.Module):
         finally:
             if was_training:
                 self.train()
+
+    _DECODE_CAP = 8192  # tokens per replay series (out ring capacity)
+
+    def _graph_decode(self, cur, state, steps: int) -> torch.Tensor:
+        """Greedy decode via a replayed hipGraph (see generate()).  The
+        captured graph + its static buffers are CACHED on the model keyed
+        by batch size, so steady-state serving is replay-only."""
+        B = cur.shape[0]
+        dev = cur.device
+        cache = getattr(self, "_dec_cache", None)
+        if cache is None or cache["B"] != B:
+            g_cur = cur.clone()
+            g_state = [(h.clone(), c.clone()) for h, c in state]
+            out_buf = torch.empty(self._DECODE_CAP, B, dtype=torch.long,
+                                  device=dev)
+            idx = torch.zeros((), dtype=torch.long, device=dev)
+
+            def body():
+                out_buf.index_copy_(0, idx.unsqueeze(0), g_cur.unsqueeze(0))
+                idx.add_(1)
+                logits, new_state = self.forward(g_cur.unsqueeze(0), g_state)
+                nxt = logits[-1].argmax(dim=-1)
+                g_cur.copy_(nxt)
+                for (h, c), (nh, nc) in zip(g_state, new_state):
+                    h.copy_(nh)
+                    c.copy_(nc)
+
+            for _ in range(2):        # warmup (workspaces at stable addrs)
+                body()
+            torch.cuda.synchronize()
+            graph = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(graph):
+                body()
+            cache = {"B": B, "graph": graph, "g_cur": g_cur,
+                     "g_state": g_state, "out_buf": out_buf, "idx": idx}
+            self._dec_cache = cache
+        # (re)prime the static buffers from this call's live state
+        cache["g_cur"].copy_(cur)
+        for (h, c), (oh, oc) in zip(cache["g_state"], state):
+            h.copy_(oh)
+            c.copy_(oc)
+        outs = []
+        done = 0
+        while done < steps:
+            n = min(steps - done, self._DECODE_CAP)
+            cache["idx"].zero_()
+            for _ in range(n):
+                cache["graph"].replay()
+            outs.append(cache["out_buf"][:n].clone())
+            done += n
+        return outs[0] if len(outs) == 1 else torch.cat(outs, dim=0)
 
     @staticmethod
     def _sample(logits, temperature, generator):

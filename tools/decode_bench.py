@@ -27,7 +27,7 @@ def main():
                       num_layers=args.layers, head="lm", vocab_size=args.vocab)
     m = LSTMModel(cfg).to("cuda").bfloat16()
     prompt = torch.randint(0, args.vocab, (args.prompt, args.B), device="cuda")
-    m.generate(prompt, steps=8)   # warmup
+    m.generate(prompt, steps=32)  # warmup (captures the decode graph)
     torch.cuda.synchronize()
     t0 = time.perf_counter()
     out = m.generate(prompt, steps=args.steps)

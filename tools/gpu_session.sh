@@ -1,6 +1,5 @@
 set -x
-mkdir -p gpurun_out
-for b in 1 16 64 256 512; do
-  timeout 200 python tools/decode_bench.py --B $b --steps 64 2>&1 | tail -1
-done | tee gpurun_out/decode_sweep.log
-timeout 400 python bench.py --gpus 1 --steps 6 --warmup 2 --preset bidir-4l --dtype fp8 2>&1 | tail -1
+python -m pytest tests/test_gpu_train.py -m gpu -q 2>&1 | tail -1
+for b in 1 16 256; do
+  timeout 200 python tools/decode_bench.py --B $b --steps 256 2>&1 | tail -1
+done
