@@ -35,6 +35,8 @@ void launch_lstm_pw_fwd(const void*, long, const void*, long, const float*,
                         int, hipStream_t);
 void launch_quant_fp8_fixed(const void*, void*, long, float, hipStream_t);
 void launch_quant_fp8_dyn(const void*, void*, long, const float*, hipStream_t);
+int launch_gemv_bf16(const void*, const void*, const float*, void*, float*,
+                     int, long, int, hipStream_t);
 void launch_quant_fp8_dyn_T(const void*, void*, long, long, const float*,
                             hipStream_t);
 void launch_xent_fwd(const void*, const long*, float*, float*, long, int,
@@ -304,6 +306,40 @@ std::vector<Tensor> lstm_seq_bwd(c10::optional<Tensor> dh_seq_opt,
 }
 
 // ---------------------------------------------------------------------
+// Serving GEMV: out (B,N) = x (B,K) @ A_t(N,K)^T + bias, the small-B
+// decode fast path (one wave per output row, full-rate weight streaming).
+// A_t is the PRE-TRANSPOSED (serving-layout) weight.  Returns the result
+// in bf16; raises if the shape is outside the fast path's envelope --
+// the python caller guards with decode_gemv_ok().
+// ---------------------------------------------------------------------
+Tensor decode_gemv(Tensor A_t, Tensor x, c10::optional<Tensor> bias) {
+    check_bf16_cuda(A_t, "A_t");
+    check_bf16_cuda(x, "x");
+    TORCH_CHECK(A_t.dim() == 2 && x.dim() == 2 && A_t.size(1) == x.size(1),
+                "decode_gemv: A_t (N,K), x (B,K)");
+    A_t = A_t.contiguous();
+    x = x.contiguous();
+    const long N = A_t.size(0), K = A_t.size(1), B = x.size(0);
+    const float* bp = nullptr;
+    Tensor bc;
+    if (bias.has_value()) {
+        bc = bias->to(at::kFloat).contiguous();
+        bp = (const float*)bc.data_ptr();
+    }
+    auto out = at::empty({B, N}, x.options());
+    int rc = launch_gemv_bf16(A_t.data_ptr(), x.data_ptr(), bp,
+                              out.data_ptr(), nullptr, (int)B, N, (int)K,
+                              cur_stream());
+    TORCH_CHECK(rc == 0, "decode_gemv: shape outside fast-path envelope "
+                "(B<=16, B*K*2 <= 160 KiB LDS, K % 512 == 0)");
+    return out;
+}
+
+bool decode_gemv_ok(long B, long K) {
+    return B <= 16 && B * K * 2 <= 160 * 1024 && K % 512 == 0;
+}
+
+// ---------------------------------------------------------------------
 // FP8 quantization helpers (experimental fp8 mode, python side: the big
 // dW/dx GEMMs in ops/lstm_layer.py).  mult = 448/amax, a DEVICE scalar.
 // ---------------------------------------------------------------------
@@ -439,6 +475,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("lstm_rec_fwd_step", &lstm_rec_fwd_step,
           "single fused recurrent step (raw_debug for GEMM-mapping tests)");
     m.def("set_persistent_fwd", [](bool v) { g_persistent_enabled = v; });
+    m.def("decode_gemv", &decode_gemv, "small-B serving GEMV",
+          py::arg("A_t"), py::arg("x"), py::arg("bias") = py::none());
+    m.def("decode_gemv_ok", &decode_gemv_ok);
     m.def("quant_fp8", &quant_fp8, "bf16 -> e4m3, device-scalar multiplier");
     m.def("quant_fp8_T", &quant_fp8_T, "bf16 (R,C) -> e4m3 (C,R) transposed");
     m.def("set_fp8_fwd", [](bool v) { g_fp8_fwd = v; },

@@ -987,6 +987,80 @@ extern "C" void launch_lstm_pw_fwd(
 }
 
 // ---------------------------------------------------------------------------
+// Serving GEMV: out (B,N) = A (N,K row-major) @ x^T (B,K) + bias, for the
+// SMALL-B decode path where hipBLASLt's GEMV leaves ~7x bandwidth on the
+// table (per-token decode is pure weight streaming).  One WAVE per output
+// row n: 64 lanes stride bf16x8 chunks over K, B accumulators per lane
+// (x is LDS-resident), cross-lane shfl reduce.  A is the PRE-TRANSPOSED
+// weight (serving layout, built once per model by the decode cache).
+// ---------------------------------------------------------------------------
+template <int BMAX>
+__global__ __launch_bounds__(256) void k_gemv_bf16(
+    const __hip_bfloat16* __restrict__ A,    // (N, K) row-major
+    const __hip_bfloat16* __restrict__ x,    // (B, K) row-major
+    const float* __restrict__ bias,          // (N,) fp32 or null
+    __hip_bfloat16* __restrict__ out,        // (B, N)
+    float* __restrict__ out_f32,             // (B, N) or null (either/or)
+    int B, long N, int K)
+{
+    extern __shared__ __attribute__((aligned(16))) char smem[];
+    __hip_bfloat16* xs = (__hip_bfloat16*)smem;      // [B][K]
+    const int tid = threadIdx.x;
+    for (int i = tid * 8; i < B * K; i += 256 * 8) {
+        *(bf16x8*)(xs + i) = *(const bf16x8*)(x + i);
+    }
+    __syncthreads();
+    const int wave = tid >> 6, lane = tid & 63;
+    const long n = (long)blockIdx.x * 4 + wave;
+    if (n >= N) return;
+    const __hip_bfloat16* ar = A + n * K;
+    float acc[BMAX];
+#pragma unroll
+    for (int b = 0; b < BMAX; ++b) acc[b] = 0.f;
+    for (int k0 = lane * 8; k0 < K; k0 += 64 * 8) {
+        const bf16x8 av = *(const bf16x8*)(ar + k0);
+#pragma unroll
+        for (int b = 0; b < BMAX; ++b) {
+            if (b < B) {
+                const bf16x8 xv = *(const bf16x8*)(xs + b * K + k0);
+#pragma unroll
+                for (int j = 0; j < 8; ++j)
+                    acc[b] += (float)av[j] * (float)xv[j];
+            }
+        }
+    }
+#pragma unroll
+    for (int b = 0; b < BMAX; ++b)
+#pragma unroll
+        for (int off = 32; off > 0; off >>= 1)
+            acc[b] += __shfl_down(acc[b], off, 64);
+    if (lane == 0) {
+        const float bv = bias ? bias[n] : 0.f;
+#pragma unroll
+        for (int b = 0; b < BMAX; ++b)
+            if (b < B) {
+                const float v = acc[b] + bv;
+                if (out) out[(long)b * N + n] = __float2bfloat16(v);
+                else out_f32[(long)b * N + n] = v;
+            }
+    }
+}
+
+extern "C" void launch_gemv_bf16(const void* A, const void* x,
+                                 const float* bias, void* out, float* out_f32,
+                                 int B, long N, int K, hipStream_t stream)
+{
+    dim3 block(256), grid((unsigned)((N + 3) / 4));
+    const size_t smem = (size_t)B * K * 2;
+#define GEMV_LAUNCH(BM)                                                           hipLaunchKernelGGL(k_gemv_bf16<BM>, grid, block, smem, stream,                    (const __hip_bfloat16*)A, (const __hip_bfloat16*)x, bias,                     (__hip_bfloat16*)out, out_f32, B, N, K)
+    if (B <= 1) GEMV_LAUNCH(1);
+    else if (B <= 4) GEMV_LAUNCH(4);
+    else if (B <= 8) GEMV_LAUNCH(8);
+    else GEMV_LAUNCH(16);
+#undef GEMV_LAUNCH
+}
+
+// ---------------------------------------------------------------------------
 // FP8 quantization for the split-path recurrent GEMM: h is tanh-bounded to
 // (-1,1), so a FIXED scale of 448 (e4m3 max) needs no amax pass; the GEMM
 // dequantizes with scale_a = 1/448.  OCP e4m3 (gfx950), not fnuz.

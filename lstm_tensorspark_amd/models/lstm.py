@@ -267,7 +267,8 @@ class LSTMModel(nn.Module):
     def generate(self, prompt: torch.Tensor, steps: int,
                  temperature: float = 0.0,
                  generator: "Optional[torch.Generator]" = None,
-                 use_graph: Optional[bool] = None) -> torch.Tensor:
+                 use_graph: Optional[bool] = None,
+                 use_gemv: Optional[bool] = None) -> torch.Tensor:
         """Autoregressive decode for the LM head (the serving path).
 
         prompt: (Tp, B) long tokens (time-major).  Returns (Tp+steps, B):
@@ -293,7 +294,8 @@ class LSTMModel(nn.Module):
             cur = self._sample(out[-1], temperature, generator)
             if use_graph:
                 return torch.cat(
-                    [prompt, self._graph_decode(cur, state, steps)], dim=0)
+                    [prompt, self._graph_decode(cur, state, steps, use_gemv)],
+                    dim=0)
             toks = [prompt]
             for _ in range(steps - 1):
                 toks.append(cur.unsqueeze(0))
@@ -307,14 +309,75 @@ class LSTMModel(nn.Module):
 
     _DECODE_CAP = 8192  # tokens per replay series (out ring capacity)
 
-    def _graph_decode(self, cur, state, steps: int) -> torch.Tensor:
+    def _gemv_cache(self):
+        """Serving weight layout: pre-transposed [W_x;W_h]^T per layer and
+        head W^T, built ONCE (doubles head-weight memory -- a serving
+        cache, not a training cost)."""
+        gc = getattr(self, "_gemv_weights", None)
+        if gc is not None:
+            return gc
+        cfg = self.cfg
+        layers = []
+        for l in range(cfg.num_layers):
+            w_x = self.weight(_pname(l, False, "W_x"))
+            w_h = self.weight(_pname(l, False, "W_h"))
+            b = self.weight(_pname(l, False, "b"))
+            wcat_t = torch.cat([w_x.detach(), w_h.detach()], dim=0).t().contiguous()
+            layers.append((wcat_t, b.detach().float()))
+        head_t = self.weight("head/W").detach().t().contiguous()
+        head_b = self.weight("head/b").detach().float()
+        gc = {"layers": layers, "head_t": head_t, "head_b": head_b}
+        self._gemv_weights = gc
+        return gc
+
+    def _gemv_step(self, cur, g_state):
+        """One greedy decode step through the streaming-GEMV kernels
+        (updates g_state in place, returns the next token)."""
+        from ..ops.extension import get_extension
+        ext = get_extension()
+        gc = self._gemv_cache()
+        H = self.cfg.hidden_size
+        x = F.embedding(cur, self.weight("embed/W"))
+        for l, (wcat_t, b) in enumerate(gc["layers"]):
+            h, c = g_state[l]
+            xcat = torch.cat([x, h], dim=1)
+            raw = ext.decode_gemv(wcat_t, xcat, b).float()
+            i = torch.sigmoid(raw[:, :H])
+            f = torch.sigmoid(raw[:, H:2 * H])
+            g = torch.tanh(raw[:, 2 * H:3 * H])
+            o = torch.sigmoid(raw[:, 3 * H:])
+            cn = f * c + i * g
+            hn = (o * torch.tanh(cn)).to(x.dtype)
+            h.copy_(hn)
+            c.copy_(cn)
+            x = hn
+        logits = ext.decode_gemv(gc["head_t"], x, gc["head_b"])
+        return logits.argmax(dim=-1)
+
+    def _gemv_eligible(self, B: int) -> bool:
+        from ..ops import extension
+        if not extension.extension_available() or self.cfg.bidirectional:
+            return False
+        ext = extension.get_extension()
+        cfg = self.cfg
+        K0 = cfg.input_size + cfg.hidden_size
+        return bool(ext.decode_gemv_ok(B, K0)
+                    and ext.decode_gemv_ok(B, cfg.hidden_size
+                                           * cfg.num_directions))
+
+    def _graph_decode(self, cur, state, steps: int,
+                      use_gemv: Optional[bool] = None) -> torch.Tensor:
         """Greedy decode via a replayed hipGraph (see generate()).  The
         captured graph + its static buffers are CACHED on the model keyed
-        by batch size, so steady-state serving is replay-only."""
+        by (batch size, gemv), so steady-state serving is replay-only.
+        use_gemv (auto: on when eligible) streams the weights through the
+        hand-written serving GEMV instead of library mm."""
         B = cur.shape[0]
         dev = cur.device
+        if use_gemv is None:
+            use_gemv = self._gemv_eligible(B)
         cache = getattr(self, "_dec_cache", None)
-        if cache is None or cache["B"] != B:
+        if cache is None or cache["B"] != B or cache["gemv"] != use_gemv:
             g_cur = cur.clone()
             g_state = [(h.clone(), c.clone()) for h, c in state]
             out_buf = torch.empty(self._DECODE_CAP, B, dtype=torch.long,
@@ -324,12 +387,15 @@ class LSTMModel(nn.Module):
             def body():
                 out_buf.index_copy_(0, idx.unsqueeze(0), g_cur.unsqueeze(0))
                 idx.add_(1)
-                logits, new_state = self.forward(g_cur.unsqueeze(0), g_state)
-                nxt = logits[-1].argmax(dim=-1)
+                if use_gemv:
+                    nxt = self._gemv_step(g_cur, g_state)
+                else:
+                    logits, new_state = self.forward(g_cur.unsqueeze(0), g_state)
+                    nxt = logits[-1].argmax(dim=-1)
+                    for (h, c), (nh, nc) in zip(g_state, new_state):
+                        h.copy_(nh)
+                        c.copy_(nc)
                 g_cur.copy_(nxt)
-                for (h, c), (nh, nc) in zip(g_state, new_state):
-                    h.copy_(nh)
-                    c.copy_(nc)
 
             for _ in range(2):        # warmup (workspaces at stable addrs)
                 body()
@@ -337,7 +403,7 @@ class LSTMModel(nn.Module):
             graph = torch.cuda.CUDAGraph()
             with torch.cuda.graph(graph):
                 body()
-            cache = {"B": B, "graph": graph, "g_cur": g_cur,
+            cache = {"B": B, "gemv": use_gemv, "graph": graph, "g_cur": g_cur,
                      "g_state": g_state, "out_buf": out_buf, "idx": idx}
             self._dec_cache = cache
         # (re)prime the static buffers from this call's live state

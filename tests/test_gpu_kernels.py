@@ -430,3 +430,17 @@ def test_fp8_layer_grads_track_bf16():
         denom = a.abs().max().item() + 1e-6
         rel = (a - b).abs().max().item() / denom
         assert rel < 0.25, f"{name} rel err {rel}"
+
+
+# ---------------------------------------------------------- serving GEMV
+@pytest.mark.parametrize("B,N,K", [(1, 4096, 1024), (4, 512, 2048),
+                                   (16, 50000, 4096)])
+def test_decode_gemv_vs_mm(ext, B, N, K):
+    g = torch.Generator(device="cpu").manual_seed(31)
+    A_t = (torch.randn(N, K, generator=g) / K ** 0.5).to("cuda").bfloat16()
+    x = torch.randn(B, K, generator=g).to("cuda").bfloat16()
+    bias = torch.randn(N, generator=g).to("cuda").float()
+    out = ext.decode_gemv(A_t, x, bias)
+    ref = (x.float() @ A_t.float().t()) + bias
+    err = (out.float() - ref).abs().max().item()
+    assert err < 0.03 * max(1.0, ref.abs().max().item()), err

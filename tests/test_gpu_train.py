@@ -112,10 +112,15 @@ def test_gpu_lm_generate():
                       head="lm", vocab_size=512)
     m = LSTMModel(cfg).to("cuda").bfloat16()
     prompt = torch.randint(0, 512, (8, 4), device="cuda")
-    out = m.generate(prompt, steps=16)            # auto: graph-captured
+    out = m.generate(prompt, steps=16, use_gemv=False)  # graph, library mm
     assert out.shape == (24, 4)
     assert torch.equal(out[:8], prompt)
     assert (out < 512).all() and (out >= 0).all()
     # graphed decode must match the eager loop token for token (greedy)
     out_eager = m.generate(prompt, steps=16, use_graph=False)
     assert torch.equal(out, out_eager)
+    # gemv serving path: same API, valid tokens, deterministic
+    g1 = m.generate(prompt, steps=16, use_gemv=True)
+    g2 = m.generate(prompt, steps=16, use_gemv=True)
+    assert torch.equal(g1, g2) and g1.shape == (24, 4)
+    assert (g1 < 512).all() and (g1 >= 0).all()
