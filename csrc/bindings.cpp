@@ -331,12 +331,12 @@ Tensor decode_gemv(Tensor A_t, Tensor x, c10::optional<Tensor> bias) {
                               out.data_ptr(), nullptr, (int)B, N, (int)K,
                               cur_stream());
     TORCH_CHECK(rc == 0, "decode_gemv: shape outside fast-path envelope "
-                "(B<=16, B*K*2 <= 160 KiB LDS, K % 512 == 0)");
+                "(B<=16, B*K*2 <= 160 KiB LDS, K % 1024 == 0)");
     return out;
 }
 
 bool decode_gemv_ok(long B, long K) {
-    return B <= 16 && B * K * 2 <= 160 * 1024 && K % 512 == 0;
+    return B <= 16 && B * K * 2 <= 160 * 1024 && K % 1024 == 0;
 }
 
 // ---------------------------------------------------------------------

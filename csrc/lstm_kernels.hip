@@ -1017,15 +1017,21 @@ __global__ __launch_bounds__(256) void k_gemv_bf16(
     float acc[BMAX];
 #pragma unroll
     for (int b = 0; b < BMAX; ++b) acc[b] = 0.f;
-    for (int k0 = lane * 8; k0 < K; k0 += 64 * 8) {
-        const bf16x8 av = *(const bf16x8*)(ar + k0);
+    // two bf16x8 loads in flight per lane per iteration (deeper MLP --
+    // a single outstanding 16B load per lane leaves the row stream
+    // latency-exposed)
+    for (int k0 = lane * 16; k0 < K; k0 += 64 * 16) {
+        const bf16x8 av0 = *(const bf16x8*)(ar + k0);
+        const bf16x8 av1 = *(const bf16x8*)(ar + k0 + 8);
 #pragma unroll
         for (int b = 0; b < BMAX; ++b) {
             if (b < B) {
-                const bf16x8 xv = *(const bf16x8*)(xs + b * K + k0);
+                const bf16x8 xv0 = *(const bf16x8*)(xs + b * K + k0);
+                const bf16x8 xv1 = *(const bf16x8*)(xs + b * K + k0 + 8);
 #pragma unroll
                 for (int j = 0; j < 8; ++j)
-                    acc[b] += (float)av[j] * (float)xv[j];
+                    acc[b] += (float)av0[j] * (float)xv0[j]
+                            + (float)av1[j] * (float)xv1[j];
             }
         }
     }
