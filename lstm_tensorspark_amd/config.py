@@ -53,6 +53,9 @@ class TrainConfig:
     # scalars), single BPTT window, dropout==0.  Ineligible configs fall
     # back to eager silently.
     cuda_graph: bool = False
+    # measure split-vs-fused recurrent path for THIS shape at init instead
+    # of the built-in heuristic (ops/autotune.py)
+    autotune: bool = False
     # fault injection (failure-detection tests): raise on this rank at this
     # step; -1 = disabled
     fault_step: int = -1

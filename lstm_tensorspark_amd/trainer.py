@@ -80,6 +80,11 @@ class Trainer:
             from .ops.extension import extension_available, get_extension
             if extension_available():
                 get_extension().set_fp8_fwd(cfg.train.dtype == "fp8")
+                if getattr(cfg.train, "autotune", False):
+                    from .ops.autotune import tune_fwd
+                    r = tune_fwd(cfg.train.batch_size, cfg.model.hidden_size)
+                    if self.ctx.is_main:
+                        self.log.info(f"autotune fwd path: {r}")
         single_window = cfg.train.bptt <= 0 or cfg.train.bptt >= cfg.train.seq_len
         enable_side_dw(getattr(cfg.dist, "side_dw", False)
                        and self.bucketer is None and single_window

@@ -124,3 +124,30 @@ def test_gpu_lm_generate():
     g2 = m.generate(prompt, steps=16, use_gemv=True)
     assert torch.equal(g1, g2) and g1.shape == (24, 4)
     assert (g1 < 512).all() and (g1 >= 0).all()
+
+
+def test_gpu_fp8_train_step():
+    """fp8 mode end-to-end trainer step: finite loss, weights update."""
+    cfg = copy_cfg(steps=2)
+    cfg.train.dtype = "fp8"
+    cfg.model.input_size = 1024
+    cfg.model.hidden_size = 1024
+    cfg.train.batch_size = 2048      # B*H >= 2M -> split+fp8 path
+    t = Trainer(cfg, device="cuda:0")
+    w0 = t.model.weight("layer0/W_h").detach().clone()
+    info = t.train_step(t.data.batch(0))
+    assert info["loss"] == info["loss"]
+    assert not torch.equal(w0, t.model.weight("layer0/W_h"))
+
+
+def test_gpu_autotune_picks_and_pins():
+    from lstm_tensorspark_amd.ops.autotune import tune_fwd
+    from lstm_tensorspark_amd.ops.extension import get_extension
+
+    ext = get_extension()
+    try:
+        r = tune_fwd(4096, 1024, T=4, iters=3)
+        assert r["choice"] in ("split", "fused")
+        assert ext.get_split_fwd() in (0, 1)
+    finally:
+        ext.set_split_fwd(-1)
