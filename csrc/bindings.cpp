@@ -37,6 +37,7 @@ void launch_quant_fp8_fixed(const void*, void*, long, float, hipStream_t);
 void launch_quant_fp8_dyn(const void*, void*, long, const float*, hipStream_t);
 int launch_gemv_bf16(const void*, const void*, const float*, void*, float*,
                      int, long, int, hipStream_t);
+int launch_colsum_bf16(const void*, float*, long, long, hipStream_t);
 void launch_quant_fp8_dyn_T(const void*, void*, long, long, const float*,
                             hipStream_t);
 void launch_xent_fwd(const void*, const long*, float*, float*, long, int,
@@ -335,6 +336,19 @@ Tensor decode_gemv(Tensor A_t, Tensor x, c10::optional<Tensor> bias) {
     return out;
 }
 
+Tensor colsum_bf16(Tensor t) {
+    // out (C,) fp32 = sum over rows of (R, C) bf16, single pass
+    check_bf16_cuda(t, "t");
+    TORCH_CHECK(t.dim() == 2, "colsum wants (R,C)");
+    t = t.contiguous();
+    const long R = t.size(0), C = t.size(1);
+    auto out = at::zeros({C}, t.options().dtype(at::kFloat));
+    int rc = launch_colsum_bf16(t.data_ptr(), (float*)out.data_ptr(), R, C,
+                                cur_stream());
+    TORCH_CHECK(rc == 0, "colsum: C % 8 != 0");
+    return out;
+}
+
 bool decode_gemv_ok(long B, long K) {
     return B <= 16 && B * K * 2 <= 160 * 1024 && K % 1024 == 0;
 }
@@ -475,6 +489,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("lstm_rec_fwd_step", &lstm_rec_fwd_step,
           "single fused recurrent step (raw_debug for GEMM-mapping tests)");
     m.def("set_persistent_fwd", [](bool v) { g_persistent_enabled = v; });
+    m.def("colsum_bf16", &colsum_bf16, "single-pass fp32 column sum");
     m.def("decode_gemv", &decode_gemv, "small-B serving GEMV",
           py::arg("A_t"), py::arg("x"), py::arg("bias") = py::none());
     m.def("decode_gemv_ok", &decode_gemv_ok);

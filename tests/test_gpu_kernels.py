@@ -444,3 +444,12 @@ def test_decode_gemv_vs_mm(ext, B, N, K):
     ref = (x.float() @ A_t.float().t()) + bias
     err = (out.float() - ref).abs().max().item()
     assert err < 0.03 * max(1.0, ref.abs().max().item()), err
+
+
+def test_colsum_vs_torch(ext):
+    g = torch.Generator(device="cpu").manual_seed(41)
+    for R, C in [(1000, 128), (4096, 4096), (33, 8)]:
+        t = (torch.randn(R, C, generator=g) * 0.1).to("cuda").bfloat16()
+        out = ext.colsum_bf16(t)
+        ref = t.float().sum(dim=0)
+        assert (out - ref).abs().max().item() < 1e-2 * max(1.0, ref.abs().max().item())
