@@ -311,10 +311,13 @@ class LSTMModel(nn.Module):
 
     def _gemv_cache(self):
         """Serving weight layout: pre-transposed [W_x;W_h]^T per layer and
-        head W^T, built ONCE (doubles head-weight memory -- a serving
-        cache, not a training cost)."""
+        head W^T (doubles head-weight memory -- a serving cache, not a
+        training cost).  Rebuilt whenever the weights version changes
+        (Trainer bumps `_weights_version` after every optimizer step), so
+        train -> generate -> train -> generate serves CURRENT weights."""
         gc = getattr(self, "_gemv_weights", None)
-        if gc is not None:
+        ver = getattr(self, "_weights_version", 0)
+        if gc is not None and gc.get("version") == ver:
             return gc
         cfg = self.cfg
         layers = []
@@ -326,8 +329,14 @@ class LSTMModel(nn.Module):
             layers.append((wcat_t, b.detach().float()))
         head_t = self.weight("head/W").detach().t().contiguous()
         head_b = self.weight("head/b").detach().float()
-        gc = {"layers": layers, "head_t": head_t, "head_b": head_b}
+        gc = {"layers": layers, "head_t": head_t, "head_b": head_b,
+              "version": ver}
         self._gemv_weights = gc
+        # the captured decode graph reads the gemv cache tensors by
+        # address: a rebuilt cache means new tensors, so drop the graph too
+        if getattr(self, "_dec_cache", None) is not None \
+                and self._dec_cache.get("gemv"):
+            self._dec_cache = None
         return gc
 
     def _gemv_step(self, cur, g_state):

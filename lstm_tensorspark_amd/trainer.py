@@ -152,6 +152,9 @@ class Trainer:
         if cfg.train.grad_clip > 0:
             norm = self.optimizer.clip_grad_norm(cfg.train.grad_clip)
         self.optimizer.step()
+        # invalidate serving-weight caches (model.generate's gemv layout)
+        self.model._weights_version = getattr(
+            self.model, "_weights_version", 0) + 1
         return {"loss": total_loss,
                 "grad_norm": float(norm) if norm is not None else None}
 
@@ -196,6 +199,8 @@ class Trainer:
             self._graph_io[0].copy_(x, non_blocking=True)
             self._graph_io[1].copy_(y, non_blocking=True)
             self._graph.replay()
+            self.model._weights_version = getattr(
+                self.model, "_weights_version", 0) + 1
         return {"loss": float(self._g_loss),
                 "grad_norm": float(self._g_norm)
                 if self._g_norm is not None else None}

@@ -125,3 +125,15 @@ def test_lm_generate_decode():
     logits, _ = m(out)
     for t in range(5, 9):
         assert torch.equal(logits[t - 1].argmax(-1), out[t])
+
+
+def test_weights_version_bumps_on_step():
+    from lstm_tensorspark_amd.config import Config, ModelConfig, TrainConfig
+    from lstm_tensorspark_amd.trainer import Trainer
+
+    cfg = Config(model=ModelConfig(input_size=16, hidden_size=16, num_classes=8),
+                 train=TrainConfig(batch_size=4, seq_len=8, log_interval=0))
+    t = Trainer(cfg, device="cpu")
+    v0 = getattr(t.model, "_weights_version", 0)
+    t.train_step(t.data.batch(0))
+    assert getattr(t.model, "_weights_version", 0) == v0 + 1
