@@ -315,5 +315,7 @@ class Trainer:
     def load(self, path: str):
         payload = load_checkpoint(path, self.model, self.optimizer)
         self.step = payload["step"]
+        self.model._weights_version = getattr(
+            self.model, "_weights_version", 0) + 1
         self.log.info(f"checkpoint loaded: {path} (step {self.step})")
         return payload
