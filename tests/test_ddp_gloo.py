@@ -130,3 +130,22 @@ def test_2rank_fp32_wire_reduction(tmp_path):
     for name in got:
         want = (per_rank[0][name] + per_rank[1][name]) / 2
         assert (got[name] - want).abs().max() < 1e-6, name
+
+
+def _eval_worker(ctx, tmpdir):
+    cfg = small_cfg()
+    t = Trainer(cfg, dist_ctx=ctx)
+    r = t.evaluate(steps=2)
+    torch.save(r, os.path.join(tmpdir, f"eval_rank{ctx.rank}.pt"))
+
+
+def test_2rank_evaluate_allreduces_mean(tmp_path):
+    """evaluate() must return the SAME rank-averaged loss on every rank."""
+    run_spawned(_eval_worker, world_size=2, backend="gloo",
+                args=(str(tmp_path),), master_port=29551)
+    r0 = torch.load(os.path.join(str(tmp_path), "eval_rank0.pt"),
+                    weights_only=False)
+    r1 = torch.load(os.path.join(str(tmp_path), "eval_rank1.pt"),
+                    weights_only=False)
+    assert abs(r0["eval_loss"] - r1["eval_loss"]) < 1e-9
+    assert r0["steps"] == 2
