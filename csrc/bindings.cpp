@@ -336,6 +336,24 @@ Tensor decode_gemv(Tensor A_t, Tensor x, c10::optional<Tensor> bias) {
     return out;
 }
 
+void decode_cell_(Tensor raw, Tensor h, Tensor c) {
+    // In-place decode cell update: raw (B,4H) bf16 = full pre-activation
+    // gates; h (B,H) bf16 and c (B,H) fp32 are the state buffers, updated
+    // in place (ONE kernel replaces the ~12 elementwise ops of the eager
+    // cell math -- per-token decode is small-kernel-latency bound).
+    check_bf16_cuda(raw, "raw");
+    check_bf16_cuda(h, "h");
+    const long B = raw.size(0), H4 = raw.size(1), H = H4 / 4;
+    TORCH_CHECK(h.size(0) == B && h.size(1) == H && c.size(1) == H,
+                "state shape mismatch");
+    TORCH_CHECK(raw.is_contiguous() && h.is_contiguous() && c.is_contiguous());
+    TORCH_CHECK(c.scalar_type() == at::kFloat);
+    launch_lstm_pw_fwd(raw.data_ptr(), H4, nullptr, 0,
+                       (const float*)c.data_ptr(), H,
+                       h.data_ptr(), H, (float*)c.data_ptr(), H,
+                       nullptr, 0, (int)B, (int)H, cur_stream());
+}
+
 Tensor colsum_bf16(Tensor t) {
     // out (C,) fp32 = sum over rows of (R, C) bf16, single pass
     check_bf16_cuda(t, "t");
@@ -490,6 +508,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "single fused recurrent step (raw_debug for GEMM-mapping tests)");
     m.def("set_persistent_fwd", [](bool v) { g_persistent_enabled = v; });
     m.def("colsum_bf16", &colsum_bf16, "single-pass fp32 column sum");
+    m.def("decode_cell_", &decode_cell_, "in-place fused decode cell update");
     m.def("decode_gemv", &decode_gemv, "small-B serving GEMV",
           py::arg("A_t"), py::arg("x"), py::arg("bias") = py::none());
     m.def("decode_gemv_ok", &decode_gemv_ok);

@@ -927,14 +927,19 @@ __global__ __launch_bounds__(256) void k_lstm_pw_fwd(
     if (row >= B) return;
     const int col0 = (int)(idx % cpr) * W;
     const __hip_bfloat16* rp = raw_rec + (long)row * rr_stride + col0;
-    const __hip_bfloat16* pp = pre + (long)row * pre_stride + col0;
+    const __hip_bfloat16* pp = pre ? pre + (long)row * pre_stride + col0
+                                   : nullptr;   // null = zero (decode path)
     fv raw[4];
 #pragma unroll
     for (int g = 0; g < 4; ++g) {
         const bfv rv = *(const bfv*)(rp + (long)g * H);
-        const bfv pv = *(const bfv*)(pp + (long)g * H);
 #pragma unroll
-        for (int j = 0; j < W; ++j) raw[g][j] = (float)rv[j] + (float)pv[j];
+        for (int j = 0; j < W; ++j) raw[g][j] = (float)rv[j];
+        if (pp) {
+            const bfv pv = *(const bfv*)(pp + (long)g * H);
+#pragma unroll
+            for (int j = 0; j < W; ++j) raw[g][j] += (float)pv[j];
+        }
     }
     const fv cp = *(const fv*)(c_prev + (long)row * cp_stride + col0);
     bfv oi, of, og, oo, oh;
@@ -955,11 +960,13 @@ __global__ __launch_bounds__(256) void k_lstm_pw_fwd(
     }
     *(bfv*)(h_out + (long)row * ho_stride + col0) = oh;
     *(fv*)(c_out + (long)row * co_stride + col0) = oc;
-    __hip_bfloat16* ap = acts + (long)row * act_stride + col0;
-    *(bfv*)(ap)               = oi;
-    *(bfv*)(ap + (long)H)     = of;
-    *(bfv*)(ap + (long)2 * H) = og;
-    *(bfv*)(ap + (long)3 * H) = oo;
+    if (acts) {   // null = decode path (no backward, no saved activations)
+        __hip_bfloat16* ap = acts + (long)row * act_stride + col0;
+        *(bfv*)(ap)               = oi;
+        *(bfv*)(ap + (long)H)     = of;
+        *(bfv*)(ap + (long)2 * H) = og;
+        *(bfv*)(ap + (long)3 * H) = oo;
+    }
 }
 
 extern "C" void launch_lstm_pw_fwd(
@@ -1047,44 +1054,49 @@ __global__ __launch_bounds__(256) void k_gemv_bf16(
     }
     __syncthreads();
     const int wave = tid >> 6, lane = tid & 63;
-    const long n = (long)blockIdx.x * 4 + wave;
-    if (n >= N) return;
-    const __hip_bfloat16* ar = A + n * K;
-    float acc[BMAX];
+    const long wave_gid = (long)blockIdx.x * 4 + wave;
+    const long nwaves = (long)gridDim.x * 4;
+    // persistent wave-loop over output rows: keeps the weight stream
+    // continuous instead of paying per-row wave ramp-up (K/1024 is only a
+    // handful of iterations per row)
+    for (long n = wave_gid; n < N; n += nwaves) {
+        const __hip_bfloat16* ar = A + n * K;
+        float acc[BMAX];
 #pragma unroll
-    for (int b = 0; b < BMAX; ++b) acc[b] = 0.f;
-    // two bf16x8 loads in flight per lane per iteration (deeper MLP --
-    // a single outstanding 16B load per lane leaves the row stream
-    // latency-exposed)
-    for (int k0 = lane * 16; k0 < K; k0 += 64 * 16) {
-        const bf16x8 av0 = *(const bf16x8*)(ar + k0);
-        const bf16x8 av1 = *(const bf16x8*)(ar + k0 + 8);
+        for (int b = 0; b < BMAX; ++b) acc[b] = 0.f;
+        // two bf16x8 loads in flight per lane per iteration (deeper MLP --
+        // a single outstanding 16B load per lane leaves the row stream
+        // latency-exposed)
+        for (int k0 = lane * 16; k0 < K; k0 += 64 * 16) {
+            const bf16x8 av0 = *(const bf16x8*)(ar + k0);
+            const bf16x8 av1 = *(const bf16x8*)(ar + k0 + 8);
 #pragma unroll
-        for (int b = 0; b < BMAX; ++b) {
-            if (b < B) {
-                const bf16x8 xv0 = *(const bf16x8*)(xs + b * K + k0);
-                const bf16x8 xv1 = *(const bf16x8*)(xs + b * K + k0 + 8);
+            for (int b = 0; b < BMAX; ++b) {
+                if (b < B) {
+                    const bf16x8 xv0 = *(const bf16x8*)(xs + b * K + k0);
+                    const bf16x8 xv1 = *(const bf16x8*)(xs + b * K + k0 + 8);
 #pragma unroll
-                for (int j = 0; j < 8; ++j)
-                    acc[b] += (float)av0[j] * (float)xv0[j]
-                            + (float)av1[j] * (float)xv1[j];
+                    for (int j = 0; j < 8; ++j)
+                        acc[b] += (float)av0[j] * (float)xv0[j]
+                                + (float)av1[j] * (float)xv1[j];
+                }
             }
         }
-    }
-#pragma unroll
-    for (int b = 0; b < BMAX; ++b)
-#pragma unroll
-        for (int off = 32; off > 0; off >>= 1)
-            acc[b] += __shfl_down(acc[b], off, 64);
-    if (lane == 0) {
-        const float bv = bias ? bias[n] : 0.f;
 #pragma unroll
         for (int b = 0; b < BMAX; ++b)
-            if (b < B) {
-                const float v = acc[b] + bv;
-                if (out) out[(long)b * N + n] = __float2bfloat16(v);
-                else out_f32[(long)b * N + n] = v;
-            }
+#pragma unroll
+            for (int off = 32; off > 0; off >>= 1)
+                acc[b] += __shfl_down(acc[b], off, 64);
+        if (lane == 0) {
+            const float bv = bias ? bias[n] : 0.f;
+#pragma unroll
+            for (int b = 0; b < BMAX; ++b)
+                if (b < B) {
+                    const float v = acc[b] + bv;
+                    if (out) out[(long)b * N + n] = __float2bfloat16(v);
+                    else out_f32[(long)b * N + n] = v;
+                }
+        }
     }
 }
 

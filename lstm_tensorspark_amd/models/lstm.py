@@ -350,16 +350,9 @@ class LSTMModel(nn.Module):
         for l, (wcat_t, b) in enumerate(gc["layers"]):
             h, c = g_state[l]
             xcat = torch.cat([x, h], dim=1)
-            raw = ext.decode_gemv(wcat_t, xcat, b).float()
-            i = torch.sigmoid(raw[:, :H])
-            f = torch.sigmoid(raw[:, H:2 * H])
-            g = torch.tanh(raw[:, 2 * H:3 * H])
-            o = torch.sigmoid(raw[:, 3 * H:])
-            cn = f * c + i * g
-            hn = (o * torch.tanh(cn)).to(x.dtype)
-            h.copy_(hn)
-            c.copy_(cn)
-            x = hn
+            raw = ext.decode_gemv(wcat_t, xcat, b)
+            ext.decode_cell_(raw, h, c)   # in-place state update, ONE kernel
+            x = h
         logits = ext.decode_gemv(gc["head_t"], x, gc["head_b"])
         return logits.argmax(dim=-1)
 
