@@ -345,7 +345,6 @@ class LSTMModel(nn.Module):
         from ..ops.extension import get_extension
         ext = get_extension()
         gc = self._gemv_cache()
-        H = self.cfg.hidden_size
         x = F.embedding(cur, self.weight("embed/W"))
         for l, (wcat_t, b) in enumerate(gc["layers"]):
             h, c = g_state[l]

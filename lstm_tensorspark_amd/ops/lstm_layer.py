@@ -33,8 +33,6 @@ exercised by the CPU test tiers.
 """
 from __future__ import annotations
 
-from typing import Optional
-
 import torch
 
 from . import extension, functional

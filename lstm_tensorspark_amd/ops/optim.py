@@ -13,7 +13,7 @@ gradient, then writes the rounded bf16 copy back into the parameter.
 from __future__ import annotations
 
 import math
-from typing import Iterable, List, Optional
+from typing import Iterable, List
 
 import torch
 
