@@ -137,3 +137,31 @@ def test_weights_version_bumps_on_step():
     v0 = getattr(t.model, "_weights_version", 0)
     t.train_step(t.data.batch(0))
     assert getattr(t.model, "_weights_version", 0) == v0 + 1
+
+
+def test_generate_temperature_sampling_cpu():
+    import torch
+    from lstm_tensorspark_amd.config import ModelConfig
+    from lstm_tensorspark_amd.models.lstm import LSTMModel
+
+    torch.manual_seed(4)
+    cfg = ModelConfig(input_size=16, hidden_size=16, num_layers=1,
+                      head="lm", vocab_size=32)
+    m = LSTMModel(cfg)
+    prompt = torch.randint(0, 32, (4, 2))
+    g = torch.Generator().manual_seed(9)
+    a = m.generate(prompt, steps=6, temperature=1.0, generator=g)
+    g2 = torch.Generator().manual_seed(9)
+    b = m.generate(prompt, steps=6, temperature=1.0, generator=g2)
+    assert torch.equal(a, b)          # seeded sampling is reproducible
+    assert a.shape == (10, 2)
+    g3 = torch.Generator().manual_seed(10)
+    c = m.generate(prompt, steps=6, temperature=1.0, generator=g3)
+    assert not torch.equal(a, c) or True  # different seed usually differs
+
+
+def test_autotune_cpu_noop():
+    from lstm_tensorspark_amd.ops.autotune import tune_fwd
+
+    r = tune_fwd(64, 64)
+    assert r["choice"] == "heuristic"
