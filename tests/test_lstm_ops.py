@@ -153,3 +153,48 @@ def test_grad_parity_vs_torch_nn_lstm():
                     - getattr(ref, f"weight_hh_l{l}{sfx}").grad.t()).abs().max() < 1e-11
             assert (model.weight(f"{name}/b").grad
                     - getattr(ref, f"bias_ih_l{l}{sfx}").grad).abs().max() < 1e-11
+
+
+def test_reverse_equals_flipped_forward():
+    """reverse=True over x must equal reverse=False over time-flipped x,
+    time-flipped back (same weights, fp64 exact)."""
+    tensors = make_inputs(3, 7, 5, 6, seed=11)
+    x, h0, c0, w_x, w_h, b = tensors
+    h_rev, hT_r, cT_r = lstm_layer(x, h0, c0, w_x, w_h, b, reverse=True)
+    h_fwd, hT_f, cT_f = lstm_layer(x.flip(0), h0, c0, w_x, w_h, b,
+                                   reverse=False)
+    assert torch.equal(h_rev, h_fwd.flip(0))
+    assert torch.equal(hT_r, hT_f) and torch.equal(cT_r, cT_f)
+
+
+def test_bptt_equals_full_when_window_covers_seq():
+    """bptt == seq_len must be bitwise-identical to no truncation."""
+    from lstm_tensorspark_amd.config import Config, ModelConfig, TrainConfig
+
+    from lstm_tensorspark_amd.trainer import Trainer
+
+    def run(bptt):
+        cfg = Config(
+            model=ModelConfig(input_size=8, hidden_size=8, num_classes=4),
+            train=TrainConfig(batch_size=3, seq_len=6, bptt=bptt,
+                              log_interval=0, seed=5))
+        t = Trainer(cfg, device="cpu")
+        t.train_step(t.data.batch(0))
+        return {k: v.detach().clone() for k, v in t.model.weight_dict().items()}
+
+    w_full = run(0)
+    w_win = run(6)
+    for k in w_full:
+        assert torch.equal(w_full[k], w_win[k]), k
+
+
+def test_carry_state_continuity():
+    """Splitting a sequence into two halves with carried (h, c) state must
+    reproduce the single-pass forward exactly (fp64)."""
+    tensors = make_inputs(2, 8, 4, 5, seed=13)
+    x, h0, c0, w_x, w_h, b = tensors
+    h_all, hT, cT = lstm_layer(x, h0, c0, w_x, w_h, b)
+    h1, hT1, cT1 = lstm_layer(x[:4], h0, c0, w_x, w_h, b)
+    h2, hT2, cT2 = lstm_layer(x[4:], hT1, cT1, w_x, w_h, b)
+    assert torch.equal(torch.cat([h1, h2], dim=0), h_all)
+    assert torch.equal(hT2, hT) and torch.equal(cT2, cT)
