@@ -22,7 +22,7 @@ import torch
 
 sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
 
-from lstm_tensorspark_amd.config import get_preset
+from lstm_tensorspark_amd.config import PRESETS, get_preset
 from lstm_tensorspark_amd.parallel.launch import cleanup, init_from_env
 from lstm_tensorspark_amd.trainer import Trainer
 
@@ -33,7 +33,7 @@ def parse_args():
                    help="world size (informational; actual world from env)")
     p.add_argument("--steps", type=int, default=20)
     p.add_argument("--warmup", type=int, default=5)
-    p.add_argument("--preset", default="mi355x-1gpu")
+    p.add_argument("--preset", default="mi355x-1gpu", choices=sorted(PRESETS))
     p.add_argument("--batch", type=int, default=0,
                    help="override per-rank batch size")
     p.add_argument("--dtype", default="",

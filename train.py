@@ -10,14 +10,15 @@ import sys
 
 sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
 
-from lstm_tensorspark_amd.config import Config, get_preset
+from lstm_tensorspark_amd.config import PRESETS, Config, get_preset
 from lstm_tensorspark_amd.parallel.launch import cleanup, init_from_env
 from lstm_tensorspark_amd.trainer import Trainer
 
 
 def main():
     p = argparse.ArgumentParser()
-    p.add_argument("--preset", default=None)
+    p.add_argument("--preset", default=None, choices=sorted(PRESETS),
+                   help="named BASELINE config")
     p.add_argument("--config", default=None, help="YAML config file")
     p.add_argument("--set", action="append", default=[], metavar="SEC.KEY=V",
                    help="config override, repeatable")
