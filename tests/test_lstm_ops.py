@@ -198,3 +198,33 @@ def test_carry_state_continuity():
     h2, hT2, cT2 = lstm_layer(x[4:], hT1, cT1, w_x, w_h, b)
     assert torch.equal(torch.cat([h1, h2], dim=0), h_all)
     assert torch.equal(hT2, hT) and torch.equal(cT2, cT)
+
+
+def test_cpu_shape_fuzz_vs_autograd():
+    """Randomized CPU fp64 fuzz: manual backward vs torch autograd on
+    arbitrary (B,T,X,H), both directions."""
+    import random
+
+    rng = random.Random(7)
+    for i in range(6):
+        B = rng.choice([1, 2, 5])
+        T = rng.choice([1, 3, 8])
+        X = rng.choice([1, 4, 9])
+        H = rng.choice([2, 6, 11])
+        reverse = bool(i % 2)
+        tensors = [t.requires_grad_(True)
+                   for t in make_inputs(B, T, X, H, seed=200 + i)]
+        g = torch.Generator().manual_seed(300 + i)
+        gout = torch.randn(T, B, H, generator=g, dtype=torch.float32).double()
+
+        hseq_r, _, _ = autograd_ref(*tensors, reverse)
+        loss_r = (hseq_r * gout).sum()
+        grads_r = torch.autograd.grad(loss_r, tensors)
+
+        t2 = [t.detach().clone().requires_grad_(True) for t in tensors]
+        hseq, _, _ = lstm_layer(*t2, reverse=reverse)
+        loss = (hseq * gout).sum()
+        grads = torch.autograd.grad(loss, t2)
+        for name, a, r in zip("x h0 c0 w_x w_h b".split(), grads, grads_r):
+            err = (a - r).abs().max().item()
+            assert err < 1e-11, f"{(B,T,X,H,reverse)} d{name}: {err}"

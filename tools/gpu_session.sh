@@ -1,7 +1,8 @@
 set -x
-mkdir -p gpurun_out
-R=$GRAFT_REPO_ROOT
-cd /tmp && export TMPDIR=/tmp
-timeout 300 rocprofv3 --kernel-trace --stats -d $R/gpurun_out/prof_lm -- python $R/bench.py --steps 3 --warmup 1 --preset lm-50k > $R/gpurun_out/prof_lm.log 2>&1
-timeout 300 rocprofv3 --kernel-trace --stats -d $R/gpurun_out/prof_bd -- python $R/bench.py --steps 3 --warmup 1 --preset bidir-4l > $R/gpurun_out/prof_bd.log 2>&1
-echo done
+python - <<'PY'
+import __graft_entry__
+__graft_entry__.smoke()
+PY
+python -m pytest tests -m gpu -q 2>&1 | tail -1
+timeout 300 python tools/kernel_bench.py --B 4096 --H 1024 --T 16 --iters 20 2>&1 | grep -E "best|seq_fwd p|seq_bwd|layer" | head -5
+timeout 300 python bench.py --steps 10 --warmup 2 2>&1 | tail -1
