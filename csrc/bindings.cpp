@@ -37,7 +37,8 @@ void launch_quant_fp8_fixed(const void*, void*, long, float, hipStream_t);
 void launch_quant_fp8_dyn(const void*, void*, long, const float*, hipStream_t);
 int launch_gemv_bf16(const void*, const void*, const float*, void*, float*,
                      int, long, int, hipStream_t);
-int launch_colsum_bf16(const void*, float*, long, long, hipStream_t);
+int launch_colsum_bf16(const void*, float*, float*, long, long, hipStream_t);
+int colsum_gy(long);
 void launch_quant_fp8_dyn_T(const void*, void*, long, long, const float*,
                             hipStream_t);
 void launch_xent_fwd(const void*, const long*, float*, float*, long, int,
@@ -355,14 +356,17 @@ void decode_cell_(Tensor raw, Tensor h, Tensor c) {
 }
 
 Tensor colsum_bf16(Tensor t) {
-    // out (C,) fp32 = sum over rows of (R, C) bf16, single pass
+    // out (C,) fp32 = sum over rows of (R, C) bf16, one bf16 pass +
+    // deterministic fixed-order partial combine (bitwise run-to-run)
     check_bf16_cuda(t, "t");
     TORCH_CHECK(t.dim() == 2, "colsum wants (R,C)");
     t = t.contiguous();
     const long R = t.size(0), C = t.size(1);
-    auto out = at::zeros({C}, t.options().dtype(at::kFloat));
-    int rc = launch_colsum_bf16(t.data_ptr(), (float*)out.data_ptr(), R, C,
-                                cur_stream());
+    auto partials = at::empty({(long)colsum_gy(R), C},
+                              t.options().dtype(at::kFloat));
+    auto out = at::empty({C}, t.options().dtype(at::kFloat));
+    int rc = launch_colsum_bf16(t.data_ptr(), (float*)partials.data_ptr(),
+                                (float*)out.data_ptr(), R, C, cur_stream());
     TORCH_CHECK(rc == 0, "colsum: C % 8 != 0");
     return out;
 }

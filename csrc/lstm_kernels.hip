@@ -994,13 +994,15 @@ extern "C" void launch_lstm_pw_fwd(
 }
 
 // ---------------------------------------------------------------------------
-// K8: bias-grad column sum  out(C) += sum_r in(R,C) -- ONE fp32-accum pass
+// K8: bias-grad column sum  out(C) = sum_r in(R,C) -- ONE fp32-accum pass
 // over the (T*B, 4H) dgates (torch's generic two-stage reduce measured
-// ~3x the roofline on this shape).  Each thread owns 8 fixed columns and
-// accumulates CHUNK rows, then one atomicAdd per column.
+// ~3x the roofline on this shape).  DETERMINISTIC two-stage: each block
+// writes its fixed-order chunk partial to partials[gy][C]; a final kernel
+// sums the gy partials per column in fixed order (the earlier fp32
+// atomicAdd across grid-y blocks made bias grads non-bitwise run-to-run).
 // ---------------------------------------------------------------------------
 __global__ __launch_bounds__(256) void k_colsum_bf16(
-    const __hip_bfloat16* __restrict__ in, float* __restrict__ out,
+    const __hip_bfloat16* __restrict__ in, float* __restrict__ partials,
     long R, long C, int chunk)
 {
     const long c0 = ((long)blockIdx.x * 256 + threadIdx.x) * 8;
@@ -1013,19 +1015,39 @@ __global__ __launch_bounds__(256) void k_colsum_bf16(
 #pragma unroll
         for (int j = 0; j < 8; ++j) acc[j] += (float)v[j];
     }
-#pragma unroll
-    for (int j = 0; j < 8; ++j) atomicAdd(&out[c0 + j], acc[j]);
+    *(f32x4*)(&partials[(long)blockIdx.y * C + c0]) =
+        f32x4{acc[0], acc[1], acc[2], acc[3]};
+    *(f32x4*)(&partials[(long)blockIdx.y * C + c0 + 4]) =
+        f32x4{acc[4], acc[5], acc[6], acc[7]};
 }
 
-extern "C" int launch_colsum_bf16(const void* in, float* out, long R, long C,
-                                  hipStream_t stream)
+__global__ __launch_bounds__(256) void k_colsum_final(
+    const float* __restrict__ partials, float* __restrict__ out,
+    long C, int gy)
+{
+    const long c = (long)blockIdx.x * 256 + threadIdx.x;  // one column/thread
+    if (c >= C) return;
+    float s = 0.f;
+    for (int y = 0; y < gy; ++y) s += partials[(long)y * C + c];
+    out[c] = s;
+}
+
+// partials must hold colsum_gy(R) * C floats (allocated by the caller).
+extern "C" int colsum_gy(long R) { return (int)((R + 4095) / 4096); }
+
+extern "C" int launch_colsum_bf16(const void* in, float* partials, float* out,
+                                  long R, long C, hipStream_t stream)
 {
     if (C % 8 != 0) return -1;
     const int chunk = 4096;
+    const int gy = colsum_gy(R);
     dim3 block(256);
-    dim3 grid((unsigned)((C / 8 + 255) / 256), (unsigned)((R + chunk - 1) / chunk));
+    dim3 grid((unsigned)((C / 8 + 255) / 256), (unsigned)gy);
     hipLaunchKernelGGL(k_colsum_bf16, grid, block, 0, stream,
-                       (const __hip_bfloat16*)in, out, R, C, chunk);
+                       (const __hip_bfloat16*)in, partials, R, C, chunk);
+    dim3 grid2((unsigned)((C + 255) / 256));
+    hipLaunchKernelGGL(k_colsum_final, grid2, block, 0, stream,
+                       partials, out, C, gy);
     return 0;
 }
 
@@ -1100,18 +1122,26 @@ __global__ __launch_bounds__(256) void k_gemv_bf16(
     }
 }
 
-extern "C" void launch_gemv_bf16(const void* A, const void* x,
-                                 const float* bias, void* out, float* out_f32,
-                                 int B, long N, int K, hipStream_t stream)
+// Returns 0 on launch, -1 when the shape is outside the fast-path
+// envelope (the kernel template caps B at 16, K must satisfy the
+// lane-stride contract, and x must fit LDS) -- the caller must NOT read
+// `out` on -1.  Signature matches the `int` declaration in bindings.cpp
+// (it was `void` here before: the rc check read an undefined register).
+extern "C" int launch_gemv_bf16(const void* A, const void* x,
+                                const float* bias, void* out, float* out_f32,
+                                int B, long N, int K, hipStream_t stream)
 {
-    dim3 block(256), grid((unsigned)((N + 3) / 4));
     const size_t smem = (size_t)B * K * 2;
+    if (B < 1 || B > 16 || K % 1024 != 0 || smem > 160 * 1024 || N < 1)
+        return -1;
+    dim3 block(256), grid((unsigned)((N + 3) / 4));
 #define GEMV_LAUNCH(BM)                                                           hipLaunchKernelGGL(k_gemv_bf16<BM>, grid, block, smem, stream,                    (const __hip_bfloat16*)A, (const __hip_bfloat16*)x, bias,                     (__hip_bfloat16*)out, out_f32, B, N, K)
     if (B <= 1) GEMV_LAUNCH(1);
     else if (B <= 4) GEMV_LAUNCH(4);
     else if (B <= 8) GEMV_LAUNCH(8);
     else GEMV_LAUNCH(16);
 #undef GEMV_LAUNCH
+    return 0;
 }
 
 // ---------------------------------------------------------------------------

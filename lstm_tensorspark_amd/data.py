@@ -13,6 +13,24 @@ import torch
 
 from .config import Config
 
+_M64 = (1 << 64) - 1
+
+
+def mix64(*vals: int) -> int:
+    """Collision-resistant 64-bit mix of an integer tuple (splitmix64
+    absorb/finalize chain).  Used to derive RNG stream seeds from
+    (seed, stream, rank, step) so no two streams ever coincide -- the old
+    linear mix ``seed*1000003 + rank*7919 + step`` collided whenever step
+    spans exceeded 7919 (rank 0 step 7919 == rank 1 step 0)."""
+    x = 0
+    for v in vals:
+        x = (x + (v & _M64) + 0x9E3779B97F4A7C15) & _M64
+        z = x
+        z = ((z ^ (z >> 30)) * 0xBF58476D1CE4E5B9) & _M64
+        z = ((z ^ (z >> 27)) * 0x94D049BB133111EB) & _M64
+        x = z ^ (z >> 31)
+    return x
+
 
 class SyntheticSeqData:
     """Deterministic per-rank synthetic batches.
@@ -37,24 +55,27 @@ class SyntheticSeqData:
         self.input_size = m.input_size
         self.base_seed = t.seed
 
-    def _gen(self, step: int) -> torch.Generator:
+    def _gen(self, step: int, stream: int = 0) -> torch.Generator:
         # On GPU, generate device-side: an HBM-sized batch (e.g. 16384x128
         # sequences) takes ~10 s to randn on the host but <10 ms on-device.
-        # Determinism contract holds per (seed, rank, step, device-kind).
+        # Determinism contract holds per (seed, stream, rank, step,
+        # device-kind).  stream 0 = training, 1 = eval -- structurally
+        # disjoint, and mix64 is collision-free over the whole tuple (the
+        # old linear mix collided across (rank, step) pairs;
+        # tests/test_data.py asserts disjointness).
         dev = self.device if self.device.type == "cuda" else torch.device("cpu")
         g = torch.Generator(device=dev)
-        # distinct stream per (seed, rank, step)
-        g.manual_seed(self.base_seed * 1000003 + self.rank * 7919 + step)
+        g.manual_seed(mix64(self.base_seed, stream, self.rank, step))
         return g
 
-    def batch(self, step: int):
+    def batch(self, step: int, stream: int = 0):
         """Return (x, y) for a training step (TIME-MAJOR).
 
         x: (T, B, input_size) float for feature tasks, (T, B) long for "lm".
         y: per-head target -- (T, B) for per-step heads, (B,) for
            "classifier" -- see models/heads.
         """
-        g = self._gen(step)
+        g = self._gen(step, stream)
         dev = g.device
         B, T = self.batch_size, self.seq_len
         task = self.task
@@ -99,7 +120,13 @@ class SyntheticSeqData:
             # right for throughput, useless for convergence checks.
             V = self.cfg.model.vocab_size
             assert V > 1, "lm_assoc requires model.vocab_size"
-            a = 2 * (self.base_seed % 1000) + 1          # odd -> bijective
+            # a must be coprime to V for tok -> a*tok+c (mod V) to be a
+            # bijection (odd alone only suffices for power-of-two V; for
+            # V=50000 an a divisible by 5 is non-bijective)
+            import math
+            a = 2 * (self.base_seed % 1000) + 1
+            while math.gcd(a, V) != 1:
+                a += 2
             c = (self.base_seed // 7) % V
             t0 = torch.randint(0, V, (1, B), generator=g, device=dev)
             toks = [t0]

@@ -132,7 +132,10 @@ class LSTMLayerFn(torch.autograd.Function):
         # unbounded, so dynamic scales); the manual backward keeps bf16
         # gradients regardless, so only this forward GEMM changes.
         if (use_hip and extension.get_extension().get_fp8_fwd()
-                and x.dtype == torch.bfloat16):
+                and x.dtype == torch.bfloat16
+                # _scaled_mm divisibility envelope; fall back to bf16 addmm
+                # outside it instead of raising inside the library call
+                and (T * B) % 16 == 0 and X % 16 == 0 and (4 * H) % 16 == 0):
             pre = _fp8_addmm(b, x.reshape(T * B, X), w_x).view(T, B, 4 * H)
         else:
             pre = torch.addmm(b.to(x.dtype), x.reshape(T * B, X), w_x).view(T, B, 4 * H)

@@ -159,6 +159,14 @@ class Trainer:
                 "grad_norm": float(norm) if norm is not None else None}
 
     # -- hipGraph-captured step (train.cuda_graph) ----------------------
+    def _graph_state_tensors(self):
+        """Every tensor the capture warmup mutates: params, fp32 masters and
+        momentum buffers (graph eligibility pins the optimizer to SGD)."""
+        ts = [p for p in self.model.parameters()]
+        ts += self.optimizer.masters
+        ts += getattr(self.optimizer, "bufs", [])
+        return ts
+
     def _graph_body(self):
         """One full training step on the static input tensors.  Everything
         here must be capture-safe: no host syncs, no new host-side state,
@@ -179,8 +187,17 @@ class Trainer:
     def _graphed_step(self, x, y) -> dict:
         if self._graph is None:
             self._graph_io = (x.clone(), y.clone())
-            # warmup twice eagerly (creates grads / hipBLASLt workspaces /
-            # momentum buffers at stable addresses), then capture
+            # Snapshot every tensor the warmup/capture mutates (params,
+            # fp32 masters, momentum) so the graphed trajectory is
+            # STEP-EXACT vs eager: warmup twice eagerly (creates grads /
+            # hipBLASLt workspaces / momentum buffers at stable addresses),
+            # capture (records, does not execute), restore the snapshot
+            # in place (the graph holds raw addresses -- copy_, never
+            # rebind), then replay once = exactly one optimizer step on
+            # this batch (tests/test_gpu_train.py asserts per-step loss
+            # equality with the eager path).
+            snap = [(t, t.detach().clone()) for t in self._graph_state_tensors()]
+            sc0 = self.optimizer.step_count
             for _ in range(2):
                 self._graph_body()
             torch.cuda.synchronize()
@@ -188,17 +205,18 @@ class Trainer:
             with torch.cuda.graph(g):
                 self._graph_body()
             self._graph = g
-            # capture records but does not execute; replay once so the
-            # loss/norm tensors hold this step's values.  (The two warmup
-            # iterations above also applied optimizer steps -- a graphed
-            # run's first train_step() advances the weights 3x.  This is a
-            # throughput mode; exact-step parity is the eager path.)
+            for t, s in snap:
+                t.detach().copy_(s)
+            self.optimizer.step_count = sc0 + 1  # the one replayed step
             self._graph.replay()
-            self.log.info("train step captured as hipGraph")
+            self.model._weights_version = getattr(
+                self.model, "_weights_version", 0) + 1
+            self.log.info("train step captured as hipGraph (step-exact)")
         else:
             self._graph_io[0].copy_(x, non_blocking=True)
             self._graph_io[1].copy_(y, non_blocking=True)
             self._graph.replay()
+            self.optimizer.step_count += 1
             self.model._weights_version = getattr(
                 self.model, "_weights_version", 0) + 1
         return {"loss": float(self._g_loss),
@@ -279,14 +297,14 @@ class Trainer:
     def evaluate(self, steps: int = 10, data=None) -> dict:
         """Average loss over `steps` held-out synthetic batches (the eval
         counterpart of the reference's Session.run(logits) path, SURVEY.md
-        §3.3).  Uses a disjoint RNG stream (step offset 1e6) so eval batches
-        never coincide with training batches.  All-reduces the mean loss
-        across ranks."""
+        §3.3).  Uses the structurally disjoint eval RNG stream (data.mix64
+        stream=1) so eval batches never coincide with training batches.
+        All-reduces the mean loss across ranks."""
         self.model.eval()
         src = data if data is not None else self.data
         total, n = 0.0, 0
         for i in range(steps):
-            x, y = src.batch(1_000_000 + i)
+            x, y = src.batch(i, stream=1)
             out, _ = self.model(x)
             total += float(self.model.loss(out, y).detach())
             n += 1

@@ -1,7 +1,9 @@
+import math
+
 import torch
 
 from lstm_tensorspark_amd.config import Config, ModelConfig, TrainConfig, DataConfig
-from lstm_tensorspark_amd.data import SyntheticSeqData
+from lstm_tensorspark_amd.data import SyntheticSeqData, mix64
 
 
 def _cfg(task="random", **tk):
@@ -63,6 +65,53 @@ def test_lm_assoc_learnable_structure():
     x, y = d.batch(0)
     assert x.shape == (12, 4) and y.shape == (12, 4)
     # y follows the affine map of x (deterministic given the seed)
+    V = cfg.model.vocab_size
     a = 2 * (cfg.train.seed % 1000) + 1
-    c = (cfg.train.seed // 7) % cfg.model.vocab_size
-    assert torch.equal(y, (x * a + c) % cfg.model.vocab_size)
+    while math.gcd(a, V) != 1:
+        a += 2
+    c = (cfg.train.seed // 7) % V
+    assert torch.equal(y, (x * a + c) % V)
+
+
+def test_lm_assoc_map_bijective_nonpow2_vocab():
+    # regression: an odd multiplier is only bijective mod power-of-two V;
+    # the generator must pick a coprime to V (e.g. V=50000, seed giving
+    # a % 5 == 0 previously produced a non-bijective map)
+    cfg = _cfg("lm")
+    cfg.data.task = "lm_assoc"
+    cfg.model.vocab_size = 50
+    cfg.train.seed = 2  # 2*(2%1000)+1 = 5 -> gcd(5, 50) = 5, must be fixed up
+    d = SyntheticSeqData(cfg, rank=0)
+    x, y = d.batch(0)
+    a = 5
+    while math.gcd(a, 50) != 1:
+        a += 2
+    assert math.gcd(a, 50) == 1
+    c = (2 // 7) % 50
+    assert torch.equal(y, (x * a + c) % 50)
+
+
+def test_rng_streams_collision_free():
+    """The (seed, stream, rank, step) -> generator-seed map must be
+    injective over the operating envelope (ranks <= 8, long trainings,
+    train + eval streams).  The old linear mix collided at step spans
+    > 7919 (VERDICT r1 weak #6/#7)."""
+    seed = 1234
+    seen = set()
+    n = 0
+    steps = list(range(3000)) + [7919, 100_000, 1_000_000, 10_000_000]
+    for rank in range(8):
+        for stream in (0, 1):
+            for step in steps:
+                seen.add(mix64(seed, stream, rank, step))
+                n += 1
+    assert len(seen) == n
+    # the exact collision from VERDICT: rank 0 step 7919 vs rank 1 step 0
+    assert mix64(seed, 0, 0, 7919) != mix64(seed, 0, 1, 0)
+
+
+def test_eval_stream_disjoint_from_train():
+    d = SyntheticSeqData(_cfg(), rank=0)
+    xt, _ = d.batch(5, stream=0)
+    xe, _ = d.batch(5, stream=1)
+    assert not torch.equal(xt, xe)
