@@ -717,6 +717,278 @@ epilogue:
 }
 
 // ---------------------------------------------------------------------------
+// 8-PHASE 256x256 fused recurrent GEMM + cell update (guide §5 "256² 8-phase
+// template", the 1.3 PF/s-class structure, adapted to the LSTM epilogue).
+// Differences from k_lstm_rec_fwd_mt256 (the ~500 TF/s round-1 kernel):
+//   - BK=64 K-tiles (stored as TWO BK=32 sub-images so the PROVEN
+//     conflict-free swz64 64-B-row chunk swizzle applies unchanged),
+//     2 LDS buffers x (A 32 KiB + B 32 KiB) = 128 KiB;
+//   - 4 phases per K-tile, each = {ds_read quadrant frags; glds-stage ONE
+//     half-tile; s_barrier; lgkmcnt(0); setprio(1); 16 MFMA; setprio(0);
+//     s_barrier} -- the fine ds_read || glds || MFMA interleave that the
+//     guide measures as THE lever (coarse phase-split = -7..-27%);
+//   - counted vmcnt ONCE per K-tile (never 0 in the main loop): the stage
+//     schedule {P1: A-lo(j+1), P2: B-lo(j+2), P3: A-hi(j+2), P4: B-hi(j+2)}
+//     leaves exactly 3 half-tiles (vmcnt(6)) in flight at each boundary and
+//     guarantees ALL of tile j+1 landed there:
+//       A-lo(j+1) staged at phase 4j (this tile's P1)    <= boundary-3
+//       B-lo(j+1)/A-hi(j+1)/B-hi(j+1) staged in tile j-1 <= boundary-4
+//     (slot-freeness: a half's LDS slot is reused exactly 2 tiles later,
+//      and its last reader phase precedes the re-stage phase: B-lo read
+//      only in P1, A-* in P1/P2, B-hi in P3 -- frags then live in VGPRs.)
+//   - per-wave output unchanged: 128 batch rows x 4 gates x 16 h-cols,
+//     gate-interleaved so the fused cell update stays lane-local; the
+//     epilogue is byte-identical to k_lstm_rec_fwd_mt256's.
+// Requires B % 256 == 0, H % 64 == 0.  Variant bit 512 in the launcher.
+// ---------------------------------------------------------------------------
+template <bool RAW_DEBUG, bool XCD_SWZ>
+__global__ __launch_bounds__(512, 2) void k_lstm_rec_fwd_8p(
+    const __hip_bfloat16* __restrict__ h_prev, long hp_stride,
+    const __hip_bfloat16* __restrict__ w_t,     // (4H, H) row-major = W_h^T
+    const __hip_bfloat16* __restrict__ pre, long pre_stride,   // (B, 4H)
+    const float* __restrict__ c_prev, long cp_stride,
+    __hip_bfloat16* __restrict__ h_out, long ho_stride,
+    float* __restrict__ c_out, long co_stride,
+    __hip_bfloat16* __restrict__ acts, long act_stride,
+    int B, int H)
+{
+    constexpr int BK = 64;          // K-tile depth (2 x 32-deep sub-images)
+    constexpr int OPBYTES = 32768;  // one operand tile: 2 sub-images x 16 KiB
+    extern __shared__ __attribute__((aligned(16))) char smem[];
+
+    const int tid = threadIdx.x;
+    const int wave = tid >> 6;
+    const int lane = tid & 63;
+    const int lr = lane & 15;   // fragment row (A) / col (B, C/D)
+    const int lk = lane >> 4;   // k-group 0..3
+    const int wm = wave & 1;    // M half: rows [wm*128, +128)
+    const int wn = wave >> 1;   // N group: h-cols [wn*16, +16)
+
+    int m_blk = blockIdx.x, n_blk = blockIdx.y;
+    if (XCD_SWZ) {
+        const int nm = gridDim.x, nwg = gridDim.x * gridDim.y;
+        const int bid = blockIdx.x + blockIdx.y * nm;
+        const int q = nwg >> 3, r = nwg & 7, xcd = bid & 7, loc = bid >> 3;
+        const int swz = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q)
+                        + loc;
+        m_blk = swz % nm;
+        n_blk = swz / nm;
+    }
+    const int m_base = m_blk * 256;
+    const long n_col0 = (long)n_blk * 64;
+    const int nsteps = H / BK;
+
+    // Stage ONE half-tile (128 rows x 64 k) of operand A (h) or B (W^T)
+    // into buffer `buf` at K offset k0: 2 glds per thread (one per 32-deep
+    // sub-image).  LDS image per operand: [kb][row][chunk] with 64-B rows
+    // and the swz64 chunk permutation applied on the SOURCE address
+    // (glds writes lane-linear; reads un-swizzle -- rule 21).
+    auto stage_half = [&](int is_b, int hf, int buf, int k0) {
+        char* const base = smem + buf * (2 * OPBYTES) + is_b * OPBYTES;
+        const int r = hf * 128 + (tid >> 2);
+        const int ch = tid & 3;
+        const int chs = (ch ^ swz64(r)) * 8;
+#pragma unroll
+        for (int kb = 0; kb < 2; ++kb) {
+            const void* src;
+            if (!is_b) {
+                src = h_prev + (long)(m_base + r) * hp_stride + (k0 + kb * 32)
+                      + chs;
+            } else {
+                const long grow = (long)(r >> 6) * H + n_col0 + (r & 63);
+                src = w_t + grow * H + (k0 + kb * 32) + chs;
+            }
+            __builtin_amdgcn_global_load_lds(
+                (const __attribute__((address_space(1))) void*)src,
+                (__attribute__((address_space(3))) void*)(
+                    base + kb * 16384 + hf * 8192 + wave * 1024), 16, 0, 0);
+        }
+    };
+
+    f32x4 acc[4][8];
+#pragma unroll
+    for (int g = 0; g < 4; ++g)
+#pragma unroll
+        for (int mf = 0; mf < 8; ++mf) acc[g][mf] = f32x4{0, 0, 0, 0};
+
+    // ---- prologue: tile 0 fully + {B-lo, A-hi, B-hi} of tile 1, counted
+    // drains so tile 0 is landed and exactly 3 half-tiles stay in flight.
+    stage_half(0, 0, 0, 0);              // A-lo(0)
+    stage_half(1, 0, 0, 0);              // B-lo(0)
+    stage_half(0, 1, 0, 0);              // A-hi(0)
+    stage_half(1, 1, 0, 0);              // B-hi(0)
+    if (nsteps > 1) {
+        asm volatile("s_waitcnt vmcnt(4)" ::: "memory");  // pace: 2 landed
+        stage_half(1, 0, 1, BK);         // B-lo(1)
+        stage_half(0, 1, 1, BK);         // A-hi(1)
+        stage_half(1, 1, 1, BK);         // B-hi(1)
+        asm volatile("s_waitcnt vmcnt(6)" ::: "memory");  // tile 0 landed
+    } else {
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    }
+    __builtin_amdgcn_s_barrier();
+
+    bf16x8 wfr[2][2];    // gate pair frags (g within pair, kb) -- reused
+    bf16x8 hlo[4][2];    // wave's m-half rows 0-63  (mi, kb)
+    bf16x8 hhi[4][2];    // wave's m-half rows 64-127
+
+    for (int j = 0; j < nsteps; ++j) {
+        const int buf = j & 1;
+        const char* ab = smem + buf * (2 * OPBYTES);
+        const char* bb = ab + OPBYTES;
+        // ---- P1: W gates 0-1 + h mi 0-3 (12 ds_reads); stage A-lo(j+1)
+#pragma unroll
+        for (int g = 0; g < 2; ++g) {
+            const int gc = g * 64 + wn * 16 + lr;
+            const int co = (lk ^ swz64(gc)) * 16;
+#pragma unroll
+            for (int kb = 0; kb < 2; ++kb)
+                wfr[g][kb] = *(const bf16x8*)(bb + kb * 16384 + gc * 64 + co);
+        }
+#pragma unroll
+        for (int mi = 0; mi < 4; ++mi) {
+            const int row = wm * 128 + mi * 16 + lr;
+            const int co = (lk ^ swz64(row)) * 16;
+#pragma unroll
+            for (int kb = 0; kb < 2; ++kb)
+                hlo[mi][kb] = *(const bf16x8*)(ab + kb * 16384 + row * 64 + co);
+        }
+        if (j + 1 < nsteps) stage_half(0, 0, (j + 1) & 1, (j + 1) * BK);
+        asm volatile("s_waitcnt lgkmcnt(8)" ::: "memory");
+        __builtin_amdgcn_s_barrier();
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+        for (int g = 0; g < 2; ++g)
+#pragma unroll
+            for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+                for (int kb = 0; kb < 2; ++kb)
+                    acc[g][mi] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        wfr[g][kb], hlo[mi][kb], acc[g][mi], 0, 0, 0);
+        __builtin_amdgcn_s_setprio(0);
+        __builtin_amdgcn_s_barrier();
+        // ---- P2: h mi 4-7 (8 ds_reads); stage B-lo(j+2)
+#pragma unroll
+        for (int mi = 0; mi < 4; ++mi) {
+            const int row = wm * 128 + (4 + mi) * 16 + lr;
+            const int co = (lk ^ swz64(row)) * 16;
+#pragma unroll
+            for (int kb = 0; kb < 2; ++kb)
+                hhi[mi][kb] = *(const bf16x8*)(ab + kb * 16384 + row * 64 + co);
+        }
+        if (j + 2 < nsteps) stage_half(1, 0, buf, (j + 2) * BK);
+        __builtin_amdgcn_s_barrier();
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+        for (int g = 0; g < 2; ++g)
+#pragma unroll
+            for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+                for (int kb = 0; kb < 2; ++kb)
+                    acc[g][4 + mi] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        wfr[g][kb], hhi[mi][kb], acc[g][4 + mi], 0, 0, 0);
+        __builtin_amdgcn_s_setprio(0);
+        __builtin_amdgcn_s_barrier();
+        // ---- P3: W gates 2-3 (4 ds_reads); stage A-hi(j+2)
+#pragma unroll
+        for (int g = 0; g < 2; ++g) {
+            const int gc = (2 + g) * 64 + wn * 16 + lr;
+            const int co = (lk ^ swz64(gc)) * 16;
+#pragma unroll
+            for (int kb = 0; kb < 2; ++kb)
+                wfr[g][kb] = *(const bf16x8*)(bb + kb * 16384 + gc * 64 + co);
+        }
+        if (j + 2 < nsteps) stage_half(0, 1, buf, (j + 2) * BK);
+        __builtin_amdgcn_s_barrier();
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+        for (int g = 0; g < 2; ++g)
+#pragma unroll
+            for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+                for (int kb = 0; kb < 2; ++kb)
+                    acc[2 + g][mi] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        wfr[g][kb], hlo[mi][kb], acc[2 + g][mi], 0, 0, 0);
+        __builtin_amdgcn_s_setprio(0);
+        __builtin_amdgcn_s_barrier();
+        // ---- P4: no ds_reads; stage B-hi(j+2); boundary vmcnt
+        if (j + 2 < nsteps) stage_half(1, 1, buf, (j + 2) * BK);
+        __builtin_amdgcn_s_barrier();
+        __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+        for (int g = 0; g < 2; ++g)
+#pragma unroll
+            for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+                for (int kb = 0; kb < 2; ++kb)
+                    acc[2 + g][4 + mi] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        wfr[g][kb], hhi[mi][kb], acc[2 + g][4 + mi], 0, 0, 0);
+        __builtin_amdgcn_s_setprio(0);
+        if (j + 1 < nsteps) {
+            if (j + 2 < nsteps)
+                asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+            else
+                asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        }
+        __builtin_amdgcn_s_barrier();
+    }
+
+    // Fused epilogue (identical to k_lstm_rec_fwd_mt256): lane owns
+    // h-columns [hcol0, hcol0+4) of batch rows (wm*128 + nf*16 + lr).
+    const int hcol0 = (int)n_col0 + wn * 16 + lk * 4;
+#pragma unroll
+    for (int nf = 0; nf < 8; ++nf) {
+        const long row = m_base + wm * 128 + nf * 16 + lr;
+        f32x4 raw[4];
+#pragma unroll
+        for (int g = 0; g < 4; ++g) {
+            const bf16x4 pv = *(const bf16x4*)(
+                pre + row * pre_stride + (long)g * H + hcol0);
+#pragma unroll
+            for (int j = 0; j < 4; ++j)
+                raw[g][j] = acc[g][nf][j] + (float)pv[j];
+        }
+        __hip_bfloat16* ap = acts + row * act_stride + hcol0;
+        if (RAW_DEBUG) {
+#pragma unroll
+            for (int g = 0; g < 4; ++g) {
+                bf16x4 o;
+#pragma unroll
+                for (int j = 0; j < 4; ++j) o[j] = (__bf16)raw[g][j];
+                *(bf16x4*)(ap + (long)g * H) = o;
+            }
+            continue;
+        }
+        const f32x4 cp = *(const f32x4*)(c_prev + row * cp_stride + hcol0);
+        bf16x4 oi, of, og, oo, oh;
+        f32x4 oc;
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+            const float gi = sigmoidf_(raw[0][j]);
+            const float gf = sigmoidf_(raw[1][j]);
+            const float gg = tanhf_(raw[2][j]);
+            const float go = sigmoidf_(raw[3][j]);
+            const float c  = gf * cp[j] + gi * gg;
+            oc[j] = c;
+            oh[j] = (__bf16)(go * tanhf_(c));
+            oi[j] = (__bf16)gi;
+            of[j] = (__bf16)gf;
+            og[j] = (__bf16)gg;
+            oo[j] = (__bf16)go;
+        }
+        *(bf16x4*)(h_out + row * ho_stride + hcol0) = oh;
+        *(f32x4*)(c_out + row * co_stride + hcol0) = oc;
+        *(bf16x4*)(ap)               = oi;
+        *(bf16x4*)(ap + (long)H)     = of;
+        *(bf16x4*)(ap + (long)2 * H) = og;
+        *(bf16x4*)(ap + (long)3 * H) = oo;
+    }
+}
+
+// ---------------------------------------------------------------------------
 // PERSISTENT whole-sequence forward: ONE kernel for the entire T-step loop.
 //
 // The recurrent weight slice stays RESIDENT in LDS for all T timesteps and
@@ -1456,6 +1728,31 @@ void launch_lstm_rec_fwd_v(
     void* acts, long act_stride, int B, int H, int raw_debug, int variant,
     hipStream_t stream)
 {
+    // 8-phase path (bit9): 256x256 tile, BK=64 counted-vmcnt glds pipeline
+    // (guide §5 template), requires B % 256 == 0.
+    if ((variant & 512) && (B % 256 == 0)) {
+        const size_t smem = 131072;  // 2 buffers x (A 32K + B 32K)
+        dim3 grid(B / 256, H / 64), block(512);
+#define P8LAUNCH(RAW_, SW_)                                                   \
+    do {                                                                      \
+        hipFuncSetAttribute((const void*)(k_lstm_rec_fwd_8p<RAW_, SW_>),      \
+                            hipFuncAttributeMaxDynamicSharedMemorySize,       \
+                            (int)smem);                                       \
+        hipLaunchKernelGGL((k_lstm_rec_fwd_8p<RAW_, SW_>), grid, block,       \
+                           smem, stream,                                      \
+            (const __hip_bfloat16*)h_prev, hp_stride,                         \
+            (const __hip_bfloat16*)w_t, (const __hip_bfloat16*)pre,           \
+            pre_stride, c_prev, cp_stride, (__hip_bfloat16*)h_out, ho_stride, \
+            c_out, co_stride, (__hip_bfloat16*)acts, act_stride, B, H);       \
+    } while (0)
+        if (raw_debug) {
+            if (variant & 8) P8LAUNCH(true, true); else P8LAUNCH(true, false);
+        } else {
+            if (variant & 8) P8LAUNCH(false, true); else P8LAUNCH(false, false);
+        }
+#undef P8LAUNCH
+        return;
+    }
     // MT256 path (bit5): 256x256 tile, 512 threads, requires B % 256 == 0.
     if ((variant & 32) && (B % 256 == 0)) {
         const size_t smem = (variant & 1)

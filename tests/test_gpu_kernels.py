@@ -63,6 +63,32 @@ def test_rec_fwd_step_full(ext, B, H):
     assert (acts.float() - acts_ref).abs().max().item() < 0.02
 
 
+@pytest.mark.parametrize("variant", [512, 520])
+@pytest.mark.parametrize("B,H", [(256, 64), (256, 128), (256, 256),
+                                 (512, 1024), (1024, 1024)])
+def test_rec_fwd_8phase_variants(ext, variant, B, H):
+    """The 8-phase BK=64 counted-vmcnt kernel (k_lstm_rec_fwd_8p) vs the
+    fp32 reference: full fused step at the nsteps = 1/2/4/16 pipeline
+    edges (prologue/epilogue drain paths), with and without XCD swizzle."""
+    h_prev, w_h, pre, c_prev = make_step_inputs(B, H, seed=4)
+    prev = ext.get_fwd_variant()
+    try:
+        ext.set_fwd_variant(variant)
+        h_out, c_out, acts = ext.lstm_rec_fwd_step(h_prev, w_h, pre, c_prev,
+                                                   False)
+    finally:
+        ext.set_fwd_variant(prev)
+    raw = pre.float() + h_prev.float() @ w_h.float()
+    i, f, g, o = raw.chunk(4, dim=1)
+    i, f, g, o = i.sigmoid(), f.sigmoid(), g.tanh(), o.sigmoid()
+    c_ref = f * c_prev + i * g
+    h_ref = o * torch.tanh(c_ref)
+    assert (c_out - c_ref).abs().max().item() < 0.03
+    assert (h_out.float() - h_ref).abs().max().item() < 0.02
+    acts_ref = torch.cat([i, f, g, o], dim=1)
+    assert (acts.float() - acts_ref).abs().max().item() < 0.02
+
+
 # ---------------------------------------------------------------- sequence
 def seq_inputs(B, T, H, seed=0, dev="cuda"):
     g = torch.Generator(device="cpu").manual_seed(seed)

@@ -71,7 +71,8 @@ def main():
     for v in [0, 1, 2, 3, 4, 5, 6, 7, 8, 9, 10, 11, 12, 13, 14, 15,
               18, 19, 26, 27,   # 16+x = BM128
               32, 40,           # MT256 3buf-BK32 (8=XCD swizzle)
-              33, 41]:          # MT256 2buf-BK64
+              33, 41,           # MT256 2buf-BK64
+              512, 520]:        # 8-phase BK64 counted-vmcnt (8=XCD swizzle)
         ext.set_fwd_variant(v)
         h_out, c_out, acts_out = ext.lstm_rec_fwd_step(h_prev, w_h, pre1,
                                                        c_prev, False)

@@ -44,11 +44,18 @@ def probe(B, H, variant):
 
 
 if __name__ == "__main__":
-    for v in [2, 32]:
-        probe(256, 64, v)
-        probe(256, 256, v)
-        probe(512, 64, v)
-        probe(4096, 1024, v)
+    variants = [int(v) for v in sys.argv[1].split(",")] if len(sys.argv) > 1 \
+        else [2, 32, 512, 520]
+    reps = int(sys.argv[2]) if len(sys.argv) > 2 else 1
+    # multi-run race screen across shapes (sync-structure edits are NEW
+    # templates per the guide's two-lane discipline): nsteps = 1, 2, 4, 16
+    for rep in range(reps):
+        for v in variants:
+            probe(256, 64, v)
+            probe(256, 128, v)
+            probe(256, 256, v)
+            probe(512, 64, v)
+            probe(4096, 1024, v)
 
 
 def dump(B, H, variant, path):
