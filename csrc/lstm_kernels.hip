@@ -741,7 +741,11 @@ epilogue:
 //     epilogue is byte-identical to k_lstm_rec_fwd_mt256's.
 // Requires B % 256 == 0, H % 64 == 0.  Variant bit 512 in the launcher.
 // ---------------------------------------------------------------------------
-template <bool RAW_DEBUG, bool XCD_SWZ>
+// ABL bitflags (µbench only): 1 = no staging/vmcnt in the main loop (reads
+// stale buffer 0 -- WRONG results; isolates compute+barrier+read cost),
+// 2 = staging but no MFMA (isolates the glds/vmcnt pipeline cost),
+// 4 = no setprio around the MFMA clusters.
+template <bool RAW_DEBUG, bool XCD_SWZ, int ABL = 0>
 __global__ __launch_bounds__(512, 2) void k_lstm_rec_fwd_8p(
     const __hip_bfloat16* __restrict__ h_prev, long hp_stride,
     const __hip_bfloat16* __restrict__ w_t,     // (4H, H) row-major = W_h^T
@@ -817,7 +821,7 @@ __global__ __launch_bounds__(512, 2) void k_lstm_rec_fwd_8p(
     stage_half(1, 0, 0, 0);              // B-lo(0)
     stage_half(0, 1, 0, 0);              // A-hi(0)
     stage_half(1, 1, 0, 0);              // B-hi(0)
-    if (nsteps > 1) {
+    if (nsteps > 1 && !(ABL & 1)) {
         asm volatile("s_waitcnt vmcnt(4)" ::: "memory");  // pace: 2 landed
         stage_half(1, 0, 1, BK);         // B-lo(1)
         stage_half(0, 1, 1, BK);         // A-hi(1)
@@ -833,7 +837,7 @@ __global__ __launch_bounds__(512, 2) void k_lstm_rec_fwd_8p(
     bf16x8 hhi[4][2];    // wave's m-half rows 64-127
 
     for (int j = 0; j < nsteps; ++j) {
-        const int buf = j & 1;
+        const int buf = (ABL & 1) ? 0 : (j & 1);
         const char* ab = smem + buf * (2 * OPBYTES);
         const char* bb = ab + OPBYTES;
         // ---- P1: W gates 0-1 + h mi 0-3 (12 ds_reads); stage A-lo(j+1)
@@ -853,11 +857,12 @@ __global__ __launch_bounds__(512, 2) void k_lstm_rec_fwd_8p(
             for (int kb = 0; kb < 2; ++kb)
                 hlo[mi][kb] = *(const bf16x8*)(ab + kb * 16384 + row * 64 + co);
         }
-        if (j + 1 < nsteps) stage_half(0, 0, (j + 1) & 1, (j + 1) * BK);
+        if (!(ABL & 1) && j + 1 < nsteps) stage_half(0, 0, (j + 1) & 1, (j + 1) * BK);
         asm volatile("s_waitcnt lgkmcnt(8)" ::: "memory");
         __builtin_amdgcn_s_barrier();
         asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-        __builtin_amdgcn_s_setprio(1);
+        if (!(ABL & 4)) __builtin_amdgcn_s_setprio(1);
+if constexpr (!(ABL & 2)) {
 #pragma unroll
         for (int g = 0; g < 2; ++g)
 #pragma unroll
@@ -866,7 +871,14 @@ __global__ __launch_bounds__(512, 2) void k_lstm_rec_fwd_8p(
                 for (int kb = 0; kb < 2; ++kb)
                     acc[g][mi] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                         wfr[g][kb], hlo[mi][kb], acc[g][mi], 0, 0, 0);
-        __builtin_amdgcn_s_setprio(0);
+        } else {
+            // keep the phase's ds_reads live (guide rule 17)
+            for (int q = 0; q < 2; ++q) {
+                asm volatile("" :: "v"(wfr[q][0]), "v"(wfr[q][1]));
+                asm volatile("" :: "v"(hlo[q][0]), "v"(hhi[q][0]));
+            }
+        }
+        if (!(ABL & 4)) __builtin_amdgcn_s_setprio(0);
         __builtin_amdgcn_s_barrier();
         // ---- P2: h mi 4-7 (8 ds_reads); stage B-lo(j+2)
 #pragma unroll
@@ -877,10 +889,11 @@ __global__ __launch_bounds__(512, 2) void k_lstm_rec_fwd_8p(
             for (int kb = 0; kb < 2; ++kb)
                 hhi[mi][kb] = *(const bf16x8*)(ab + kb * 16384 + row * 64 + co);
         }
-        if (j + 2 < nsteps) stage_half(1, 0, buf, (j + 2) * BK);
+        if (!(ABL & 1) && j + 2 < nsteps) stage_half(1, 0, buf, (j + 2) * BK);
         __builtin_amdgcn_s_barrier();
         asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-        __builtin_amdgcn_s_setprio(1);
+        if (!(ABL & 4)) __builtin_amdgcn_s_setprio(1);
+if constexpr (!(ABL & 2)) {
 #pragma unroll
         for (int g = 0; g < 2; ++g)
 #pragma unroll
@@ -889,7 +902,14 @@ __global__ __launch_bounds__(512, 2) void k_lstm_rec_fwd_8p(
                 for (int kb = 0; kb < 2; ++kb)
                     acc[g][4 + mi] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                         wfr[g][kb], hhi[mi][kb], acc[g][4 + mi], 0, 0, 0);
-        __builtin_amdgcn_s_setprio(0);
+        } else {
+            // keep the phase's ds_reads live (guide rule 17)
+            for (int q = 0; q < 2; ++q) {
+                asm volatile("" :: "v"(wfr[q][0]), "v"(wfr[q][1]));
+                asm volatile("" :: "v"(hlo[q][0]), "v"(hhi[q][0]));
+            }
+        }
+        if (!(ABL & 4)) __builtin_amdgcn_s_setprio(0);
         __builtin_amdgcn_s_barrier();
         // ---- P3: W gates 2-3 (4 ds_reads); stage A-hi(j+2)
 #pragma unroll
@@ -900,10 +920,11 @@ __global__ __launch_bounds__(512, 2) void k_lstm_rec_fwd_8p(
             for (int kb = 0; kb < 2; ++kb)
                 wfr[g][kb] = *(const bf16x8*)(bb + kb * 16384 + gc * 64 + co);
         }
-        if (j + 2 < nsteps) stage_half(0, 1, buf, (j + 2) * BK);
+        if (!(ABL & 1) && j + 2 < nsteps) stage_half(0, 1, buf, (j + 2) * BK);
         __builtin_amdgcn_s_barrier();
         asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-        __builtin_amdgcn_s_setprio(1);
+        if (!(ABL & 4)) __builtin_amdgcn_s_setprio(1);
+if constexpr (!(ABL & 2)) {
 #pragma unroll
         for (int g = 0; g < 2; ++g)
 #pragma unroll
@@ -912,12 +933,20 @@ __global__ __launch_bounds__(512, 2) void k_lstm_rec_fwd_8p(
                 for (int kb = 0; kb < 2; ++kb)
                     acc[2 + g][mi] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                         wfr[g][kb], hlo[mi][kb], acc[2 + g][mi], 0, 0, 0);
-        __builtin_amdgcn_s_setprio(0);
+        } else {
+            // keep the phase's ds_reads live (guide rule 17)
+            for (int q = 0; q < 2; ++q) {
+                asm volatile("" :: "v"(wfr[q][0]), "v"(wfr[q][1]));
+                asm volatile("" :: "v"(hlo[q][0]), "v"(hhi[q][0]));
+            }
+        }
+        if (!(ABL & 4)) __builtin_amdgcn_s_setprio(0);
         __builtin_amdgcn_s_barrier();
         // ---- P4: no ds_reads; stage B-hi(j+2); boundary vmcnt
-        if (j + 2 < nsteps) stage_half(1, 1, buf, (j + 2) * BK);
+        if (!(ABL & 1) && j + 2 < nsteps) stage_half(1, 1, buf, (j + 2) * BK);
         __builtin_amdgcn_s_barrier();
-        __builtin_amdgcn_s_setprio(1);
+        if (!(ABL & 4)) __builtin_amdgcn_s_setprio(1);
+if constexpr (!(ABL & 2)) {
 #pragma unroll
         for (int g = 0; g < 2; ++g)
 #pragma unroll
@@ -926,8 +955,15 @@ __global__ __launch_bounds__(512, 2) void k_lstm_rec_fwd_8p(
                 for (int kb = 0; kb < 2; ++kb)
                     acc[2 + g][4 + mi] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                         wfr[g][kb], hhi[mi][kb], acc[2 + g][4 + mi], 0, 0, 0);
-        __builtin_amdgcn_s_setprio(0);
-        if (j + 1 < nsteps) {
+        } else {
+            // keep the phase's ds_reads live (guide rule 17)
+            for (int q = 0; q < 2; ++q) {
+                asm volatile("" :: "v"(wfr[q][0]), "v"(wfr[q][1]));
+                asm volatile("" :: "v"(hlo[q][0]), "v"(hhi[q][0]));
+            }
+        }
+        if (!(ABL & 4)) __builtin_amdgcn_s_setprio(0);
+        if (!(ABL & 1) && j + 1 < nsteps) {
             if (j + 2 < nsteps)
                 asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
             else
@@ -1733,22 +1769,31 @@ void launch_lstm_rec_fwd_v(
     if ((variant & 512) && (B % 256 == 0)) {
         const size_t smem = 131072;  // 2 buffers x (A 32K + B 32K)
         dim3 grid(B / 256, H / 64), block(512);
-#define P8LAUNCH(RAW_, SW_)                                                   \
+#define P8LAUNCH(RAW_, SW_, ABL_)                                             \
     do {                                                                      \
-        hipFuncSetAttribute((const void*)(k_lstm_rec_fwd_8p<RAW_, SW_>),      \
+        hipFuncSetAttribute((const void*)(k_lstm_rec_fwd_8p<RAW_, SW_, ABL_>),\
                             hipFuncAttributeMaxDynamicSharedMemorySize,       \
                             (int)smem);                                       \
-        hipLaunchKernelGGL((k_lstm_rec_fwd_8p<RAW_, SW_>), grid, block,       \
+        hipLaunchKernelGGL((k_lstm_rec_fwd_8p<RAW_, SW_, ABL_>), grid, block, \
                            smem, stream,                                      \
             (const __hip_bfloat16*)h_prev, hp_stride,                         \
             (const __hip_bfloat16*)w_t, (const __hip_bfloat16*)pre,           \
             pre_stride, c_prev, cp_stride, (__hip_bfloat16*)h_out, ho_stride, \
             c_out, co_stride, (__hip_bfloat16*)acts, act_stride, B, H);       \
     } while (0)
+        if (variant & 64) { P8LAUNCH(true, false, 1); return; }   // no-stage
+        if (variant & 128) { P8LAUNCH(true, false, 2); return; }  // no-MFMA
+        if (variant & 4) {                                        // no-setprio
+            if (raw_debug) P8LAUNCH(true, false, 4);
+            else P8LAUNCH(false, false, 4);
+            return;
+        }
         if (raw_debug) {
-            if (variant & 8) P8LAUNCH(true, true); else P8LAUNCH(true, false);
+            if (variant & 8) P8LAUNCH(true, true, 0);
+            else P8LAUNCH(true, false, 0);
         } else {
-            if (variant & 8) P8LAUNCH(false, true); else P8LAUNCH(false, false);
+            if (variant & 8) P8LAUNCH(false, true, 0);
+            else P8LAUNCH(false, false, 0);
         }
 #undef P8LAUNCH
         return;
