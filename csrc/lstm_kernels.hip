@@ -830,7 +830,7 @@ __global__ __launch_bounds__(512, 2) void k_lstm_rec_fwd_8p(
     } else {
         asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     }
-    __builtin_amdgcn_s_barrier();
+    if (!(ABL & 8)) __builtin_amdgcn_s_barrier();
 
     bf16x8 wfr[2][2];    // gate pair frags (g within pair, kb) -- reused
     bf16x8 hlo[4][2];    // wave's m-half rows 0-63  (mi, kb)
@@ -859,7 +859,7 @@ __global__ __launch_bounds__(512, 2) void k_lstm_rec_fwd_8p(
         }
         if (!(ABL & 1) && j + 1 < nsteps) stage_half(0, 0, (j + 1) & 1, (j + 1) * BK);
         asm volatile("s_waitcnt lgkmcnt(8)" ::: "memory");
-        __builtin_amdgcn_s_barrier();
+        if (!(ABL & 8)) __builtin_amdgcn_s_barrier();
         asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
         if (!(ABL & 4)) __builtin_amdgcn_s_setprio(1);
 if constexpr (!(ABL & 2)) {
@@ -879,7 +879,7 @@ if constexpr (!(ABL & 2)) {
             }
         }
         if (!(ABL & 4)) __builtin_amdgcn_s_setprio(0);
-        __builtin_amdgcn_s_barrier();
+        if (!(ABL & 8)) __builtin_amdgcn_s_barrier();
         // ---- P2: h mi 4-7 (8 ds_reads); stage B-lo(j+2)
 #pragma unroll
         for (int mi = 0; mi < 4; ++mi) {
@@ -890,7 +890,7 @@ if constexpr (!(ABL & 2)) {
                 hhi[mi][kb] = *(const bf16x8*)(ab + kb * 16384 + row * 64 + co);
         }
         if (!(ABL & 1) && j + 2 < nsteps) stage_half(1, 0, buf, (j + 2) * BK);
-        __builtin_amdgcn_s_barrier();
+        if (!(ABL & 8)) __builtin_amdgcn_s_barrier();
         asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
         if (!(ABL & 4)) __builtin_amdgcn_s_setprio(1);
 if constexpr (!(ABL & 2)) {
@@ -910,7 +910,7 @@ if constexpr (!(ABL & 2)) {
             }
         }
         if (!(ABL & 4)) __builtin_amdgcn_s_setprio(0);
-        __builtin_amdgcn_s_barrier();
+        if (!(ABL & 8)) __builtin_amdgcn_s_barrier();
         // ---- P3: W gates 2-3 (4 ds_reads); stage A-hi(j+2)
 #pragma unroll
         for (int g = 0; g < 2; ++g) {
@@ -921,7 +921,7 @@ if constexpr (!(ABL & 2)) {
                 wfr[g][kb] = *(const bf16x8*)(bb + kb * 16384 + gc * 64 + co);
         }
         if (!(ABL & 1) && j + 2 < nsteps) stage_half(0, 1, buf, (j + 2) * BK);
-        __builtin_amdgcn_s_barrier();
+        if (!(ABL & 8)) __builtin_amdgcn_s_barrier();
         asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
         if (!(ABL & 4)) __builtin_amdgcn_s_setprio(1);
 if constexpr (!(ABL & 2)) {
@@ -941,10 +941,10 @@ if constexpr (!(ABL & 2)) {
             }
         }
         if (!(ABL & 4)) __builtin_amdgcn_s_setprio(0);
-        __builtin_amdgcn_s_barrier();
+        if (!(ABL & 8)) __builtin_amdgcn_s_barrier();
         // ---- P4: no ds_reads; stage B-hi(j+2); boundary vmcnt
         if (!(ABL & 1) && j + 2 < nsteps) stage_half(1, 1, buf, (j + 2) * BK);
-        __builtin_amdgcn_s_barrier();
+        if (!(ABL & 8)) __builtin_amdgcn_s_barrier();
         if (!(ABL & 4)) __builtin_amdgcn_s_setprio(1);
 if constexpr (!(ABL & 2)) {
 #pragma unroll
@@ -969,7 +969,7 @@ if constexpr (!(ABL & 2)) {
             else
                 asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
         }
-        __builtin_amdgcn_s_barrier();
+        if (!(ABL & 8)) __builtin_amdgcn_s_barrier();
     }
 
     // Fused epilogue (identical to k_lstm_rec_fwd_mt256): lane owns
@@ -1781,8 +1781,10 @@ void launch_lstm_rec_fwd_v(
             pre_stride, c_prev, cp_stride, (__hip_bfloat16*)h_out, ho_stride, \
             c_out, co_stride, (__hip_bfloat16*)acts, act_stride, B, H);       \
     } while (0)
+        if ((variant & 192) == 192) { P8LAUNCH(true, false, 3); return; }
         if (variant & 64) { P8LAUNCH(true, false, 1); return; }   // no-stage
         if (variant & 128) { P8LAUNCH(true, false, 2); return; }  // no-MFMA
+        if (variant & 256) { P8LAUNCH(true, false, 8); return; }  // no-barrier
         if (variant & 4) {                                        // no-setprio
             if (raw_debug) P8LAUNCH(true, false, 4);
             else P8LAUNCH(false, false, 4);

@@ -165,3 +165,25 @@ def test_autotune_cpu_noop():
 
     r = tune_fwd(64, 64)
     assert r["choice"] == "heuristic"
+
+
+def test_rccl_sweep_tool_degenerate(tmp_path):
+    """tools/rccl_sweep.py must run at world_size=1 (no PG) so it is one
+    torchrun invocation away from the real sweep."""
+    import json
+    from tools.rccl_sweep import main
+    out = str(tmp_path / "sweep.jsonl")
+    main(["--buckets", "4", "--batch", "2", "--steps", "1",
+          "--hidden", "32", "--seq-len", "8", "--out", out])
+    rows = [json.loads(l) for l in open(out)]
+    assert any(r["kind"] == "step" for r in rows)
+
+
+def test_overlap_probe_tool_degenerate(tmp_path):
+    """tools/overlap_probe.py world_size=1: records bucket-ready times and
+    the overlapped-vs-deferred A/B."""
+    from tools.overlap_probe import main
+    out = str(tmp_path / "ov.jsonl")
+    r = main(["--batch", "2", "--steps", "3", "--out", out])
+    assert r["per_step"]["buckets"], "no bucket-ready timestamps recorded"
+    assert r["per_step"]["bwd_ms"] > 0

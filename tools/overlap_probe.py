@@ -44,13 +44,17 @@ class TimingBucketer:
         self.ready_events = {}
         self.ready_wall = {}
         orig = bucketer._on_grad_ready
+        # own countdown: at world_size=1 the bucketer's hook early-returns
+        # without touching .pending, so readiness is tracked here
+        self._left = {}
 
         def hook(p):
             bucket = bucketer._param_bucket[p]
-            was = bucket.pending
+            i = bucketer.buckets.index(bucket)
+            self._left[i] = self._left.get(i, len(bucket.params)) - 1
             orig(p)
-            if was == 1:  # this grad completed the bucket
-                i = bucketer.buckets.index(bucket)
+            if self._left[i] == 0:  # this grad completed the bucket
+                self._left[i] = len(bucket.params)
                 if self.on_gpu:
                     ev = torch.cuda.Event(enable_timing=True)
                     ev.record()

@@ -65,11 +65,13 @@ def bench_allreduce(ctx, sizes_mb, iters=20, dtype=torch.bfloat16):
     return rows
 
 
-def bench_step(ctx, bucket_mb, reduce_dtype, batch, steps=8, warmup=3):
+def bench_step(ctx, bucket_mb, reduce_dtype, batch, steps=8, warmup=3,
+               hidden=1024, seq_len=128):
     cfg = Config(
-        model=ModelConfig(input_size=1024, hidden_size=1024, num_layers=2),
-        train=TrainConfig(batch_size=batch, seq_len=128, lr=0.01,
-                          steps=steps, log_interval=0, dtype="bf16"),
+        model=ModelConfig(input_size=hidden, hidden_size=hidden, num_layers=2),
+        train=TrainConfig(batch_size=batch, seq_len=seq_len, lr=0.01,
+                          steps=steps, log_interval=0,
+                          dtype="bf16" if ctx.device.type == "cuda" else "fp32"),
         data=DataConfig(task="random"),
     )
     cfg.dist.bucket_mb = bucket_mb
@@ -107,6 +109,8 @@ def main(argv=None):
                     help="per-GPU batch (default: 16384 GPU / 8 CPU)")
     ap.add_argument("--sizes-mb", default="1,4,8,16,32,64")
     ap.add_argument("--steps", type=int, default=8)
+    ap.add_argument("--hidden", type=int, default=1024)
+    ap.add_argument("--seq-len", type=int, default=128)
     ap.add_argument("--out", default="")
     args = ap.parse_args(argv)
 
@@ -118,7 +122,8 @@ def main(argv=None):
         rows += bench_allreduce(ctx, [int(s) for s in args.sizes_mb.split(",")])
     for bmb in [int(b) for b in args.buckets.split(",")]:
         for rd in ("", "fp32"):
-            rows.append(bench_step(ctx, bmb, rd, batch, steps=args.steps))
+            rows.append(bench_step(ctx, bmb, rd, batch, steps=args.steps,
+                                   hidden=args.hidden, seq_len=args.seq_len))
             if ctx.is_main:
                 print(json.dumps(rows[-1]), flush=True)
     if ctx.is_main:
