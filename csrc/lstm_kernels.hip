@@ -1025,6 +1025,224 @@ if constexpr (!(ABL & 2)) {
 }
 
 // ---------------------------------------------------------------------------
+// 2-PHASE merged variant of the 8-phase kernel: same 256x256 BK=64 tile and
+// LDS images, but TWO phases per K-tile of 32 MFMA each -- half the barrier
+// crossings, double-length matrix segments (the 4-phase ablation showed the
+// barrier+read scaffold, not MFMA, dominating).  No setprio (measured
+// neutral-to-negative on this GEMM shape, guide T5 "NOT GEMM").
+//   Ph1: ds_read A mi0-7 (16) + W g01 (4); stage {A-lo(j+1), B-hi(j+1)};
+//        barrier; lgkmcnt(0); 32 MFMA (g01 x mi0-7 x kb); barrier.
+//   Ph2: ds_read W g23 (4); stage {B-lo(j+2), A-hi(j+2)};
+//        barrier; lgkmcnt(0); 32 MFMA (g23); boundary vmcnt(4); barrier.
+// Slot-freeness: B-hi(j+1) stages in Ph1(j) because its slot (B-hi(j-1))
+// was last read in Ph2(j-1); B-lo/A-hi(j+2) stage in Ph2(j) after their
+// slots' last reads in Ph1(j).  Boundary vmcnt(4) leaves exactly Ph2(j)'s
+// two stages in flight and guarantees ALL of tile j+1 landed.
+// ---------------------------------------------------------------------------
+template <bool RAW_DEBUG, bool XCD_SWZ>
+__global__ __launch_bounds__(512, 2) void k_lstm_rec_fwd_2ph(
+    const __hip_bfloat16* __restrict__ h_prev, long hp_stride,
+    const __hip_bfloat16* __restrict__ w_t,     // (4H, H) row-major = W_h^T
+    const __hip_bfloat16* __restrict__ pre, long pre_stride,   // (B, 4H)
+    const float* __restrict__ c_prev, long cp_stride,
+    __hip_bfloat16* __restrict__ h_out, long ho_stride,
+    float* __restrict__ c_out, long co_stride,
+    __hip_bfloat16* __restrict__ acts, long act_stride,
+    int B, int H)
+{
+    constexpr int BK = 64;
+    constexpr int OPBYTES = 32768;
+    extern __shared__ __attribute__((aligned(16))) char smem[];
+
+    const int tid = threadIdx.x;
+    const int wave = tid >> 6;
+    const int lane = tid & 63;
+    const int lr = lane & 15;
+    const int lk = lane >> 4;
+    const int wm = wave & 1;
+    const int wn = wave >> 1;
+
+    int m_blk = blockIdx.x, n_blk = blockIdx.y;
+    if (XCD_SWZ) {
+        const int nm = gridDim.x, nwg = gridDim.x * gridDim.y;
+        const int bid = blockIdx.x + blockIdx.y * nm;
+        const int q = nwg >> 3, r = nwg & 7, xcd = bid & 7, loc = bid >> 3;
+        const int swz = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q)
+                        + loc;
+        m_blk = swz % nm;
+        n_blk = swz / nm;
+    }
+    const int m_base = m_blk * 256;
+    const long n_col0 = (long)n_blk * 64;
+    const int nsteps = H / BK;
+
+    auto stage_half = [&](int is_b, int hf, int buf, int k0) {
+        char* const base = smem + buf * (2 * OPBYTES) + is_b * OPBYTES;
+        const int r = hf * 128 + (tid >> 2);
+        const int ch = tid & 3;
+        const int chs = (ch ^ swz64(r)) * 8;
+#pragma unroll
+        for (int kb = 0; kb < 2; ++kb) {
+            const void* src;
+            if (!is_b) {
+                src = h_prev + (long)(m_base + r) * hp_stride + (k0 + kb * 32)
+                      + chs;
+            } else {
+                const long grow = (long)(r >> 6) * H + n_col0 + (r & 63);
+                src = w_t + grow * H + (k0 + kb * 32) + chs;
+            }
+            __builtin_amdgcn_global_load_lds(
+                (const __attribute__((address_space(1))) void*)src,
+                (__attribute__((address_space(3))) void*)(
+                    base + kb * 16384 + hf * 8192 + wave * 1024), 16, 0, 0);
+        }
+    };
+
+    f32x4 acc[4][8];
+#pragma unroll
+    for (int g = 0; g < 4; ++g)
+#pragma unroll
+        for (int mf = 0; mf < 8; ++mf) acc[g][mf] = f32x4{0, 0, 0, 0};
+
+    // prologue: tile 0 fully + Ph2(-1)'s {B-lo(1), A-hi(1)}
+    stage_half(0, 0, 0, 0);
+    stage_half(1, 1, 0, 0);
+    stage_half(1, 0, 0, 0);
+    stage_half(0, 1, 0, 0);
+    if (nsteps > 1) {
+        asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+        stage_half(1, 0, 1, BK);         // B-lo(1)
+        stage_half(0, 1, 1, BK);         // A-hi(1)
+        asm volatile("s_waitcnt vmcnt(4)" ::: "memory");  // tile 0 landed
+    } else {
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    }
+    __builtin_amdgcn_s_barrier();
+
+    bf16x8 wfr[2][2];
+    bf16x8 hfr[8][2];
+
+    for (int j = 0; j < nsteps; ++j) {
+        const int buf = j & 1;
+        const char* ab = smem + buf * (2 * OPBYTES);
+        const char* bb = ab + OPBYTES;
+        // ---- Ph1: A mi0-7 + W g01 (20 ds_reads); stage A-lo/B-hi(j+1)
+#pragma unroll
+        for (int g = 0; g < 2; ++g) {
+            const int gc = g * 64 + wn * 16 + lr;
+            const int co = (lk ^ swz64(gc)) * 16;
+#pragma unroll
+            for (int kb = 0; kb < 2; ++kb)
+                wfr[g][kb] = *(const bf16x8*)(bb + kb * 16384 + gc * 64 + co);
+        }
+#pragma unroll
+        for (int mi = 0; mi < 8; ++mi) {
+            const int row = wm * 128 + mi * 16 + lr;
+            const int co = (lk ^ swz64(row)) * 16;
+#pragma unroll
+            for (int kb = 0; kb < 2; ++kb)
+                hfr[mi][kb] = *(const bf16x8*)(ab + kb * 16384 + row * 64 + co);
+        }
+        if (j + 1 < nsteps) {
+            stage_half(0, 0, (j + 1) & 1, (j + 1) * BK);  // A-lo(j+1)
+            stage_half(1, 1, (j + 1) & 1, (j + 1) * BK);  // B-hi(j+1)
+        }
+        asm volatile("s_waitcnt lgkmcnt(8)" ::: "memory");
+        __builtin_amdgcn_s_barrier();
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+#pragma unroll
+        for (int g = 0; g < 2; ++g)
+#pragma unroll
+            for (int mi = 0; mi < 8; ++mi)
+#pragma unroll
+                for (int kb = 0; kb < 2; ++kb)
+                    acc[g][mi] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        wfr[g][kb], hfr[mi][kb], acc[g][mi], 0, 0, 0);
+        __builtin_amdgcn_s_barrier();
+        // ---- Ph2: W g23 (4 ds_reads); stage B-lo/A-hi(j+2)
+#pragma unroll
+        for (int g = 0; g < 2; ++g) {
+            const int gc = (2 + g) * 64 + wn * 16 + lr;
+            const int co = (lk ^ swz64(gc)) * 16;
+#pragma unroll
+            for (int kb = 0; kb < 2; ++kb)
+                wfr[g][kb] = *(const bf16x8*)(bb + kb * 16384 + gc * 64 + co);
+        }
+        if (j + 2 < nsteps) {
+            stage_half(1, 0, buf, (j + 2) * BK);  // B-lo(j+2)
+            stage_half(0, 1, buf, (j + 2) * BK);  // A-hi(j+2)
+        }
+        __builtin_amdgcn_s_barrier();
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+#pragma unroll
+        for (int g = 0; g < 2; ++g)
+#pragma unroll
+            for (int mi = 0; mi < 8; ++mi)
+#pragma unroll
+                for (int kb = 0; kb < 2; ++kb)
+                    acc[2 + g][mi] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        wfr[g][kb], hfr[mi][kb], acc[2 + g][mi], 0, 0, 0);
+        if (j + 1 < nsteps) {
+            if (j + 2 < nsteps)
+                asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+            else
+                asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        }
+        __builtin_amdgcn_s_barrier();
+    }
+
+    // epilogue identical to k_lstm_rec_fwd_8p
+    const int hcol0 = (int)n_col0 + wn * 16 + lk * 4;
+#pragma unroll
+    for (int nf = 0; nf < 8; ++nf) {
+        const long row = m_base + wm * 128 + nf * 16 + lr;
+        f32x4 raw[4];
+#pragma unroll
+        for (int g = 0; g < 4; ++g) {
+            const bf16x4 pv = *(const bf16x4*)(
+                pre + row * pre_stride + (long)g * H + hcol0);
+#pragma unroll
+            for (int j = 0; j < 4; ++j)
+                raw[g][j] = acc[g][nf][j] + (float)pv[j];
+        }
+        __hip_bfloat16* ap = acts + row * act_stride + hcol0;
+        if (RAW_DEBUG) {
+#pragma unroll
+            for (int g = 0; g < 4; ++g) {
+                bf16x4 o;
+#pragma unroll
+                for (int j = 0; j < 4; ++j) o[j] = (__bf16)raw[g][j];
+                *(bf16x4*)(ap + (long)g * H) = o;
+            }
+            continue;
+        }
+        const f32x4 cp = *(const f32x4*)(c_prev + row * cp_stride + hcol0);
+        bf16x4 oi, of, og, oo, oh;
+        f32x4 oc;
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+            const float gi = sigmoidf_(raw[0][j]);
+            const float gf = sigmoidf_(raw[1][j]);
+            const float gg = tanhf_(raw[2][j]);
+            const float go = sigmoidf_(raw[3][j]);
+            const float c  = gf * cp[j] + gi * gg;
+            oc[j] = c;
+            oh[j] = (__bf16)(go * tanhf_(c));
+            oi[j] = (__bf16)gi;
+            of[j] = (__bf16)gf;
+            og[j] = (__bf16)gg;
+            oo[j] = (__bf16)go;
+        }
+        *(bf16x4*)(h_out + row * ho_stride + hcol0) = oh;
+        *(f32x4*)(c_out + row * co_stride + hcol0) = oc;
+        *(bf16x4*)(ap)               = oi;
+        *(bf16x4*)(ap + (long)H)     = of;
+        *(bf16x4*)(ap + (long)2 * H) = og;
+        *(bf16x4*)(ap + (long)3 * H) = oo;
+    }
+}
+
+// ---------------------------------------------------------------------------
 // PERSISTENT whole-sequence forward: ONE kernel for the entire T-step loop.
 //
 // The recurrent weight slice stays RESIDENT in LDS for all T timesteps and
@@ -1764,6 +1982,30 @@ void launch_lstm_rec_fwd_v(
     void* acts, long act_stride, int B, int H, int raw_debug, int variant,
     hipStream_t stream)
 {
+    // 2-phase merged path (bit10): 256x256 BK=64, 32-MFMA segments
+    if ((variant & 1024) && (B % 256 == 0)) {
+        const size_t smem = 131072;
+        dim3 grid(B / 256, H / 64), block(512);
+#define P2LAUNCH(RAW_, SW_)                                                   \
+    do {                                                                      \
+        hipFuncSetAttribute((const void*)(k_lstm_rec_fwd_2ph<RAW_, SW_>),     \
+                            hipFuncAttributeMaxDynamicSharedMemorySize,       \
+                            (int)smem);                                       \
+        hipLaunchKernelGGL((k_lstm_rec_fwd_2ph<RAW_, SW_>), grid, block,      \
+                           smem, stream,                                      \
+            (const __hip_bfloat16*)h_prev, hp_stride,                         \
+            (const __hip_bfloat16*)w_t, (const __hip_bfloat16*)pre,           \
+            pre_stride, c_prev, cp_stride, (__hip_bfloat16*)h_out, ho_stride, \
+            c_out, co_stride, (__hip_bfloat16*)acts, act_stride, B, H);       \
+    } while (0)
+        if (raw_debug) {
+            if (variant & 8) P2LAUNCH(true, true); else P2LAUNCH(true, false);
+        } else {
+            if (variant & 8) P2LAUNCH(false, true); else P2LAUNCH(false, false);
+        }
+#undef P2LAUNCH
+        return;
+    }
     // 8-phase path (bit9): 256x256 tile, BK=64 counted-vmcnt glds pipeline
     // (guide §5 template), requires B % 256 == 0.
     if ((variant & 512) && (B % 256 == 0)) {
