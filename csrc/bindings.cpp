@@ -48,6 +48,10 @@ void launch_xent_bwd(const void*, const long*, const float*, void*,
 int launch_lstm_seq_fwd_persistent(void*, const void*, const void*,
                                    const float*, float*, void*,
                                    unsigned*, int, int, int, int, hipStream_t);
+int launch_lstm_dh_pw_bwd(const void*, long, const void*, long, const void*,
+                          long, const void*, long, const float*, long,
+                          const float*, long, void*, long, float*, long,
+                          void*, long, int, int, int, int, hipStream_t);
 }
 
 namespace {
@@ -63,6 +67,9 @@ bool g_persistent_enabled = false;  // per-step glds path is faster (kbench6)
 // fused wins to B=512 (18.7 vs 22.0 us), ~tie at B=1024, split wins from
 // B=2048; at H=2048 split wins from B=1024: 1107 vs 1209 ms/step).
 int g_split_fwd = -1;
+// Fused backward step: -1/1 = on where eligible (B%256==0, H%256==0),
+// 0 = off (split hipBLASLt dh GEMM + k_lstm_pw_bwd).
+int g_fused_bwd = -1;
 // FP8 recurrent GEMM (experimental, MI355X 2x-bf16 MFMA rate): quantize
 // h (tanh-bounded, FIXED scale 448) and W_h (per-call amax scale) to OCP
 // e4m3 and run the per-step GEMM through hipBLASLt _scaled_mm (measured
@@ -263,10 +270,20 @@ std::vector<Tensor> lstm_seq_bwd(c10::optional<Tensor> dh_seq_opt,
     auto dgates = at::empty({T, B, H4}, acts.options());
     auto dc = at::empty({B, H}, acts.options().dtype(at::kFloat));
     auto dh_buf = at::empty({B, H}, acts.options());
-    // W_h^T as a CONTIGUOUS copy (one 8 MB transpose per layer-backward):
-    // hipBLASLt runs the per-step (B,4H)x(4H,H) grad GEMM ~1.4x faster on
-    // the plain-NN layout than on the transposed view (980 vs 670 TF/s).
-    auto w_h_t = w_h.t().contiguous();
+    auto w_hc = w_h.contiguous();
+    // Fused backward step [K5a+K4]: hand MFMA dh GEMM with the pointwise
+    // backward fused into its epilogue (k_lstm_dh_pw_bwd) -- one kernel
+    // per timestep, fp32 dh carry in registers, and no W_h^T transpose
+    // copy (the kernel reads the (H,4H) layout directly).
+    const bool fused = (g_fused_bwd != 0) && (B % 256 == 0) && (H % 256 == 0);
+    Tensor w_h_t;
+    if (!fused) {
+        // split fallback: W_h^T as a CONTIGUOUS copy (one 8 MB transpose
+        // per layer-backward): hipBLASLt runs the per-step (B,4H)x(4H,H)
+        // grad GEMM ~1.4x faster on the plain-NN layout than on the
+        // transposed view (980 vs 670 TF/s).
+        w_h_t = w_h.t().contiguous();
+    }
 
     auto* act_p = (const char*)acts.data_ptr();
     auto* cs_p = (const float*)c_seq.data_ptr();
@@ -283,6 +300,20 @@ std::vector<Tensor> lstm_seq_bwd(c10::optional<Tensor> dh_seq_opt,
             cprev_p = cs_p + t_prev * B * H;
         } else {
             cprev_p = (const float*)c0c.data_ptr();
+        }
+        if (fused && !first) {
+            // dgates of the step handled in the PREVIOUS iteration
+            const long t_nx = reverse ? (t - 1) : (t + 1);
+            int rc = launch_lstm_dh_pw_bwd(
+                dg_p + 2 * t_nx * B * H4, H4, w_hc.data_ptr(), H4,
+                has_up ? up_p + 2 * t * B * H : nullptr, H,
+                act_p + 2 * t * B * H4, H4,
+                cs_p + t * B * H, H, cprev_p, H,
+                dg_p + 2 * t * B * H4, H4,
+                (float*)dc.data_ptr(), H,
+                nullptr, 0, (int)B, (int)H, /*raw=*/0, /*swz=*/1, stream);
+            TORCH_CHECK(rc == 0, "fused bwd launch failed");
+            continue;
         }
         const void* carry_p = nullptr;
         long carry_stride = H;
@@ -301,8 +332,21 @@ std::vector<Tensor> lstm_seq_bwd(c10::optional<Tensor> dh_seq_opt,
             first ? 1 : 0,
             (first && has_cT) ? (const float*)d_cT.data_ptr() : nullptr, H,
             stream);
-        // recurrent hidden-grad GEMM [K5a]: dh_{t-1} = dgates_t @ W_h^T
-        at::mm_out(dh_buf, dgates.select(0, t), w_h_t);
+        if (!fused) {
+            // recurrent hidden-grad GEMM [K5a]: dh_{t-1} = dgates_t @ W_h^T
+            at::mm_out(dh_buf, dgates.select(0, t), w_h_t);
+        }
+    }
+    if (fused) {
+        // chain-end dh0 = dgates_{t_last} @ W_h^T via the RAW fused kernel
+        const long t0 = reverse ? (T - 1) : 0;
+        int rc = launch_lstm_dh_pw_bwd(
+            dg_p + 2 * t0 * B * H4, H4, w_hc.data_ptr(), H4,
+            nullptr, 0, nullptr, 0, nullptr, 0, nullptr, 0,
+            nullptr, 0, nullptr, 0,
+            dh_buf.data_ptr(), H, (int)B, (int)H, /*raw=*/1, /*swz=*/1,
+            stream);
+        TORCH_CHECK(rc == 0, "fused bwd raw launch failed");
     }
     return {dgates, dh_buf, dc};  // (dgates, dh0, dc0)
 }
@@ -524,6 +568,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("set_split_fwd", [](int v) { g_split_fwd = v; },
           "-1 auto (B>=2048), 0 fused hand kernel, 1 hipBLASLt+pointwise");
     m.def("get_split_fwd", []() { return g_split_fwd; });
+    m.def("set_fused_bwd", [](int v) { g_fused_bwd = v; },
+          "fused dh-GEMM+pointwise backward: -1/1 on where eligible, 0 off");
+    m.def("get_fused_bwd", []() { return g_fused_bwd; });
     m.def("get_persistent_fwd", []() { return g_persistent_enabled; });
     m.def("set_fwd_variant", &set_fwd_variant,
           "select rec_fwd kernel variant (bit0 STAGE_A, bit1 XCD_SWZ, bit2 BM32)");

@@ -1243,6 +1243,268 @@ __global__ __launch_bounds__(512, 2) void k_lstm_rec_fwd_2ph(
 }
 
 // ---------------------------------------------------------------------------
+// FUSED BACKWARD STEP [K5a+K4]: dh_t = dgates_{t+1} @ W_h^T on MFMA, with
+// the pointwise backward of step t fused into the epilogue -- one kernel
+// replaces the hipBLASLt dh GEMM + k_lstm_pw_bwd pair per timestep AND
+// keeps the (B,H) dh carry in registers (no bf16 round trip; the carry is
+// fp32 = strictly better numerics than the split path's bf16 dh_buf).
+//
+// GEMM mapping mirrors k_lstm_rec_fwd_2ph exactly (256x256 output tile,
+// BK=64 as two swz64 BK=32 sub-images, 2 phases/tile, counted vmcnt):
+//   - M = batch rows of dgates_{t+1} (the "A image"),
+//   - N = 256 dh columns; the "B image" rows are W_h rows n_col0+gc
+//     ((H, 4H) row-major: k = gate-dim is contiguous -- NO transpose copy),
+//   - K = 4H (64 K-tiles at H=1024).
+// The wave's 4 column-frags are cf*64 + wn*16 (+lr), NOT gates: the
+// epilogue loops cf to cover 4 dh column groups, each lane owning 4
+// consecutive dh columns -> every pw access is bf16x4/f32x4.
+// RAW_DEBUG: write dh only (used for the chain-end dh0 and numerics tests).
+// ---------------------------------------------------------------------------
+template <bool RAW_DEBUG, bool XCD_SWZ>
+__global__ __launch_bounds__(512, 2) void k_lstm_dh_pw_bwd(
+    const __hip_bfloat16* __restrict__ dg_next, long dgn_stride,  // (B,4H)
+    const __hip_bfloat16* __restrict__ w_h, long wh_stride,       // (H,4H)
+    const __hip_bfloat16* __restrict__ dh_up, long up_stride,     // nullable
+    const __hip_bfloat16* __restrict__ acts, long act_stride,     // (B,4H)
+    const float* __restrict__ c_cur, long cc_stride,
+    const float* __restrict__ c_prev, long cp_stride,
+    __hip_bfloat16* __restrict__ dgates_out, long dgo_stride,     // (B,4H)
+    float* __restrict__ dc, long dc_stride,                       // in-place
+    __hip_bfloat16* __restrict__ dh_out, long dho_stride,         // RAW only
+    int B, int H)
+{
+    constexpr int BK = 64;
+    constexpr int OPBYTES = 32768;
+    extern __shared__ __attribute__((aligned(16))) char smem[];
+
+    const int tid = threadIdx.x;
+    const int wave = tid >> 6;
+    const int lane = tid & 63;
+    const int lr = lane & 15;
+    const int lk = lane >> 4;
+    const int wm = wave & 1;
+    const int wn = wave >> 1;
+
+    int m_blk = blockIdx.x, n_blk = blockIdx.y;
+    if (XCD_SWZ) {
+        const int nm = gridDim.x, nwg = gridDim.x * gridDim.y;
+        const int bid = blockIdx.x + blockIdx.y * nm;
+        const int q = nwg >> 3, r = nwg & 7, xcd = bid & 7, loc = bid >> 3;
+        const int swz = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q)
+                        + loc;
+        m_blk = swz % nm;
+        n_blk = swz / nm;
+    }
+    const int m_base = m_blk * 256;
+    const long n_col0 = (long)n_blk * 256;   // 256 dh columns per tile
+    const int K = 4 * H;
+    const int nsteps = K / BK;
+
+    auto stage_half = [&](int is_b, int hf, int buf, int k0) {
+        char* const base = smem + buf * (2 * OPBYTES) + is_b * OPBYTES;
+        const int r = hf * 128 + (tid >> 2);
+        const int ch = tid & 3;
+        const int chs = (ch ^ swz64(r)) * 8;
+#pragma unroll
+        for (int kb = 0; kb < 2; ++kb) {
+            const void* src;
+            if (!is_b) {
+                src = dg_next + (long)(m_base + r) * dgn_stride + (k0 + kb * 32)
+                      + chs;
+            } else {
+                src = w_h + (n_col0 + r) * wh_stride + (k0 + kb * 32) + chs;
+            }
+            __builtin_amdgcn_global_load_lds(
+                (const __attribute__((address_space(1))) void*)src,
+                (__attribute__((address_space(3))) void*)(
+                    base + kb * 16384 + hf * 8192 + wave * 1024), 16, 0, 0);
+        }
+    };
+
+    f32x4 acc[4][8];
+#pragma unroll
+    for (int g = 0; g < 4; ++g)
+#pragma unroll
+        for (int mf = 0; mf < 8; ++mf) acc[g][mf] = f32x4{0, 0, 0, 0};
+
+    stage_half(0, 0, 0, 0);
+    stage_half(1, 1, 0, 0);
+    stage_half(1, 0, 0, 0);
+    stage_half(0, 1, 0, 0);
+    if (nsteps > 1) {
+        asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+        stage_half(1, 0, 1, BK);
+        stage_half(0, 1, 1, BK);
+        asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+    } else {
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    }
+    __builtin_amdgcn_s_barrier();
+
+    bf16x8 wfr[2][2];
+    bf16x8 hfr[8][2];
+
+    for (int j = 0; j < nsteps; ++j) {
+        const int buf = j & 1;
+        const char* ab = smem + buf * (2 * OPBYTES);
+        const char* bb = ab + OPBYTES;
+#pragma unroll
+        for (int g = 0; g < 2; ++g) {
+            const int gc = g * 64 + wn * 16 + lr;
+            const int co = (lk ^ swz64(gc)) * 16;
+#pragma unroll
+            for (int kb = 0; kb < 2; ++kb)
+                wfr[g][kb] = *(const bf16x8*)(bb + kb * 16384 + gc * 64 + co);
+        }
+#pragma unroll
+        for (int mi = 0; mi < 8; ++mi) {
+            const int row = wm * 128 + mi * 16 + lr;
+            const int co = (lk ^ swz64(row)) * 16;
+#pragma unroll
+            for (int kb = 0; kb < 2; ++kb)
+                hfr[mi][kb] = *(const bf16x8*)(ab + kb * 16384 + row * 64 + co);
+        }
+        if (j + 1 < nsteps) {
+            stage_half(0, 0, (j + 1) & 1, (j + 1) * BK);
+            stage_half(1, 1, (j + 1) & 1, (j + 1) * BK);
+        }
+        asm volatile("s_waitcnt lgkmcnt(8)" ::: "memory");
+        __builtin_amdgcn_s_barrier();
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+#pragma unroll
+        for (int g = 0; g < 2; ++g)
+#pragma unroll
+            for (int mi = 0; mi < 8; ++mi)
+#pragma unroll
+                for (int kb = 0; kb < 2; ++kb)
+                    acc[g][mi] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        wfr[g][kb], hfr[mi][kb], acc[g][mi], 0, 0, 0);
+        __builtin_amdgcn_s_barrier();
+#pragma unroll
+        for (int g = 0; g < 2; ++g) {
+            const int gc = (2 + g) * 64 + wn * 16 + lr;
+            const int co = (lk ^ swz64(gc)) * 16;
+#pragma unroll
+            for (int kb = 0; kb < 2; ++kb)
+                wfr[g][kb] = *(const bf16x8*)(bb + kb * 16384 + gc * 64 + co);
+        }
+        if (j + 2 < nsteps) {
+            stage_half(1, 0, buf, (j + 2) * BK);
+            stage_half(0, 1, buf, (j + 2) * BK);
+        }
+        __builtin_amdgcn_s_barrier();
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+#pragma unroll
+        for (int g = 0; g < 2; ++g)
+#pragma unroll
+            for (int mi = 0; mi < 8; ++mi)
+#pragma unroll
+                for (int kb = 0; kb < 2; ++kb)
+                    acc[2 + g][mi] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        wfr[g][kb], hfr[mi][kb], acc[2 + g][mi], 0, 0, 0);
+        if (j + 1 < nsteps) {
+            if (j + 2 < nsteps)
+                asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+            else
+                asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        }
+        __builtin_amdgcn_s_barrier();
+    }
+
+    // Epilogue: acc[cf][nf] holds dh for 4 dh-column groups (cf*64 apart);
+    // fused pointwise backward of step t at each (row, hcol).
+#pragma unroll
+    for (int cf = 0; cf < 4; ++cf) {
+        const long hcol0 = n_col0 + cf * 64 + wn * 16 + lk * 4;
+#pragma unroll
+        for (int nf = 0; nf < 8; ++nf) {
+            const long row = m_base + wm * 128 + nf * 16 + lr;
+            f32x4 dh = acc[cf][nf];
+            if (RAW_DEBUG) {
+                bf16x4 o;
+#pragma unroll
+                for (int jj = 0; jj < 4; ++jj) o[jj] = (__bf16)dh[jj];
+                *(bf16x4*)(dh_out + row * dho_stride + hcol0) = o;
+                continue;
+            }
+            if (dh_up) {
+                const bf16x4 uv = *(const bf16x4*)(
+                    dh_up + row * up_stride + hcol0);
+#pragma unroll
+                for (int jj = 0; jj < 4; ++jj) dh[jj] += (float)uv[jj];
+            }
+            const f32x4 cc = *(const f32x4*)(c_cur + row * cc_stride + hcol0);
+            const f32x4 cp = *(const f32x4*)(c_prev + row * cp_stride + hcol0);
+            const f32x4 dcv = *(const f32x4*)(dc + row * dc_stride + hcol0);
+            const __hip_bfloat16* ap = acts + row * act_stride + hcol0;
+            const bf16x4 vi = *(const bf16x4*)(ap);
+            const bf16x4 vf = *(const bf16x4*)(ap + (long)H);
+            const bf16x4 vg = *(const bf16x4*)(ap + (long)2 * H);
+            const bf16x4 vo = *(const bf16x4*)(ap + (long)3 * H);
+            bf16x4 out_i, out_f, out_g, out_o;
+            f32x4 dc_out;
+#pragma unroll
+            for (int jj = 0; jj < 4; ++jj) {
+                const float i = (float)vi[jj], f = (float)vf[jj];
+                const float g = (float)vg[jj], o = (float)vo[jj];
+                const float tc = tanhf_(cc[jj]);
+                const float d_o = dh[jj] * tc;
+                float d_c = dcv[jj] + dh[jj] * o * (1.0f - tc * tc);
+                const float d_i = d_c * g;
+                const float d_g = d_c * i;
+                const float d_f = d_c * cp[jj];
+                dc_out[jj] = d_c * f;
+                out_i[jj] = (__bf16)(d_i * i * (1.0f - i));
+                out_f[jj] = (__bf16)(d_f * f * (1.0f - f));
+                out_g[jj] = (__bf16)(d_g * (1.0f - g * g));
+                out_o[jj] = (__bf16)(d_o * o * (1.0f - o));
+            }
+            __hip_bfloat16* dgp = dgates_out + row * dgo_stride + hcol0;
+            *(bf16x4*)(dgp)               = out_i;
+            *(bf16x4*)(dgp + (long)H)     = out_f;
+            *(bf16x4*)(dgp + (long)2 * H) = out_g;
+            *(bf16x4*)(dgp + (long)3 * H) = out_o;
+            *(f32x4*)(dc + row * dc_stride + hcol0) = dc_out;
+        }
+    }
+}
+
+extern "C" int launch_lstm_dh_pw_bwd(
+    const void* dg_next, long dgn_stride, const void* w_h, long wh_stride,
+    const void* dh_up, long up_stride, const void* acts, long act_stride,
+    const float* c_cur, long cc_stride, const float* c_prev, long cp_stride,
+    void* dgates_out, long dgo_stride, float* dc, long dc_stride,
+    void* dh_out, long dho_stride, int B, int H, int raw, int xcd_swz,
+    hipStream_t stream)
+{
+    if (B % 256 != 0 || H % 64 != 0) return -1;
+    const size_t smem = 131072;
+    dim3 grid(B / 256, H / 256), block(512);
+    if (H % 256 != 0) return -1;
+#define DHLAUNCH(RAW_, SW_)                                                   \
+    do {                                                                      \
+        hipFuncSetAttribute((const void*)(k_lstm_dh_pw_bwd<RAW_, SW_>),       \
+                            hipFuncAttributeMaxDynamicSharedMemorySize,       \
+                            (int)smem);                                       \
+        hipLaunchKernelGGL((k_lstm_dh_pw_bwd<RAW_, SW_>), grid, block, smem,  \
+                           stream,                                            \
+            (const __hip_bfloat16*)dg_next, dgn_stride,                       \
+            (const __hip_bfloat16*)w_h, wh_stride,                            \
+            (const __hip_bfloat16*)dh_up, up_stride,                          \
+            (const __hip_bfloat16*)acts, act_stride, c_cur, cc_stride,        \
+            c_prev, cp_stride, (__hip_bfloat16*)dgates_out, dgo_stride,       \
+            dc, dc_stride, (__hip_bfloat16*)dh_out, dho_stride, B, H);        \
+    } while (0)
+    if (raw) {
+        if (xcd_swz) DHLAUNCH(true, true); else DHLAUNCH(true, false);
+    } else {
+        if (xcd_swz) DHLAUNCH(false, true); else DHLAUNCH(false, false);
+    }
+#undef DHLAUNCH
+    return 0;
+}
+
+// ---------------------------------------------------------------------------
 // PERSISTENT whole-sequence forward: ONE kernel for the entire T-step loop.
 //
 // The recurrent weight slice stays RESIDENT in LDS for all T timesteps and

@@ -143,6 +143,37 @@ def test_seq_bwd_vs_ref(ext, reverse):
     assert (dc0 - dc0_r).abs().max().item() < 0.05
 
 
+@pytest.mark.parametrize("reverse", [False, True])
+@pytest.mark.parametrize("B,T,H", [(256, 6, 256), (512, 3, 1024)])
+def test_seq_bwd_fused_vs_ref(ext, reverse, B, T, H):
+    """Fused dh-GEMM+pointwise backward (k_lstm_dh_pw_bwd) vs the fp32
+    reference AND vs the split path on identical saved tensors."""
+    from lstm_tensorspark_amd.ops import functional
+
+    pre, w_h, h0, c0 = seq_inputs(B, T, H, seed=5)
+    h_r, c_r, a_r, hT_r, cT_r = functional.ref_seq_fwd(pre, w_h, h0, c0, reverse)
+    g = torch.Generator(device="cpu").manual_seed(6)
+    d_hseq = (torch.randn(T, B, H, generator=g) * 0.1).to("cuda").bfloat16()
+    prev = ext.get_fused_bwd()
+    try:
+        ext.set_fused_bwd(1)
+        dg, dh0, dc0 = ext.lstm_seq_bwd(d_hseq, None, None, w_h, a_r, c_r,
+                                        c0, reverse)
+        ext.set_fused_bwd(0)
+        dg_s, dh0_s, dc0_s = ext.lstm_seq_bwd(d_hseq, None, None, w_h, a_r,
+                                              c_r, c0, reverse)
+    finally:
+        ext.set_fused_bwd(prev)
+    dg_r, dh0_r, dc0_r = functional.ref_seq_bwd(d_hseq, None, None, w_h, a_r,
+                                                c_r, c0, reverse)
+    assert (dg.float() - dg_r.float()).abs().max().item() < 0.05
+    assert (dh0.float() - dh0_r.float()).abs().max().item() < 0.05
+    assert (dc0 - dc0_r).abs().max().item() < 0.05
+    # fused vs split agree to bf16 rounding (fused keeps the dh carry fp32)
+    assert (dg.float() - dg_s.float()).abs().max().item() < 0.05
+    assert (dh0.float() - dh0_s.float()).abs().max().item() < 0.05
+
+
 def test_seq_bwd_none_grads(ext):
     from lstm_tensorspark_amd.ops import functional
 
