@@ -1039,7 +1039,7 @@ if constexpr (!(ABL & 2)) {
 // slots' last reads in Ph1(j).  Boundary vmcnt(4) leaves exactly Ph2(j)'s
 // two stages in flight and guarantees ALL of tile j+1 landed.
 // ---------------------------------------------------------------------------
-template <bool RAW_DEBUG, bool XCD_SWZ>
+template <bool RAW_DEBUG, bool XCD_SWZ, int TUNE = 0>
 __global__ __launch_bounds__(512, 2) void k_lstm_rec_fwd_2ph(
     const __hip_bfloat16* __restrict__ h_prev, long hp_stride,
     const __hip_bfloat16* __restrict__ w_t,     // (4H, H) row-major = W_h^T
@@ -1127,6 +1127,10 @@ __global__ __launch_bounds__(512, 2) void k_lstm_rec_fwd_2ph(
         const char* ab = smem + buf * (2 * OPBYTES);
         const char* bb = ab + OPBYTES;
         // ---- Ph1: A mi0-7 + W g01 (20 ds_reads); stage A-lo/B-hi(j+1)
+        if ((TUNE & 2) && j + 1 < nsteps) {
+            stage_half(0, 0, (j + 1) & 1, (j + 1) * BK);  // A-lo(j+1)
+            stage_half(1, 1, (j + 1) & 1, (j + 1) * BK);  // B-hi(j+1)
+        }
 #pragma unroll
         for (int g = 0; g < 2; ++g) {
             const int gc = g * 64 + wn * 16 + lr;
@@ -1143,11 +1147,12 @@ __global__ __launch_bounds__(512, 2) void k_lstm_rec_fwd_2ph(
             for (int kb = 0; kb < 2; ++kb)
                 hfr[mi][kb] = *(const bf16x8*)(ab + kb * 16384 + row * 64 + co);
         }
-        if (j + 1 < nsteps) {
+        if (!(TUNE & 2) && j + 1 < nsteps) {
             stage_half(0, 0, (j + 1) & 1, (j + 1) * BK);  // A-lo(j+1)
             stage_half(1, 1, (j + 1) & 1, (j + 1) * BK);  // B-hi(j+1)
         }
-        asm volatile("s_waitcnt lgkmcnt(8)" ::: "memory");
+        if (!(TUNE & 1))
+            asm volatile("s_waitcnt lgkmcnt(8)" ::: "memory");
         __builtin_amdgcn_s_barrier();
         asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
 #pragma unroll
@@ -1160,6 +1165,10 @@ __global__ __launch_bounds__(512, 2) void k_lstm_rec_fwd_2ph(
                         wfr[g][kb], hfr[mi][kb], acc[g][mi], 0, 0, 0);
         __builtin_amdgcn_s_barrier();
         // ---- Ph2: W g23 (4 ds_reads); stage B-lo/A-hi(j+2)
+        if ((TUNE & 2) && j + 2 < nsteps) {
+            stage_half(1, 0, buf, (j + 2) * BK);  // B-lo(j+2)
+            stage_half(0, 1, buf, (j + 2) * BK);  // A-hi(j+2)
+        }
 #pragma unroll
         for (int g = 0; g < 2; ++g) {
             const int gc = (2 + g) * 64 + wn * 16 + lr;
@@ -1168,7 +1177,7 @@ __global__ __launch_bounds__(512, 2) void k_lstm_rec_fwd_2ph(
             for (int kb = 0; kb < 2; ++kb)
                 wfr[g][kb] = *(const bf16x8*)(bb + kb * 16384 + gc * 64 + co);
         }
-        if (j + 2 < nsteps) {
+        if (!(TUNE & 2) && j + 2 < nsteps) {
             stage_half(1, 0, buf, (j + 2) * BK);  // B-lo(j+2)
             stage_half(0, 1, buf, (j + 2) * BK);  // A-hi(j+2)
         }
@@ -2248,23 +2257,33 @@ void launch_lstm_rec_fwd_v(
     if ((variant & 1024) && (B % 256 == 0)) {
         const size_t smem = 131072;
         dim3 grid(B / 256, H / 64), block(512);
-#define P2LAUNCH(RAW_, SW_)                                                   \
+#define P2LAUNCH(RAW_, SW_, TN_)                                              \
     do {                                                                      \
-        hipFuncSetAttribute((const void*)(k_lstm_rec_fwd_2ph<RAW_, SW_>),     \
+        hipFuncSetAttribute((const void*)(k_lstm_rec_fwd_2ph<RAW_, SW_, TN_>),\
                             hipFuncAttributeMaxDynamicSharedMemorySize,       \
                             (int)smem);                                       \
-        hipLaunchKernelGGL((k_lstm_rec_fwd_2ph<RAW_, SW_>), grid, block,      \
+        hipLaunchKernelGGL((k_lstm_rec_fwd_2ph<RAW_, SW_, TN_>), grid, block, \
                            smem, stream,                                      \
             (const __hip_bfloat16*)h_prev, hp_stride,                         \
             (const __hip_bfloat16*)w_t, (const __hip_bfloat16*)pre,           \
             pre_stride, c_prev, cp_stride, (__hip_bfloat16*)h_out, ho_stride, \
             c_out, co_stride, (__hip_bfloat16*)acts, act_stride, B, H);       \
     } while (0)
+#define P2DISPATCH(RAW_, SW_)                                                 \
+    switch (variant & 3) {                                                    \
+        case 1: P2LAUNCH(RAW_, SW_, 1); break;                                \
+        case 2: P2LAUNCH(RAW_, SW_, 2); break;                                \
+        case 3: P2LAUNCH(RAW_, SW_, 3); break;                                \
+        default: P2LAUNCH(RAW_, SW_, 0); break;                               \
+    }
         if (raw_debug) {
-            if (variant & 8) P2LAUNCH(true, true); else P2LAUNCH(true, false);
+            if (variant & 8) { P2DISPATCH(true, true); }
+            else { P2DISPATCH(true, false); }
         } else {
-            if (variant & 8) P2LAUNCH(false, true); else P2LAUNCH(false, false);
+            if (variant & 8) { P2DISPATCH(false, true); }
+            else { P2DISPATCH(false, false); }
         }
+#undef P2DISPATCH
 #undef P2LAUNCH
         return;
     }
