@@ -1200,54 +1200,98 @@ __global__ __launch_bounds__(512, 2) void k_lstm_rec_fwd_2ph(
         __builtin_amdgcn_s_barrier();
     }
 
-    // epilogue identical to k_lstm_rec_fwd_8p
-    const int hcol0 = (int)n_col0 + wn * 16 + lk * 4;
+    // Epilogue. RAW mode keeps the direct per-lane form (numerics tests).
+    // The CELL path stages acc through LDS (free after the K-loop) so that
+    // every global access is a coalesced >=16-B op: the direct form's 8-B
+    // accesses (lane = 4 cols of one row, rows 16 apart across lanes) were
+    // store-issue-bound -- measured ~60 us of the B=16384 step.
+    if (RAW_DEBUG) {
+        const int hcol0 = (int)n_col0 + wn * 16 + lk * 4;
 #pragma unroll
-    for (int nf = 0; nf < 8; ++nf) {
-        const long row = m_base + wm * 128 + nf * 16 + lr;
-        f32x4 raw[4];
-#pragma unroll
-        for (int g = 0; g < 4; ++g) {
-            const bf16x4 pv = *(const bf16x4*)(
-                pre + row * pre_stride + (long)g * H + hcol0);
-#pragma unroll
-            for (int j = 0; j < 4; ++j)
-                raw[g][j] = acc[g][nf][j] + (float)pv[j];
-        }
-        __hip_bfloat16* ap = acts + row * act_stride + hcol0;
-        if (RAW_DEBUG) {
+        for (int nf = 0; nf < 8; ++nf) {
+            const long row = m_base + wm * 128 + nf * 16 + lr;
+            __hip_bfloat16* ap = acts + row * act_stride + hcol0;
 #pragma unroll
             for (int g = 0; g < 4; ++g) {
+                const bf16x4 pv = *(const bf16x4*)(
+                    pre + row * pre_stride + (long)g * H + hcol0);
                 bf16x4 o;
 #pragma unroll
-                for (int j = 0; j < 4; ++j) o[j] = (__bf16)raw[g][j];
+                for (int jj = 0; jj < 4; ++jj)
+                    o[jj] = (__bf16)(acc[g][nf][jj] + (float)pv[jj]);
                 *(bf16x4*)(ap + (long)g * H) = o;
             }
-            continue;
         }
-        const f32x4 cp = *(const f32x4*)(c_prev + row * cp_stride + hcol0);
-        bf16x4 oi, of, og, oo, oh;
-        f32x4 oc;
+        return;
+    }
+    float* const img = (float*)smem;        // [64 rows][256 gate-cols] fp32
+    const int wr_col = wn * 16 + lk * 4;    // this lane's 4 cols (per gate)
 #pragma unroll
-        for (int j = 0; j < 4; ++j) {
-            const float gi = sigmoidf_(raw[0][j]);
-            const float gf = sigmoidf_(raw[1][j]);
-            const float gg = tanhf_(raw[2][j]);
-            const float go = sigmoidf_(raw[3][j]);
-            const float c  = gf * cp[j] + gi * gg;
-            oc[j] = c;
-            oh[j] = (__bf16)(go * tanhf_(c));
-            oi[j] = (__bf16)gi;
-            of[j] = (__bf16)gf;
-            og[j] = (__bf16)gg;
-            oo[j] = (__bf16)go;
+    for (int rd = 0; rd < 4; ++rd) {
+        if (rd) __syncthreads();            // previous round's reads done
+        // write rows nf = 2rd, 2rd+1 of this wave's m-half; chunk index
+        // XORed with (row & 3) << 2 to spread ds_write bank slots
+#pragma unroll
+        for (int nn = 0; nn < 2; ++nn) {
+            const int nf = rd * 2 + nn;
+            const int irow = wm * 32 + nn * 16 + lr;
+#pragma unroll
+            for (int g = 0; g < 4; ++g) {
+                const int chunk = (g * 16 + wn * 4 + lk) ^ ((irow & 3) << 2);
+                *(f32x4*)(img + irow * 256 + chunk * 4) = acc[g][nf];
+            }
         }
-        *(bf16x4*)(h_out + row * ho_stride + hcol0) = oh;
-        *(f32x4*)(c_out + row * co_stride + hcol0) = oc;
-        *(bf16x4*)(ap)               = oi;
-        *(bf16x4*)(ap + (long)H)     = of;
-        *(bf16x4*)(ap + (long)2 * H) = og;
-        *(bf16x4*)(ap + (long)3 * H) = oo;
+        __syncthreads();
+        // read pass: thread = (row = tid>>3, 8 h-cols = (tid&7)*8)
+        const int irow = tid >> 3;
+        const int hc = (tid & 7) * 8;
+        const int wm_r = irow >> 5, rem = irow & 31;
+        const long grow = m_base + wm_r * 128 + (rd * 2 + (rem >> 4)) * 16
+                          + (rem & 15);
+        float raw[4][8];
+#pragma unroll
+        for (int g = 0; g < 4; ++g) {
+#pragma unroll
+            for (int hh = 0; hh < 2; ++hh) {
+                const int chunk = (g * 16 + (hc >> 2) + hh)
+                                  ^ ((irow & 3) << 2);
+                const f32x4 v = *(const f32x4*)(img + irow * 256 + chunk * 4);
+#pragma unroll
+                for (int jj = 0; jj < 4; ++jj) raw[g][hh * 4 + jj] = v[jj];
+            }
+            const bf16x8 pv = *(const bf16x8*)(
+                pre + grow * pre_stride + (long)g * H + n_col0 + hc);
+#pragma unroll
+            for (int jj = 0; jj < 8; ++jj) raw[g][jj] += (float)pv[jj];
+        }
+        const long hcol = n_col0 + hc;
+        const f32x4 cp0 = *(const f32x4*)(c_prev + grow * cp_stride + hcol);
+        const f32x4 cp1 = *(const f32x4*)(c_prev + grow * cp_stride + hcol + 4);
+        bf16x8 oi, of, og, oo, oh;
+        f32x4 oc0, oc1;
+#pragma unroll
+        for (int jj = 0; jj < 8; ++jj) {
+            const float gi = sigmoidf_(raw[0][jj]);
+            const float gf = sigmoidf_(raw[1][jj]);
+            const float gg = tanhf_(raw[2][jj]);
+            const float go = sigmoidf_(raw[3][jj]);
+            const float cpv = jj < 4 ? cp0[jj] : cp1[jj - 4];
+            const float c = gf * cpv + gi * gg;
+            if (jj < 4) oc0[jj] = c; else oc1[jj - 4] = c;
+            oh[jj] = (__bf16)(go * tanhf_(c));
+            oi[jj] = (__bf16)gi;
+            of[jj] = (__bf16)gf;
+            og[jj] = (__bf16)gg;
+            oo[jj] = (__bf16)go;
+        }
+        *(bf16x8*)(h_out + grow * ho_stride + hcol) = oh;
+        *(f32x4*)(c_out + grow * co_stride + hcol) = oc0;
+        *(f32x4*)(c_out + grow * co_stride + hcol + 4) = oc1;
+        __hip_bfloat16* ap = acts + grow * act_stride + hcol;
+        *(bf16x8*)(ap)               = oi;
+        *(bf16x8*)(ap + (long)H)     = of;
+        *(bf16x8*)(ap + (long)2 * H) = og;
+        *(bf16x8*)(ap + (long)3 * H) = oo;
     }
 }
 
@@ -1357,6 +1401,10 @@ __global__ __launch_bounds__(512, 2) void k_lstm_dh_pw_bwd(
         const int buf = j & 1;
         const char* ab = smem + buf * (2 * OPBYTES);
         const char* bb = ab + OPBYTES;
+        if (j + 1 < nsteps) {
+            stage_half(0, 0, (j + 1) & 1, (j + 1) * BK);
+            stage_half(1, 1, (j + 1) & 1, (j + 1) * BK);
+        }
 #pragma unroll
         for (int g = 0; g < 2; ++g) {
             const int gc = g * 64 + wn * 16 + lr;
@@ -1373,11 +1421,6 @@ __global__ __launch_bounds__(512, 2) void k_lstm_dh_pw_bwd(
             for (int kb = 0; kb < 2; ++kb)
                 hfr[mi][kb] = *(const bf16x8*)(ab + kb * 16384 + row * 64 + co);
         }
-        if (j + 1 < nsteps) {
-            stage_half(0, 0, (j + 1) & 1, (j + 1) * BK);
-            stage_half(1, 1, (j + 1) & 1, (j + 1) * BK);
-        }
-        asm volatile("s_waitcnt lgkmcnt(8)" ::: "memory");
         __builtin_amdgcn_s_barrier();
         asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
 #pragma unroll
@@ -1389,6 +1432,10 @@ __global__ __launch_bounds__(512, 2) void k_lstm_dh_pw_bwd(
                     acc[g][mi] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                         wfr[g][kb], hfr[mi][kb], acc[g][mi], 0, 0, 0);
         __builtin_amdgcn_s_barrier();
+        if (j + 2 < nsteps) {
+            stage_half(1, 0, buf, (j + 2) * BK);
+            stage_half(0, 1, buf, (j + 2) * BK);
+        }
 #pragma unroll
         for (int g = 0; g < 2; ++g) {
             const int gc = (2 + g) * 64 + wn * 16 + lr;
@@ -1396,10 +1443,6 @@ __global__ __launch_bounds__(512, 2) void k_lstm_dh_pw_bwd(
 #pragma unroll
             for (int kb = 0; kb < 2; ++kb)
                 wfr[g][kb] = *(const bf16x8*)(bb + kb * 16384 + gc * 64 + co);
-        }
-        if (j + 2 < nsteps) {
-            stage_half(1, 0, buf, (j + 2) * BK);
-            stage_half(0, 1, buf, (j + 2) * BK);
         }
         __builtin_amdgcn_s_barrier();
         asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
@@ -1420,60 +1463,102 @@ __global__ __launch_bounds__(512, 2) void k_lstm_dh_pw_bwd(
         __builtin_amdgcn_s_barrier();
     }
 
-    // Epilogue: acc[cf][nf] holds dh for 4 dh-column groups (cf*64 apart);
-    // fused pointwise backward of step t at each (row, hcol).
+    // Epilogue. RAW: direct per-lane dh store (chain-end dh0 + tests).
+    // CELL path: LDS-bounce (see k_lstm_rec_fwd_2ph) -- the direct form's
+    // 8-B scattered accesses made the fused step slower than split.
+    if (RAW_DEBUG) {
 #pragma unroll
-    for (int cf = 0; cf < 4; ++cf) {
-        const long hcol0 = n_col0 + cf * 64 + wn * 16 + lk * 4;
+        for (int cf = 0; cf < 4; ++cf) {
+            const long hcol0 = n_col0 + cf * 64 + wn * 16 + lk * 4;
 #pragma unroll
-        for (int nf = 0; nf < 8; ++nf) {
-            const long row = m_base + wm * 128 + nf * 16 + lr;
-            f32x4 dh = acc[cf][nf];
-            if (RAW_DEBUG) {
+            for (int nf = 0; nf < 8; ++nf) {
+                const long row = m_base + wm * 128 + nf * 16 + lr;
                 bf16x4 o;
 #pragma unroll
-                for (int jj = 0; jj < 4; ++jj) o[jj] = (__bf16)dh[jj];
+                for (int jj = 0; jj < 4; ++jj) o[jj] = (__bf16)acc[cf][nf][jj];
                 *(bf16x4*)(dh_out + row * dho_stride + hcol0) = o;
-                continue;
             }
+        }
+        return;
+    }
+    float* const img = (float*)smem;        // [64 rows][256 dh-cols] fp32
+#pragma unroll
+    for (int rd = 0; rd < 4; ++rd) {
+        if (rd) __syncthreads();
+#pragma unroll
+        for (int nn = 0; nn < 2; ++nn) {
+            const int nf = rd * 2 + nn;
+            const int irow = wm * 32 + nn * 16 + lr;
+#pragma unroll
+            for (int cf = 0; cf < 4; ++cf) {
+                const int chunk = (cf * 16 + wn * 4 + lk) ^ ((irow & 3) << 2);
+                *(f32x4*)(img + irow * 256 + chunk * 4) = acc[cf][nf];
+            }
+        }
+        __syncthreads();
+        // 64 rows x 32 8-col chunks = 2048 tasks, 4 per thread
+#pragma unroll
+        for (int sub = 0; sub < 4; ++sub) {
+            const int irow = sub * 16 + (tid >> 5);
+            const int hc = (tid & 31) * 8;
+            const int wm_r = irow >> 5, rem = irow & 31;
+            const long grow = m_base + wm_r * 128 + (rd * 2 + (rem >> 4)) * 16
+                              + (rem & 15);
+            float dhv[8];
+#pragma unroll
+            for (int hh = 0; hh < 2; ++hh) {
+                const int chunk = ((hc >> 2) + hh) ^ ((irow & 3) << 2);
+                const f32x4 v = *(const f32x4*)(img + irow * 256 + chunk * 4);
+#pragma unroll
+                for (int jj = 0; jj < 4; ++jj) dhv[hh * 4 + jj] = v[jj];
+            }
+            const long hcol = n_col0 + hc;
             if (dh_up) {
-                const bf16x4 uv = *(const bf16x4*)(
-                    dh_up + row * up_stride + hcol0);
+                const bf16x8 uv = *(const bf16x8*)(
+                    dh_up + grow * up_stride + hcol);
 #pragma unroll
-                for (int jj = 0; jj < 4; ++jj) dh[jj] += (float)uv[jj];
+                for (int jj = 0; jj < 8; ++jj) dhv[jj] += (float)uv[jj];
             }
-            const f32x4 cc = *(const f32x4*)(c_cur + row * cc_stride + hcol0);
-            const f32x4 cp = *(const f32x4*)(c_prev + row * cp_stride + hcol0);
-            const f32x4 dcv = *(const f32x4*)(dc + row * dc_stride + hcol0);
-            const __hip_bfloat16* ap = acts + row * act_stride + hcol0;
-            const bf16x4 vi = *(const bf16x4*)(ap);
-            const bf16x4 vf = *(const bf16x4*)(ap + (long)H);
-            const bf16x4 vg = *(const bf16x4*)(ap + (long)2 * H);
-            const bf16x4 vo = *(const bf16x4*)(ap + (long)3 * H);
-            bf16x4 out_i, out_f, out_g, out_o;
-            f32x4 dc_out;
+            const f32x4 cc0 = *(const f32x4*)(c_cur + grow * cc_stride + hcol);
+            const f32x4 cc1 = *(const f32x4*)(c_cur + grow * cc_stride + hcol + 4);
+            const f32x4 cp0 = *(const f32x4*)(c_prev + grow * cp_stride + hcol);
+            const f32x4 cp1 = *(const f32x4*)(c_prev + grow * cp_stride + hcol + 4);
+            const f32x4 dv0 = *(const f32x4*)(dc + grow * dc_stride + hcol);
+            const f32x4 dv1 = *(const f32x4*)(dc + grow * dc_stride + hcol + 4);
+            const __hip_bfloat16* ap = acts + grow * act_stride + hcol;
+            const bf16x8 vi = *(const bf16x8*)(ap);
+            const bf16x8 vf = *(const bf16x8*)(ap + (long)H);
+            const bf16x8 vg = *(const bf16x8*)(ap + (long)2 * H);
+            const bf16x8 vo = *(const bf16x8*)(ap + (long)3 * H);
+            bf16x8 out_i, out_f, out_g, out_o;
+            f32x4 dc0, dc1;
 #pragma unroll
-            for (int jj = 0; jj < 4; ++jj) {
+            for (int jj = 0; jj < 8; ++jj) {
                 const float i = (float)vi[jj], f = (float)vf[jj];
                 const float g = (float)vg[jj], o = (float)vo[jj];
-                const float tc = tanhf_(cc[jj]);
-                const float d_o = dh[jj] * tc;
-                float d_c = dcv[jj] + dh[jj] * o * (1.0f - tc * tc);
+                const float ccv = jj < 4 ? cc0[jj] : cc1[jj - 4];
+                const float cpv = jj < 4 ? cp0[jj] : cp1[jj - 4];
+                const float dcv = jj < 4 ? dv0[jj] : dv1[jj - 4];
+                const float tc = tanhf_(ccv);
+                const float d_o = dhv[jj] * tc;
+                float d_c = dcv + dhv[jj] * o * (1.0f - tc * tc);
                 const float d_i = d_c * g;
                 const float d_g = d_c * i;
-                const float d_f = d_c * cp[jj];
-                dc_out[jj] = d_c * f;
+                const float d_f = d_c * cpv;
+                const float dco = d_c * f;
+                if (jj < 4) dc0[jj] = dco; else dc1[jj - 4] = dco;
                 out_i[jj] = (__bf16)(d_i * i * (1.0f - i));
                 out_f[jj] = (__bf16)(d_f * f * (1.0f - f));
                 out_g[jj] = (__bf16)(d_g * (1.0f - g * g));
                 out_o[jj] = (__bf16)(d_o * o * (1.0f - o));
             }
-            __hip_bfloat16* dgp = dgates_out + row * dgo_stride + hcol0;
-            *(bf16x4*)(dgp)               = out_i;
-            *(bf16x4*)(dgp + (long)H)     = out_f;
-            *(bf16x4*)(dgp + (long)2 * H) = out_g;
-            *(bf16x4*)(dgp + (long)3 * H) = out_o;
-            *(f32x4*)(dc + row * dc_stride + hcol0) = dc_out;
+            __hip_bfloat16* dgp = dgates_out + grow * dgo_stride + hcol;
+            *(bf16x8*)(dgp)               = out_i;
+            *(bf16x8*)(dgp + (long)H)     = out_f;
+            *(bf16x8*)(dgp + (long)2 * H) = out_g;
+            *(bf16x8*)(dgp + (long)3 * H) = out_o;
+            *(f32x4*)(dc + grow * dc_stride + hcol) = dc0;
+            *(f32x4*)(dc + grow * dc_stride + hcol + 4) = dc1;
         }
     }
 }
@@ -2497,7 +2582,7 @@ void launch_lstm_rec_fwd(
     hipStream_t stream)
 {
     int v = g_fwd_variant;
-    if (v < 0) v = (B % 256 == 0 && B >= 2048) ? 40   // MT256 + XCD swizzle
+    if (v < 0) v = (B % 256 == 0 && B >= 2048) ? 1027  // 2ph TUNE3 (kbench r2)
                  : (B >= 2048) ? 10 : 14;
     launch_lstm_rec_fwd_v(h_prev, hp_stride, w_t, pre, pre_stride, c_prev,
                           cp_stride, h_out, ho_stride, c_out, co_stride, acts,
