@@ -109,10 +109,18 @@ std::vector<Tensor> lstm_seq_fwd(Tensor pre, Tensor w_h, Tensor h0, Tensor c0,
     pre = pre.contiguous();
     auto h0c = h0.contiguous();
     auto c0c = c0.contiguous();
+    // Forward path choice: the hand-written fused 2ph kernel (256x256
+    // tile) beats the split hipBLASLt+pointwise path where its grid fills
+    // the 256-CU chip (measured 204 vs 258 us/step at B=16384 H=1024);
+    // below that, split wins at B*H >= 2M and the BM64/BM32 fused kernels
+    // win smaller still (round-1 sweeps).
+    const bool fill2ph = (B % 256 == 0) && (H % 64 == 0)
+                         && (B / 256) * (H / 64) >= 256;
+    const bool split_auto = (B * H >= 2097152) && !fill2ph;
     // W_h^T (4H, H): one transposed copy per layer call so each lane's MFMA
     // B-fragment is a contiguous 16-byte load (fused path only).
     const bool need_wt = g_persistent_enabled
-        || !((g_split_fwd < 0) ? (B * H >= 2097152) : (g_split_fwd != 0));
+        || !((g_split_fwd < 0) ? split_auto : (g_split_fwd != 0));
     auto w_t = need_wt ? w_h.t().contiguous() : w_h;
 
     auto opts_bf = pre.options();
@@ -154,7 +162,7 @@ std::vector<Tensor> lstm_seq_fwd(Tensor pre, Tensor w_h, Tensor h0, Tensor c0,
         }
     }
 
-    const bool split = (g_split_fwd < 0) ? (B * H >= 2097152) : (g_split_fwd != 0);
+    const bool split = (g_split_fwd < 0) ? split_auto : (g_split_fwd != 0);
     const bool fp8 = g_fp8_fwd && split;
     at::Tensor raw_buf, h8, w8_cm, sc_a, sc_b;
     if (split) raw_buf = at::empty({B, H4}, opts_bf);
@@ -275,7 +283,11 @@ std::vector<Tensor> lstm_seq_bwd(c10::optional<Tensor> dh_seq_opt,
     // backward fused into its epilogue (k_lstm_dh_pw_bwd) -- one kernel
     // per timestep, fp32 dh carry in registers, and no W_h^T transpose
     // copy (the kernel reads the (H,4H) layout directly).
-    const bool fused = (g_fused_bwd != 0) && (B % 256 == 0) && (H % 256 == 0);
+    // auto (-1): only where the (B/256, H/256) grid fills enough of the
+    // 256-CU chip; explicit 1 forces it on any eligible shape (tests).
+    const bool fused = (B % 256 == 0) && (H % 256 == 0)
+        && (g_fused_bwd > 0
+            || (g_fused_bwd < 0 && (B / 256) * (H / 256) >= 192));
     Tensor w_h_t;
     if (!fused) {
         // split fallback: W_h^T as a CONTIGUOUS copy (one 8 MB transpose

@@ -2582,7 +2582,8 @@ void launch_lstm_rec_fwd(
     hipStream_t stream)
 {
     int v = g_fwd_variant;
-    if (v < 0) v = (B % 256 == 0 && B >= 2048) ? 1027  // 2ph TUNE3 (kbench r2)
+    if (v < 0) v = (B % 256 == 0 && (B / 256) * (H / 64) >= 256)
+                     ? 1027                        // 2ph TUNE3 (kbench r2)
                  : (B >= 2048) ? 10 : 14;
     launch_lstm_rec_fwd_v(h_prev, hp_stride, w_t, pre, pre_stride, c_prev,
                           cp_stride, h_out, ho_stride, c_out, co_stride, acts,
