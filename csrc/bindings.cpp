@@ -389,7 +389,7 @@ Tensor decode_gemv(Tensor A_t, Tensor x, c10::optional<Tensor> bias) {
                               out.data_ptr(), nullptr, (int)B, N, (int)K,
                               cur_stream());
     TORCH_CHECK(rc == 0, "decode_gemv: shape outside fast-path envelope "
-                "(B<=16, B*K*2 <= 160 KiB LDS, K % 1024 == 0)");
+                "(B<=16, B*K*2 <= 160 KiB LDS, K % 16 == 0)");
     return out;
 }
 

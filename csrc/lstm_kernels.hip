@@ -2014,7 +2014,11 @@ extern "C" int launch_gemv_bf16(const void* A, const void* x,
                                 int B, long N, int K, hipStream_t stream)
 {
     const size_t smem = (size_t)B * K * 2;
-    if (B < 1 || B > 16 || K % 1024 != 0 || smem > 160 * 1024 || N < 1)
+    // correctness envelope: the lane loop strides bf16x8 PAIRS (K % 16);
+    // decode_gemv_ok additionally demands K % 1024 (full-lane PERF
+    // envelope) for the auto path -- an explicit use_gemv with smaller K
+    // is valid, just underutilizes lanes.
+    if (B < 1 || B > 16 || K % 16 != 0 || smem > 160 * 1024 || N < 1)
         return -1;
     dim3 block(256), grid((unsigned)((N + 3) / 4));
 #define GEMV_LAUNCH(BM)                                                           hipLaunchKernelGGL(k_gemv_bf16<BM>, grid, block, smem, stream,                    (const __hip_bfloat16*)A, (const __hip_bfloat16*)x, bias,                     (__hip_bfloat16*)out, out_f32, B, N, K)
