@@ -41,6 +41,9 @@ int launch_colsum_bf16(const void*, float*, float*, long, long, hipStream_t);
 int colsum_gy(long);
 void launch_quant_fp8_dyn_T(const void*, void*, long, long, const float*,
                             hipStream_t);
+int launch_quant_fp8_pack(const void*, void*, void*, float*, long, long,
+                          const float*, hipStream_t);
+void launch_colsum_final(const float*, float*, long, int, hipStream_t);
 void launch_xent_fwd(const void*, const long*, float*, float*, long, int,
                      hipStream_t);
 void launch_xent_bwd(const void*, const long*, const float*, void*,
@@ -444,6 +447,33 @@ Tensor quant_fp8(Tensor t, Tensor mult) {
     return out;
 }
 
+std::vector<Tensor> quant_fp8_pack(Tensor t, Tensor mult) {
+    // (R,C) bf16 -> {out8 (R,C) e4m3, out8T (C,R) e4m3, db (C,) fp32} in
+    // ONE pass (fp8 wgrad path).  Raises outside the divisibility
+    // envelope -- the python caller guards.
+    check_bf16_cuda(t, "t");
+    TORCH_CHECK(t.dim() == 2, "quant_fp8_pack wants (R,C)");
+    t = t.contiguous();
+    const long R = t.size(0), C = t.size(1);
+    TORCH_CHECK(R % 512 == 0 && C % 128 == 0,
+                "quant_fp8_pack: R % 512 == 0 && C % 128 == 0 required");
+    auto opts8 = t.options().dtype(at::kFloat8_e4m3fn);
+    auto out8 = at::empty({R, C}, opts8);
+    auto out8T = at::empty({C, R}, opts8);
+    auto partials = at::empty({R / 512, C}, t.options().dtype(at::kFloat));
+    auto db = at::empty({C}, t.options().dtype(at::kFloat));
+    int rc = launch_quant_fp8_pack(t.data_ptr(), out8.data_ptr(),
+                                   out8T.data_ptr(),
+                                   (float*)partials.data_ptr(), R, C,
+                                   (const float*)mult.data_ptr(),
+                                   cur_stream());
+    TORCH_CHECK(rc == 0, "quant_fp8_pack launch failed");
+    launch_colsum_final((const float*)partials.data_ptr(),
+                        (float*)db.data_ptr(), C, (int)(R / 512),
+                        cur_stream());
+    return {out8, out8T, db};
+}
+
 Tensor quant_fp8_T(Tensor t, Tensor mult) {
     check_bf16_cuda(t, "t");
     TORCH_CHECK(t.dim() == 2, "quant_fp8_T wants (R,C)");
@@ -574,6 +604,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("decode_gemv_ok", &decode_gemv_ok);
     m.def("quant_fp8", &quant_fp8, "bf16 -> e4m3, device-scalar multiplier");
     m.def("quant_fp8_T", &quant_fp8_T, "bf16 (R,C) -> e4m3 (C,R) transposed");
+    m.def("quant_fp8_pack", &quant_fp8_pack,
+          "one-pass bf16 (R,C) -> {e4m3 (R,C), e4m3 (C,R), colsum fp32}");
     m.def("set_fp8_fwd", [](bool v) { g_fp8_fwd = v; },
           "experimental: fp8(e4m3) recurrent GEMM in the split forward");
     m.def("get_fp8_fwd", []() { return g_fp8_fwd; });

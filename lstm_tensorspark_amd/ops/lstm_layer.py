@@ -181,8 +181,17 @@ class LSTMLayerFn(torch.autograd.Function):
             dg_amax = dg_flat.abs().amax().float().clamp_min(1e-8)
             mult_dg = 448.0 / dg_amax
             scale_dg = dg_amax / 448.0
-            dg8cm = ext.quant_fp8_T(dg_flat, mult_dg).t()  # (TB,4H) col-major
             s448 = torch.full((), 1.0 / 448.0, device=dev)
+            # ONE fused pass over dgates -> {dg8 row-major, dg8^T, db}
+            # (k_quant_fp8_pack; the old 3-pass chain measured 53 ms vs an
+            # 8 ms roofline -- DEFERRED.md r1 item 2).  Envelope guarded.
+            packed = None
+            if (T * B) % 512 == 0 and (4 * H) % 128 == 0:
+                dg8, dg8t, db_f32 = ext.quant_fp8_pack(dg_flat, mult_dg)
+                dg8cm = dg8t.t()                   # (TB,4H) col-major view
+            else:
+                dg8cm = ext.quant_fp8_T(dg_flat, mult_dg).t()
+                dg8 = db_f32 = None
             if ng[4]:
                 hprev = h_prev_view(h_all, ctx.reverse).reshape(T * B, H)
                 hp8t = ext.quant_fp8_T(hprev, torch.full((), 448.0, device=dev))
@@ -196,9 +205,11 @@ class LSTMLayerFn(torch.autograd.Function):
                                         scale_b=scale_dg,
                                         out_dtype=torch.bfloat16)
             if ng[5]:
-                db = dg_flat.sum(dim=0, dtype=torch.float32).to(w_x.dtype)
+                db = (db_f32 if db_f32 is not None
+                      else dg_flat.sum(dim=0, dtype=torch.float32)).to(w_x.dtype)
             if ng[0]:
-                dg8 = ext.quant_fp8(dg_flat, mult_dg)
+                if dg8 is None:
+                    dg8 = ext.quant_fp8(dg_flat, mult_dg)
                 wx_amax = w_x.abs().amax().float().clamp_min(1e-8)
                 wx8 = (w_x * (448.0 / wx_amax)).to(f8)
                 dx = torch._scaled_mm(dg8, wx8.t(), scale_a=scale_dg,

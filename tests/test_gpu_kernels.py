@@ -527,6 +527,24 @@ def test_colsum_vs_torch(ext):
         assert (out - ref).abs().max().item() < 1e-2 * max(1.0, ref.abs().max().item())
 
 
+def test_quant_fp8_pack_vs_separate(ext):
+    """One-pass pack == {quant_fp8, quant_fp8_T, fp32 colsum} bitwise for
+    the fp8 outputs and to fp32-order tolerance for db."""
+    g = torch.Generator(device="cpu").manual_seed(43)
+    R, C = 1024, 256
+    t = (torch.randn(R, C, generator=g) * 2.0).to("cuda").bfloat16()
+    amax = t.abs().amax().float().clamp_min(1e-8)
+    mult = 448.0 / amax
+    dg8, dg8t, db = ext.quant_fp8_pack(t, mult)
+    ref8 = ext.quant_fp8(t, mult)
+    ref8t = ext.quant_fp8_T(t, mult)
+    assert torch.equal(dg8.view(torch.uint8), ref8.view(torch.uint8))
+    assert torch.equal(dg8t.view(torch.uint8), ref8t.view(torch.uint8))
+    ref_db = t.float().sum(dim=0)
+    assert (db - ref_db).abs().max().item() < 1e-2 * max(
+        1.0, ref_db.abs().max().item())
+
+
 def test_colsum_deterministic_run_to_run(ext):
     # the two-stage fixed-order reduction must be bitwise reproducible
     # (the old fp32 atomicAdd across grid-y blocks was not)
