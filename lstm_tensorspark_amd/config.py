@@ -41,6 +41,9 @@ class TrainConfig:
     adam_eps: float = 1e-8
     grad_clip: float = 0.0        # global-norm clip; 0 => off
     dtype: str = "fp32"           # "fp32" | "bf16" | "fp8" (bf16 + e4m3 recurrent GEMM)
+    fp8_head: bool = False        # e4m3 LM-head forward GEMM (rowwise scales,
+                                  # bf16 backward); quality-verified with bf16
+                                  # recurrence -- the lm-50k speed knob
     steps: int = 100
     log_interval: int = 10
     ckpt_interval: int = 0        # 0 => only at end if ckpt_dir set

@@ -257,9 +257,13 @@ class LSTMModel(nn.Module):
 
     @staticmethod
     def _fp8_head() -> bool:
+        # r2: independent of the fp8 RECURRENT switch -- the r1 "fp8 head
+        # quality failure" was misattributed: isolation runs showed the
+        # fp8 recurrent path stalls 50k-way LM learning (loss stuck at
+        # ~10.3 vs bf16's 0.29 @120 steps on lm_assoc) while bf16
+        # recurrent + fp8 head converges cleanly (profiles/r02_fp8_lm.md).
         from ..ops import extension
-        return (_FP8_HEAD and extension.extension_available()
-                and extension.get_extension().get_fp8_fwd())
+        return _FP8_HEAD and extension.extension_available()
 
     def loss(self, output: torch.Tensor, y: torch.Tensor) -> torch.Tensor:
         head = self.cfg.head

@@ -80,6 +80,9 @@ class Trainer:
             from .ops.extension import extension_available, get_extension
             if extension_available():
                 get_extension().set_fp8_fwd(cfg.train.dtype == "fp8")
+                from .models import lstm as _lstm_mod
+                _lstm_mod.enable_fp8_head(
+                    getattr(cfg.train, "fp8_head", False))
                 if getattr(cfg.train, "autotune", False):
                     from .ops.autotune import tune_fwd
                     r = tune_fwd(cfg.train.batch_size, cfg.model.hidden_size)
