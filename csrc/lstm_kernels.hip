@@ -1039,7 +1039,7 @@ if constexpr (!(ABL & 2)) {
 // slots' last reads in Ph1(j).  Boundary vmcnt(4) leaves exactly Ph2(j)'s
 // two stages in flight and guarantees ALL of tile j+1 landed.
 // ---------------------------------------------------------------------------
-template <bool RAW_DEBUG, bool XCD_SWZ, int TUNE = 0>
+template <bool RAW_DEBUG, bool XCD_SWZ, int TUNE = 0, int BM = 256>
 __global__ __launch_bounds__(512, 2) void k_lstm_rec_fwd_2ph(
     const __hip_bfloat16* __restrict__ h_prev, long hp_stride,
     const __hip_bfloat16* __restrict__ w_t,     // (4H, H) row-major = W_h^T
@@ -1051,7 +1051,13 @@ __global__ __launch_bounds__(512, 2) void k_lstm_rec_fwd_2ph(
     int B, int H)
 {
     constexpr int BK = 64;
-    constexpr int OPBYTES = 32768;
+    constexpr int MI = BM / 32;        // m-fragments per wave
+    constexpr int MHALF = BM / 2;      // rows per wave m-half
+    constexpr int ABYTES = BM * 128;   // A operand tile (2 kb sub-images)
+    constexpr int BUFB = ABYTES + 32768;
+    // per-wave glds per stage: B halves 2, A halves BM/128 (at BM=128 each
+    // wave issues ONE -- waves 0-3 fill kb=0, waves 4-7 kb=1)
+    constexpr int NB = 2 + BM / 128;   // boundary vmcnt: Ph2's stages
     extern __shared__ __attribute__((aligned(16))) char smem[];
 
     const int tid = threadIdx.x;
@@ -1072,37 +1078,60 @@ __global__ __launch_bounds__(512, 2) void k_lstm_rec_fwd_2ph(
         m_blk = swz % nm;
         n_blk = swz / nm;
     }
-    const int m_base = m_blk * 256;
+    const int m_base = m_blk * BM;
     const long n_col0 = (long)n_blk * 64;
     const int nsteps = H / BK;
 
     auto stage_half = [&](int is_b, int hf, int buf, int k0) {
-        char* const base = smem + buf * (2 * OPBYTES) + is_b * OPBYTES;
-        const int r = hf * 128 + (tid >> 2);
-        const int ch = tid & 3;
-        const int chs = (ch ^ swz64(r)) * 8;
+        char* const base = smem + buf * BUFB + is_b * ABYTES;
+        if (is_b) {
+            const int r = hf * 128 + (tid >> 2);
+            const int ch = tid & 3;
+            const int chs = (ch ^ swz64(r)) * 8;
 #pragma unroll
-        for (int kb = 0; kb < 2; ++kb) {
-            const void* src;
-            if (!is_b) {
-                src = h_prev + (long)(m_base + r) * hp_stride + (k0 + kb * 32)
-                      + chs;
-            } else {
+            for (int kb = 0; kb < 2; ++kb) {
                 const long grow = (long)(r >> 6) * H + n_col0 + (r & 63);
-                src = w_t + grow * H + (k0 + kb * 32) + chs;
+                const void* src = w_t + grow * H + (k0 + kb * 32) + chs;
+                __builtin_amdgcn_global_load_lds(
+                    (const __attribute__((address_space(1))) void*)src,
+                    (__attribute__((address_space(3))) void*)(
+                        base + kb * 16384 + hf * 8192 + wave * 1024),
+                    16, 0, 0);
             }
+        } else if (BM == 256) {
+            const int r = hf * 128 + (tid >> 2);
+            const int chs = ((tid & 3) ^ swz64(r)) * 8;
+#pragma unroll
+            for (int kb = 0; kb < 2; ++kb) {
+                const void* src = h_prev + (long)(m_base + r) * hp_stride
+                                  + (k0 + kb * 32) + chs;
+                __builtin_amdgcn_global_load_lds(
+                    (const __attribute__((address_space(1))) void*)src,
+                    (__attribute__((address_space(3))) void*)(
+                        base + kb * 16384 + hf * 8192 + wave * 1024),
+                    16, 0, 0);
+            }
+        } else {
+            // BM=128: one glds per thread; waves 0-3 fill kb=0, 4-7 kb=1
+            const int kb = wave >> 2;
+            const int t2 = tid & 255;
+            const int r = hf * 64 + (t2 >> 2);
+            const int chs = ((t2 & 3) ^ swz64(r)) * 8;
+            const void* src = h_prev + (long)(m_base + r) * hp_stride
+                              + (k0 + kb * 32) + chs;
             __builtin_amdgcn_global_load_lds(
                 (const __attribute__((address_space(1))) void*)src,
                 (__attribute__((address_space(3))) void*)(
-                    base + kb * 16384 + hf * 8192 + wave * 1024), 16, 0, 0);
+                    base + kb * (ABYTES / 2) + hf * (ABYTES / 4)
+                    + (wave & 3) * 1024), 16, 0, 0);
         }
     };
 
-    f32x4 acc[4][8];
+    f32x4 acc[4][MI];
 #pragma unroll
     for (int g = 0; g < 4; ++g)
 #pragma unroll
-        for (int mf = 0; mf < 8; ++mf) acc[g][mf] = f32x4{0, 0, 0, 0};
+        for (int mf = 0; mf < MI; ++mf) acc[g][mf] = f32x4{0, 0, 0, 0};
 
     // prologue: tile 0 fully + Ph2(-1)'s {B-lo(1), A-hi(1)}
     stage_half(0, 0, 0, 0);
@@ -1110,22 +1139,23 @@ __global__ __launch_bounds__(512, 2) void k_lstm_rec_fwd_2ph(
     stage_half(1, 0, 0, 0);
     stage_half(0, 1, 0, 0);
     if (nsteps > 1) {
-        asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+        asm volatile("s_waitcnt vmcnt(%0)" :: "i"(NB) : "memory");
         stage_half(1, 0, 1, BK);         // B-lo(1)
         stage_half(0, 1, 1, BK);         // A-hi(1)
-        asm volatile("s_waitcnt vmcnt(4)" ::: "memory");  // tile 0 landed
+        // tile 0 landed (NB = Ph2-equivalent stages left in flight)
+        asm volatile("s_waitcnt vmcnt(%0)" :: "i"(NB) : "memory");
     } else {
         asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     }
     __builtin_amdgcn_s_barrier();
 
     bf16x8 wfr[2][2];
-    bf16x8 hfr[8][2];
+    bf16x8 hfr[MI][2];
 
     for (int j = 0; j < nsteps; ++j) {
         const int buf = j & 1;
-        const char* ab = smem + buf * (2 * OPBYTES);
-        const char* bb = ab + OPBYTES;
+        const char* ab = smem + buf * BUFB;
+        const char* bb = ab + ABYTES;
         // ---- Ph1: A mi0-7 + W g01 (20 ds_reads); stage A-lo/B-hi(j+1)
         if ((TUNE & 2) && j + 1 < nsteps) {
             stage_half(0, 0, (j + 1) & 1, (j + 1) * BK);  // A-lo(j+1)
@@ -1140,12 +1170,13 @@ __global__ __launch_bounds__(512, 2) void k_lstm_rec_fwd_2ph(
                 wfr[g][kb] = *(const bf16x8*)(bb + kb * 16384 + gc * 64 + co);
         }
 #pragma unroll
-        for (int mi = 0; mi < 8; ++mi) {
-            const int row = wm * 128 + mi * 16 + lr;
+        for (int mi = 0; mi < MI; ++mi) {
+            const int row = wm * MHALF + mi * 16 + lr;
             const int co = (lk ^ swz64(row)) * 16;
 #pragma unroll
             for (int kb = 0; kb < 2; ++kb)
-                hfr[mi][kb] = *(const bf16x8*)(ab + kb * 16384 + row * 64 + co);
+                hfr[mi][kb] = *(const bf16x8*)(
+                    ab + kb * (ABYTES / 2) + row * 64 + co);
         }
         if (!(TUNE & 2) && j + 1 < nsteps) {
             stage_half(0, 0, (j + 1) & 1, (j + 1) * BK);  // A-lo(j+1)
@@ -1158,7 +1189,7 @@ __global__ __launch_bounds__(512, 2) void k_lstm_rec_fwd_2ph(
 #pragma unroll
         for (int g = 0; g < 2; ++g)
 #pragma unroll
-            for (int mi = 0; mi < 8; ++mi)
+            for (int mi = 0; mi < MI; ++mi)
 #pragma unroll
                 for (int kb = 0; kb < 2; ++kb)
                     acc[g][mi] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
@@ -1186,14 +1217,14 @@ __global__ __launch_bounds__(512, 2) void k_lstm_rec_fwd_2ph(
 #pragma unroll
         for (int g = 0; g < 2; ++g)
 #pragma unroll
-            for (int mi = 0; mi < 8; ++mi)
+            for (int mi = 0; mi < MI; ++mi)
 #pragma unroll
                 for (int kb = 0; kb < 2; ++kb)
                     acc[2 + g][mi] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                         wfr[g][kb], hfr[mi][kb], acc[2 + g][mi], 0, 0, 0);
         if (j + 1 < nsteps) {
             if (j + 2 < nsteps)
-                asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+                asm volatile("s_waitcnt vmcnt(%0)" :: "i"(NB) : "memory");
             else
                 asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
         }
@@ -1208,8 +1239,8 @@ __global__ __launch_bounds__(512, 2) void k_lstm_rec_fwd_2ph(
     if (RAW_DEBUG) {
         const int hcol0 = (int)n_col0 + wn * 16 + lk * 4;
 #pragma unroll
-        for (int nf = 0; nf < 8; ++nf) {
-            const long row = m_base + wm * 128 + nf * 16 + lr;
+        for (int nf = 0; nf < MI; ++nf) {
+            const long row = m_base + wm * MHALF + nf * 16 + lr;
             __hip_bfloat16* ap = acts + row * act_stride + hcol0;
 #pragma unroll
             for (int g = 0; g < 4; ++g) {
@@ -1226,8 +1257,9 @@ __global__ __launch_bounds__(512, 2) void k_lstm_rec_fwd_2ph(
     }
     float* const img = (float*)smem;        // [64 rows][256 gate-cols] fp32
     const int wr_col = wn * 16 + lk * 4;    // this lane's 4 cols (per gate)
+    (void)wr_col;
 #pragma unroll
-    for (int rd = 0; rd < 4; ++rd) {
+    for (int rd = 0; rd < MI / 2; ++rd) {
         if (rd) __syncthreads();            // previous round's reads done
         // write rows nf = 2rd, 2rd+1 of this wave's m-half; chunk index
         // XORed with (row & 3) << 2 to spread ds_write bank slots
@@ -1246,7 +1278,7 @@ __global__ __launch_bounds__(512, 2) void k_lstm_rec_fwd_2ph(
         const int irow = tid >> 3;
         const int hc = (tid & 7) * 8;
         const int wm_r = irow >> 5, rem = irow & 31;
-        const long grow = m_base + wm_r * 128 + (rd * 2 + (rem >> 4)) * 16
+        const long grow = m_base + wm_r * MHALF + (rd * 2 + (rem >> 4)) * 16
                           + (rem & 15);
         float raw[4][8];
 #pragma unroll
@@ -2448,36 +2480,45 @@ void launch_lstm_rec_fwd_v(
     void* acts, long act_stride, int B, int H, int raw_debug, int variant,
     hipStream_t stream)
 {
-    // 2-phase merged path (bit10): 256x256 BK=64, 32-MFMA segments
-    if ((variant & 1024) && (B % 256 == 0)) {
-        const size_t smem = 131072;
-        dim3 grid(B / 256, H / 64), block(512);
-#define P2LAUNCH(RAW_, SW_, TN_)                                              \
+    // 2-phase merged path (bit10): BMx256 BK=64, 32-MFMA segments;
+    // bit11 selects the BM=128 row tile (grids that under-fill at BM=256)
+    if ((variant & 1024)
+        && (B % ((variant & 2048) ? 128 : 256) == 0)) {
+        const int bm = (variant & 2048) ? 128 : 256;
+        const size_t smem = (size_t)2 * (bm * 128 + 32768);
+        dim3 grid(B / bm, H / 64), block(512);
+#define P2LAUNCH(RAW_, SW_, TN_, BM_)                                         \
     do {                                                                      \
-        hipFuncSetAttribute((const void*)(k_lstm_rec_fwd_2ph<RAW_, SW_, TN_>),\
-                            hipFuncAttributeMaxDynamicSharedMemorySize,       \
-                            (int)smem);                                       \
-        hipLaunchKernelGGL((k_lstm_rec_fwd_2ph<RAW_, SW_, TN_>), grid, block, \
-                           smem, stream,                                      \
+        hipFuncSetAttribute(                                                  \
+            (const void*)(k_lstm_rec_fwd_2ph<RAW_, SW_, TN_, BM_>),           \
+            hipFuncAttributeMaxDynamicSharedMemorySize, (int)smem);           \
+        hipLaunchKernelGGL((k_lstm_rec_fwd_2ph<RAW_, SW_, TN_, BM_>), grid,   \
+                           block, smem, stream,                               \
             (const __hip_bfloat16*)h_prev, hp_stride,                         \
             (const __hip_bfloat16*)w_t, (const __hip_bfloat16*)pre,           \
             pre_stride, c_prev, cp_stride, (__hip_bfloat16*)h_out, ho_stride, \
             c_out, co_stride, (__hip_bfloat16*)acts, act_stride, B, H);       \
     } while (0)
-#define P2DISPATCH(RAW_, SW_)                                                 \
+#define P2DISPATCH(RAW_, SW_, BM_)                                            \
     switch (variant & 3) {                                                    \
-        case 1: P2LAUNCH(RAW_, SW_, 1); break;                                \
-        case 2: P2LAUNCH(RAW_, SW_, 2); break;                                \
-        case 3: P2LAUNCH(RAW_, SW_, 3); break;                                \
-        default: P2LAUNCH(RAW_, SW_, 0); break;                               \
+        case 1: P2LAUNCH(RAW_, SW_, 1, BM_); break;                           \
+        case 2: P2LAUNCH(RAW_, SW_, 2, BM_); break;                           \
+        case 3: P2LAUNCH(RAW_, SW_, 3, BM_); break;                           \
+        default: P2LAUNCH(RAW_, SW_, 0, BM_); break;                          \
     }
+#define P2DISPATCH2(RAW_, SW_)                                                \
+    do {                                                                      \
+        if (variant & 2048) { P2DISPATCH(RAW_, SW_, 128); }                   \
+        else { P2DISPATCH(RAW_, SW_, 256); }                                  \
+    } while (0)
         if (raw_debug) {
-            if (variant & 8) { P2DISPATCH(true, true); }
-            else { P2DISPATCH(true, false); }
+            if (variant & 8) { P2DISPATCH2(true, true); }
+            else { P2DISPATCH2(true, false); }
         } else {
-            if (variant & 8) { P2DISPATCH(false, true); }
-            else { P2DISPATCH(false, false); }
+            if (variant & 8) { P2DISPATCH2(false, true); }
+            else { P2DISPATCH2(false, false); }
         }
+#undef P2DISPATCH2
 #undef P2DISPATCH
 #undef P2LAUNCH
         return;

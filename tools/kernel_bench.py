@@ -73,7 +73,8 @@ def main():
               32, 40,           # MT256 3buf-BK32 (8=XCD swizzle)
               33, 41,           # MT256 2buf-BK64
               512, 520,         # 8-phase BK64 counted-vmcnt (8=XCD swizzle)
-              1024, 1032]:      # 2-phase merged BK64 (8=XCD swizzle)
+              1024, 1032,       # 2-phase merged BK64 (8=XCD swizzle)
+              3072, 3075]:      # 2-phase BM=128 row tile (3=TUNE3)
         ext.set_fwd_variant(v)
         h_out, c_out, acts_out = ext.lstm_rec_fwd_step(h_prev, w_h, pre1,
                                                        c_prev, False)
