@@ -71,6 +71,25 @@ def test_gpu_checkpoint_resume_bitwise(tmp_path):
         assert torch.equal(vA, vB), f"resume not bitwise for {k}"
 
 
+def test_gpu_fused_path_convergence():
+    """Training convergence THROUGH the r2 fused kernels (2ph fwd +
+    dh_pw_bwd bwd engage only when the grid fills the chip, so the small
+    copy-task tests never exercise them end-to-end): B=4096, h=1024 on
+    the copy task must learn."""
+    cfg = Config(
+        model=ModelConfig(input_size=64, hidden_size=1024, num_layers=2,
+                          num_classes=8),
+        train=TrainConfig(batch_size=4096, seq_len=16, lr=0.3, momentum=0.9,
+                          steps=30, log_interval=0, seed=9, dtype="bf16"),
+        data=DataConfig(task="copy", copy_delay=1),
+    )
+    t = Trainer(cfg, device="cuda:0")
+    first = t.train_step(t.data.batch(0))["loss"]
+    t.fit(30)
+    last = t.train_step(t.data.batch(31))["loss"]
+    assert last < first * 0.7, (first, last)
+
+
 def test_graft_smoke_entry():
     import __graft_entry__
 
