@@ -119,7 +119,9 @@ std::vector<Tensor> lstm_seq_fwd(Tensor pre, Tensor w_h, Tensor h0, Tensor c0,
     // win smaller still (round-1 sweeps).
     const bool fill2ph = (B % 256 == 0) && (H % 64 == 0)
                          && (B / 256) * (H / 64) >= 256;
-    const bool split_auto = (B * H >= 2097152) && !fill2ph;
+    // fp8 mode's e4m3 recurrent GEMM lives on the split path -- keep it
+    // there (measured faster than the bf16 fused kernel at 2x MFMA rate)
+    const bool split_auto = (B * H >= 2097152) && (!fill2ph || g_fp8_fwd);
     // W_h^T (4H, H): one transposed copy per layer call so each lane's MFMA
     // B-fragment is a contiguous 16-byte load (fused path only).
     const bool need_wt = g_persistent_enabled
