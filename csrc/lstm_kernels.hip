@@ -1328,6 +1328,212 @@ __global__ __launch_bounds__(512, 2) void k_lstm_rec_fwd_2ph(
 }
 
 // ---------------------------------------------------------------------------
+// OVERLAP variant [r2 experiment]: 128x256 tile, BK=32, 256 threads
+// (4 waves, 1M x 4N), THREE 24-KiB LDS buffers = 72 KiB/block -> TWO
+// blocks co-resident per CU (VGPR ~200 = 2 waves/SIMD).  While one
+// block's waves sit in an MFMA segment, the partner block's waves run
+// their HBM-bound staging / LDS-bounce epilogue -- the mem-beside-matrix
+// pairing the pipeline-locked 512-thread kernels cannot get.  One phase
+// per K-tile: {12 ds_reads; stage tile j+2 (6 glds, 3rd buffer);
+// s_barrier; lgkmcnt(0); 32 MFMA; vmcnt(6) boundary; s_barrier}.
+// ---------------------------------------------------------------------------
+template <bool RAW_DEBUG, bool XCD_SWZ>
+__global__ __launch_bounds__(256, 2) void k_lstm_rec_fwd_ov(
+    const __hip_bfloat16* __restrict__ h_prev, long hp_stride,
+    const __hip_bfloat16* __restrict__ w_t,     // (4H, H) row-major = W_h^T
+    const __hip_bfloat16* __restrict__ pre, long pre_stride,   // (B, 4H)
+    const float* __restrict__ c_prev, long cp_stride,
+    __hip_bfloat16* __restrict__ h_out, long ho_stride,
+    float* __restrict__ c_out, long co_stride,
+    __hip_bfloat16* __restrict__ acts, long act_stride,
+    int B, int H)
+{
+    constexpr int BK = 32;
+    constexpr int ABYTES = 128 * 64;      // 8 KiB (128 rows x 32 k bf16)
+    constexpr int BBYTES = 256 * 64;      // 16 KiB (256 gate-cols x 32 k)
+    constexpr int BUFB = ABYTES + BBYTES; // 24 KiB; x3 buffers = 72 KiB
+    extern __shared__ __attribute__((aligned(16))) char smem[];
+
+    const int tid = threadIdx.x;
+    const int wave = tid >> 6;
+    const int lane = tid & 63;
+    const int lr = lane & 15;
+    const int lk = lane >> 4;
+    const int wn = wave;               // 4 N groups of 16 h-cols
+
+    int m_blk = blockIdx.x, n_blk = blockIdx.y;
+    if (XCD_SWZ) {
+        const int nm = gridDim.x, nwg = gridDim.x * gridDim.y;
+        const int bid = blockIdx.x + blockIdx.y * nm;
+        const int q = nwg >> 3, r = nwg & 7, xcd = bid & 7, loc = bid >> 3;
+        const int swz = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q)
+                        + loc;
+        m_blk = swz % nm;
+        n_blk = swz / nm;
+    }
+    const int m_base = m_blk * 128;
+    const long n_col0 = (long)n_blk * 64;
+    const int nsteps = H / BK;
+
+    // stage one whole BK=32 K-tile (A 8 KiB + B 16 KiB, 6 glds per thread)
+    auto stage = [&](int buf, int k0) {
+        char* const ab = smem + buf * BUFB;
+        char* const bb = ab + ABYTES;
+#pragma unroll
+        for (int s2 = 0; s2 < 2; ++s2) {
+            const int ci = s2 * 256 + tid;
+            const int r = ci >> 2, ch = ci & 3;
+            const void* src = h_prev + (long)(m_base + r) * hp_stride + k0
+                              + ((ch ^ swz64(r)) * 8);
+            __builtin_amdgcn_global_load_lds(
+                (const __attribute__((address_space(1))) void*)src,
+                (__attribute__((address_space(3))) void*)(
+                    ab + s2 * 4096 + wave * 1024), 16, 0, 0);
+        }
+#pragma unroll
+        for (int s2 = 0; s2 < 4; ++s2) {
+            const int ci = s2 * 256 + tid;
+            const int gc = ci >> 2, ch = ci & 3;
+            const long grow = (long)(gc >> 6) * H + n_col0 + (gc & 63);
+            const void* src = w_t + grow * H + k0 + ((ch ^ swz64(gc)) * 8);
+            __builtin_amdgcn_global_load_lds(
+                (const __attribute__((address_space(1))) void*)src,
+                (__attribute__((address_space(3))) void*)(
+                    bb + s2 * 4096 + wave * 1024), 16, 0, 0);
+        }
+    };
+
+    f32x4 acc[4][8];
+#pragma unroll
+    for (int g = 0; g < 4; ++g)
+#pragma unroll
+        for (int mf = 0; mf < 8; ++mf) acc[g][mf] = f32x4{0, 0, 0, 0};
+
+    stage(0, 0);
+    if (nsteps > 1) stage(1, BK);
+    asm volatile("s_waitcnt vmcnt(6)" ::: "memory");  // tile 0 landed
+    __builtin_amdgcn_s_barrier();
+
+    for (int j = 0; j < nsteps; ++j) {
+        const int buf = j % 3;
+        const char* ab = smem + buf * BUFB;
+        const char* bb = ab + ABYTES;
+        bf16x8 wfr[4], hfr[8];
+#pragma unroll
+        for (int g = 0; g < 4; ++g) {
+            const int gc = g * 64 + wn * 16 + lr;
+            wfr[g] = *(const bf16x8*)(bb + gc * 64 + ((lk ^ swz64(gc)) * 16));
+        }
+#pragma unroll
+        for (int mi = 0; mi < 8; ++mi) {
+            const int row = mi * 16 + lr;
+            hfr[mi] = *(const bf16x8*)(ab + row * 64 + ((lk ^ swz64(row)) * 16));
+        }
+        if (j + 2 < nsteps) stage((j + 2) % 3, (j + 2) * BK);
+        __builtin_amdgcn_s_barrier();
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+#pragma unroll
+        for (int g = 0; g < 4; ++g)
+#pragma unroll
+            for (int mi = 0; mi < 8; ++mi)
+                acc[g][mi] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    wfr[g], hfr[mi], acc[g][mi], 0, 0, 0);
+        if (j + 1 < nsteps) {
+            if (j + 2 < nsteps)
+                asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+            else
+                asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        }
+        __builtin_amdgcn_s_barrier();
+    }
+
+    // Epilogue. RAW: direct stores; CELL: LDS-bounce (own 72 KiB now free)
+    if (RAW_DEBUG) {
+        const int hcol0 = (int)n_col0 + wn * 16 + lk * 4;
+#pragma unroll
+        for (int nf = 0; nf < 8; ++nf) {
+            const long row = m_base + nf * 16 + lr;
+            __hip_bfloat16* ap = acts + row * act_stride + hcol0;
+#pragma unroll
+            for (int g = 0; g < 4; ++g) {
+                const bf16x4 pv = *(const bf16x4*)(
+                    pre + row * pre_stride + (long)g * H + hcol0);
+                bf16x4 o;
+#pragma unroll
+                for (int jj = 0; jj < 4; ++jj)
+                    o[jj] = (__bf16)(acc[g][nf][jj] + (float)pv[jj]);
+                *(bf16x4*)(ap + (long)g * H) = o;
+            }
+        }
+        return;
+    }
+    float* const img = (float*)smem;        // [32 rows][256 gate-cols] fp32
+#pragma unroll
+    for (int rd = 0; rd < 4; ++rd) {
+        if (rd) __syncthreads();
+#pragma unroll
+        for (int nn = 0; nn < 2; ++nn) {
+            const int nf = rd * 2 + nn;
+            const int irow = nn * 16 + lr;
+#pragma unroll
+            for (int g = 0; g < 4; ++g) {
+                const int chunk = (g * 16 + wn * 4 + lk) ^ ((irow & 3) << 2);
+                *(f32x4*)(img + irow * 256 + chunk * 4) = acc[g][nf];
+            }
+        }
+        __syncthreads();
+        // 32 rows x 8 h-col chunks = 256 tasks, one per thread
+        const int irow = tid >> 3;
+        const int hc = (tid & 7) * 8;
+        const long grow = m_base + (rd * 2 + (irow >> 4)) * 16 + (irow & 15);
+        float raw[4][8];
+#pragma unroll
+        for (int g = 0; g < 4; ++g) {
+#pragma unroll
+            for (int hh = 0; hh < 2; ++hh) {
+                const int chunk = (g * 16 + (hc >> 2) + hh)
+                                  ^ ((irow & 3) << 2);
+                const f32x4 v = *(const f32x4*)(img + irow * 256 + chunk * 4);
+#pragma unroll
+                for (int jj = 0; jj < 4; ++jj) raw[g][hh * 4 + jj] = v[jj];
+            }
+            const bf16x8 pv = *(const bf16x8*)(
+                pre + grow * pre_stride + (long)g * H + n_col0 + hc);
+#pragma unroll
+            for (int jj = 0; jj < 8; ++jj) raw[g][jj] += (float)pv[jj];
+        }
+        const long hcol = n_col0 + hc;
+        const f32x4 cp0 = *(const f32x4*)(c_prev + grow * cp_stride + hcol);
+        const f32x4 cp1 = *(const f32x4*)(c_prev + grow * cp_stride + hcol + 4);
+        bf16x8 oi, of, og, oo, oh;
+        f32x4 oc0, oc1;
+#pragma unroll
+        for (int jj = 0; jj < 8; ++jj) {
+            const float gi = sigmoidf_(raw[0][jj]);
+            const float gf = sigmoidf_(raw[1][jj]);
+            const float gg = tanhf_(raw[2][jj]);
+            const float go = sigmoidf_(raw[3][jj]);
+            const float cpv = jj < 4 ? cp0[jj] : cp1[jj - 4];
+            const float c = gf * cpv + gi * gg;
+            if (jj < 4) oc0[jj] = c; else oc1[jj - 4] = c;
+            oh[jj] = (__bf16)(go * tanhf_(c));
+            oi[jj] = (__bf16)gi;
+            of[jj] = (__bf16)gf;
+            og[jj] = (__bf16)gg;
+            oo[jj] = (__bf16)go;
+        }
+        *(bf16x8*)(h_out + grow * ho_stride + hcol) = oh;
+        *(f32x4*)(c_out + grow * co_stride + hcol) = oc0;
+        *(f32x4*)(c_out + grow * co_stride + hcol + 4) = oc1;
+        __hip_bfloat16* ap = acts + grow * act_stride + hcol;
+        *(bf16x8*)(ap)               = oi;
+        *(bf16x8*)(ap + (long)H)     = of;
+        *(bf16x8*)(ap + (long)2 * H) = og;
+        *(bf16x8*)(ap + (long)3 * H) = oo;
+    }
+}
+
+// ---------------------------------------------------------------------------
 // FUSED BACKWARD STEP [K5a+K4]: dh_t = dgates_{t+1} @ W_h^T on MFMA, with
 // the pointwise backward of step t fused into the epilogue -- one kernel
 // replaces the hipBLASLt dh GEMM + k_lstm_pw_bwd pair per timestep AND
@@ -2480,6 +2686,30 @@ void launch_lstm_rec_fwd_v(
     void* acts, long act_stride, int B, int H, int raw_debug, int variant,
     hipStream_t stream)
 {
+    // overlap path (bit12): 128x256 BK=32, 256 threads, 2 blocks/CU
+    if ((variant & 4096) && (B % 128 == 0)) {
+        const size_t smem = 3 * (128 * 64 + 256 * 64);  // 72 KiB
+        dim3 grid(B / 128, H / 64), block(256);
+#define OVLAUNCH(RAW_, SW_)                                                   \
+    do {                                                                      \
+        hipFuncSetAttribute((const void*)(k_lstm_rec_fwd_ov<RAW_, SW_>),      \
+                            hipFuncAttributeMaxDynamicSharedMemorySize,       \
+                            (int)smem);                                       \
+        hipLaunchKernelGGL((k_lstm_rec_fwd_ov<RAW_, SW_>), grid, block,       \
+                           smem, stream,                                      \
+            (const __hip_bfloat16*)h_prev, hp_stride,                         \
+            (const __hip_bfloat16*)w_t, (const __hip_bfloat16*)pre,           \
+            pre_stride, c_prev, cp_stride, (__hip_bfloat16*)h_out, ho_stride, \
+            c_out, co_stride, (__hip_bfloat16*)acts, act_stride, B, H);       \
+    } while (0)
+        if (raw_debug) {
+            if (variant & 8) OVLAUNCH(true, true); else OVLAUNCH(true, false);
+        } else {
+            if (variant & 8) OVLAUNCH(false, true); else OVLAUNCH(false, false);
+        }
+#undef OVLAUNCH
+        return;
+    }
     // 2-phase merged path (bit10): BMx256 BK=64, 32-MFMA segments;
     // bit11 selects the BM=128 row tile (grids that under-fill at BM=256)
     if ((variant & 1024)

@@ -74,7 +74,8 @@ def main():
               33, 41,           # MT256 2buf-BK64
               512, 520,         # 8-phase BK64 counted-vmcnt (8=XCD swizzle)
               1024, 1032,       # 2-phase merged BK64 (8=XCD swizzle)
-              3072, 3075]:      # 2-phase BM=128 row tile (3=TUNE3)
+              3072, 3075,       # 2-phase BM=128 row tile (3=TUNE3)
+              4096, 4104]:      # overlap 128x256 BK32, 2 blocks/CU
         ext.set_fwd_variant(v)
         h_out, c_out, acts_out = ext.lstm_rec_fwd_step(h_prev, w_h, pre1,
                                                        c_prev, False)
