@@ -1256,8 +1256,6 @@ __global__ __launch_bounds__(512, 2) void k_lstm_rec_fwd_2ph(
         return;
     }
     float* const img = (float*)smem;        // [64 rows][256 gate-cols] fp32
-    const int wr_col = wn * 16 + lk * 4;    // this lane's 4 cols (per gate)
-    (void)wr_col;
 #pragma unroll
     for (int rd = 0; rd < MI / 2; ++rd) {
         if (rd) __syncthreads();            // previous round's reads done
