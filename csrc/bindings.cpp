@@ -55,6 +55,11 @@ int launch_lstm_dh_pw_bwd(const void*, long, const void*, long, const void*,
                           long, const void*, long, const float*, long,
                           const float*, long, void*, long, float*, long,
                           void*, long, int, int, int, int, hipStream_t);
+int launch_lstm_dh_pw_bwd_ov(const void*, long, const void*, long,
+                             const void*, long, const void*, long,
+                             const float*, long, const float*, long, void*,
+                             long, float*, long, void*, long, int, int, int,
+                             int, hipStream_t);
 }
 
 namespace {
@@ -117,8 +122,12 @@ std::vector<Tensor> lstm_seq_fwd(Tensor pre, Tensor w_h, Tensor h0, Tensor c0,
     // the 256-CU chip (measured 204 vs 258 us/step at B=16384 H=1024);
     // below that, split wins at B*H >= 2M and the BM64/BM32 fused kernels
     // win smaller still (round-1 sweeps).
-    const bool fill2ph = (B % 256 == 0) && (H % 64 == 0)
-                         && (B / 256) * (H / 64) >= 256;
+    // The fused 256-col-tile kernels win where (a) their grid fills the
+    // chip AND (b) K=H is small enough that hipBLASLt's deep-K advantage
+    // doesn't apply (measured: fused wins H=1024 for B=4096..32768, ties
+    // H=2048 at B=2048, LOSES H=8192 -- library GEMM ~1 PF/s at K>=2048).
+    const bool fill2ph = (B % 128 == 0) && (H % 64 == 0) && (H <= 1536)
+                         && (B / 128) * (H / 64) >= 512;
     // fp8 mode's e4m3 recurrent GEMM lives on the split path -- keep it
     // there (measured faster than the bf16 fused kernel at 2x MFMA rate)
     const bool split_auto = (B * H >= 2097152) && (!fill2ph || g_fp8_fwd);
@@ -289,10 +298,13 @@ std::vector<Tensor> lstm_seq_bwd(c10::optional<Tensor> dh_seq_opt,
     // per timestep, fp32 dh carry in registers, and no W_h^T transpose
     // copy (the kernel reads the (H,4H) layout directly).
     // auto (-1): only where the (B/256, H/256) grid fills enough of the
-    // 256-CU chip; explicit 1 forces it on any eligible shape (tests).
+    // 256-CU chip AND H is in the measured-parity regime (H=1024: tie
+    // with split + fewer launches; large H unmeasured -> stay split);
+    // explicit 1 forces it on any eligible shape (tests).
     const bool fused = (B % 256 == 0) && (H % 256 == 0)
         && (g_fused_bwd > 0
-            || (g_fused_bwd < 0 && (B / 256) * (H / 256) >= 192));
+            || (g_fused_bwd < 0 && H <= 1536
+                && (B / 256) * (H / 256) >= 192));
     Tensor w_h_t;
     if (!fused) {
         // split fallback: W_h^T as a CONTIGUOUS copy (one 8 MB transpose
@@ -321,14 +333,25 @@ std::vector<Tensor> lstm_seq_bwd(c10::optional<Tensor> dh_seq_opt,
         if (fused && !first) {
             // dgates of the step handled in the PREVIOUS iteration
             const long t_nx = reverse ? (t - 1) : (t + 1);
-            int rc = launch_lstm_dh_pw_bwd(
+            // overlap (2 blocks/CU) kernel first; 512-thread kernel as
+            // the fallback for shapes it cannot tile (B % 128 != 0)
+            int rc = launch_lstm_dh_pw_bwd_ov(
                 dg_p + 2 * t_nx * B * H4, H4, w_hc.data_ptr(), H4,
                 has_up ? up_p + 2 * t * B * H : nullptr, H,
                 act_p + 2 * t * B * H4, H4,
                 cs_p + t * B * H, H, cprev_p, H,
                 dg_p + 2 * t * B * H4, H4,
                 (float*)dc.data_ptr(), H,
-                nullptr, 0, (int)B, (int)H, /*raw=*/0, /*swz=*/1, stream);
+                nullptr, 0, (int)B, (int)H, /*raw=*/0, /*swz=*/0, stream);
+            if (rc != 0)
+                rc = launch_lstm_dh_pw_bwd(
+                    dg_p + 2 * t_nx * B * H4, H4, w_hc.data_ptr(), H4,
+                    has_up ? up_p + 2 * t * B * H : nullptr, H,
+                    act_p + 2 * t * B * H4, H4,
+                    cs_p + t * B * H, H, cprev_p, H,
+                    dg_p + 2 * t * B * H4, H4,
+                    (float*)dc.data_ptr(), H,
+                    nullptr, 0, (int)B, (int)H, /*raw=*/0, /*swz=*/1, stream);
             TORCH_CHECK(rc == 0, "fused bwd launch failed");
             continue;
         }
@@ -357,12 +380,19 @@ std::vector<Tensor> lstm_seq_bwd(c10::optional<Tensor> dh_seq_opt,
     if (fused) {
         // chain-end dh0 = dgates_{t_last} @ W_h^T via the RAW fused kernel
         const long t0 = reverse ? (T - 1) : 0;
-        int rc = launch_lstm_dh_pw_bwd(
+        int rc = launch_lstm_dh_pw_bwd_ov(
             dg_p + 2 * t0 * B * H4, H4, w_hc.data_ptr(), H4,
             nullptr, 0, nullptr, 0, nullptr, 0, nullptr, 0,
             nullptr, 0, nullptr, 0,
-            dh_buf.data_ptr(), H, (int)B, (int)H, /*raw=*/1, /*swz=*/1,
+            dh_buf.data_ptr(), H, (int)B, (int)H, /*raw=*/1, /*swz=*/0,
             stream);
+        if (rc != 0)
+            rc = launch_lstm_dh_pw_bwd(
+                dg_p + 2 * t0 * B * H4, H4, w_hc.data_ptr(), H4,
+                nullptr, 0, nullptr, 0, nullptr, 0, nullptr, 0,
+                nullptr, 0, nullptr, 0,
+                dh_buf.data_ptr(), H, (int)B, (int)H, /*raw=*/1, /*swz=*/1,
+                stream);
         TORCH_CHECK(rc == 0, "fused bwd raw launch failed");
     }
     return {dgates, dh_buf, dc};  // (dgates, dh0, dc0)

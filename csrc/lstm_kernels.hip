@@ -1799,6 +1799,252 @@ __global__ __launch_bounds__(512, 2) void k_lstm_dh_pw_bwd(
     }
 }
 
+// Overlap variant of the fused backward step: 128x256 dh tile, BK=32,
+// 256 threads, 3 LDS buffers (72 KiB) -> 2 blocks/CU so the pointwise
+// epilogue's HBM traffic hides under the partner block's MFMA segments.
+// Same operand mapping as k_lstm_dh_pw_bwd.
+template <bool RAW_DEBUG, bool XCD_SWZ>
+__global__ __launch_bounds__(256, 2) void k_lstm_dh_pw_bwd_ov(
+    const __hip_bfloat16* __restrict__ dg_next, long dgn_stride,  // (B,4H)
+    const __hip_bfloat16* __restrict__ w_h, long wh_stride,       // (H,4H)
+    const __hip_bfloat16* __restrict__ dh_up, long up_stride,     // nullable
+    const __hip_bfloat16* __restrict__ acts, long act_stride,     // (B,4H)
+    const float* __restrict__ c_cur, long cc_stride,
+    const float* __restrict__ c_prev, long cp_stride,
+    __hip_bfloat16* __restrict__ dgates_out, long dgo_stride,     // (B,4H)
+    float* __restrict__ dc, long dc_stride,                       // in-place
+    __hip_bfloat16* __restrict__ dh_out, long dho_stride,         // RAW only
+    int B, int H)
+{
+    constexpr int BK = 32;
+    constexpr int ABYTES = 128 * 64;
+    constexpr int BBYTES = 256 * 64;
+    constexpr int BUFB = ABYTES + BBYTES;
+    extern __shared__ __attribute__((aligned(16))) char smem[];
+
+    const int tid = threadIdx.x;
+    const int wave = tid >> 6;
+    const int lane = tid & 63;
+    const int lr = lane & 15;
+    const int lk = lane >> 4;
+    const int wn = wave;
+
+    int m_blk = blockIdx.x, n_blk = blockIdx.y;
+    if (XCD_SWZ) {
+        const int nm = gridDim.x, nwg = gridDim.x * gridDim.y;
+        const int bid = blockIdx.x + blockIdx.y * nm;
+        const int q = nwg >> 3, r = nwg & 7, xcd = bid & 7, loc = bid >> 3;
+        const int swz = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q)
+                        + loc;
+        m_blk = swz % nm;
+        n_blk = swz / nm;
+    }
+    const int m_base = m_blk * 128;
+    const long n_col0 = (long)n_blk * 256;
+    const int K = 4 * H;
+    const int nsteps = K / BK;
+
+    auto stage = [&](int buf, int k0) {
+        char* const ab = smem + buf * BUFB;
+        char* const bb = ab + ABYTES;
+#pragma unroll
+        for (int s2 = 0; s2 < 2; ++s2) {
+            const int ci = s2 * 256 + tid;
+            const int r = ci >> 2, ch = ci & 3;
+            const void* src = dg_next + (long)(m_base + r) * dgn_stride + k0
+                              + ((ch ^ swz64(r)) * 8);
+            __builtin_amdgcn_global_load_lds(
+                (const __attribute__((address_space(1))) void*)src,
+                (__attribute__((address_space(3))) void*)(
+                    ab + s2 * 4096 + wave * 1024), 16, 0, 0);
+        }
+#pragma unroll
+        for (int s2 = 0; s2 < 4; ++s2) {
+            const int ci = s2 * 256 + tid;
+            const int gc = ci >> 2, ch = ci & 3;
+            const void* src = w_h + (n_col0 + gc) * wh_stride + k0
+                              + ((ch ^ swz64(gc)) * 8);
+            __builtin_amdgcn_global_load_lds(
+                (const __attribute__((address_space(1))) void*)src,
+                (__attribute__((address_space(3))) void*)(
+                    bb + s2 * 4096 + wave * 1024), 16, 0, 0);
+        }
+    };
+
+    f32x4 acc[4][8];
+#pragma unroll
+    for (int g = 0; g < 4; ++g)
+#pragma unroll
+        for (int mf = 0; mf < 8; ++mf) acc[g][mf] = f32x4{0, 0, 0, 0};
+
+    stage(0, 0);
+    if (nsteps > 1) stage(1, BK);
+    asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+
+    for (int j = 0; j < nsteps; ++j) {
+        const int buf = j % 3;
+        const char* ab = smem + buf * BUFB;
+        const char* bb = ab + ABYTES;
+        bf16x8 wfr[4], hfr[8];
+#pragma unroll
+        for (int g = 0; g < 4; ++g) {
+            const int gc = g * 64 + wn * 16 + lr;
+            wfr[g] = *(const bf16x8*)(bb + gc * 64 + ((lk ^ swz64(gc)) * 16));
+        }
+#pragma unroll
+        for (int mi = 0; mi < 8; ++mi) {
+            const int row = mi * 16 + lr;
+            hfr[mi] = *(const bf16x8*)(ab + row * 64 + ((lk ^ swz64(row)) * 16));
+        }
+        if (j + 2 < nsteps) stage((j + 2) % 3, (j + 2) * BK);
+        __builtin_amdgcn_s_barrier();
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+#pragma unroll
+        for (int g = 0; g < 4; ++g)
+#pragma unroll
+            for (int mi = 0; mi < 8; ++mi)
+                acc[g][mi] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    wfr[g], hfr[mi], acc[g][mi], 0, 0, 0);
+        if (j + 1 < nsteps) {
+            if (j + 2 < nsteps)
+                asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+            else
+                asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        }
+        __builtin_amdgcn_s_barrier();
+    }
+
+    if (RAW_DEBUG) {
+#pragma unroll
+        for (int cf = 0; cf < 4; ++cf) {
+            const long hcol0 = n_col0 + cf * 64 + wn * 16 + lk * 4;
+#pragma unroll
+            for (int nf = 0; nf < 8; ++nf) {
+                const long row = m_base + nf * 16 + lr;
+                bf16x4 o;
+#pragma unroll
+                for (int jj = 0; jj < 4; ++jj) o[jj] = (__bf16)acc[cf][nf][jj];
+                *(bf16x4*)(dh_out + row * dho_stride + hcol0) = o;
+            }
+        }
+        return;
+    }
+    float* const img = (float*)smem;        // [32 rows][256 dh-cols] fp32
+#pragma unroll
+    for (int rd = 0; rd < 4; ++rd) {
+        if (rd) __syncthreads();
+#pragma unroll
+        for (int nn = 0; nn < 2; ++nn) {
+            const int nf = rd * 2 + nn;
+            const int irow = nn * 16 + lr;
+#pragma unroll
+            for (int cf = 0; cf < 4; ++cf) {
+                const int chunk = (cf * 16 + wn * 4 + lk) ^ ((irow & 3) << 2);
+                *(f32x4*)(img + irow * 256 + chunk * 4) = acc[cf][nf];
+            }
+        }
+        __syncthreads();
+        // 32 rows x 32 8-col chunks = 1024 tasks, 4 per thread
+#pragma unroll
+        for (int sub = 0; sub < 4; ++sub) {
+            const int irow = sub * 8 + (tid >> 5);
+            const int hc = (tid & 31) * 8;
+            const long grow = m_base + (rd * 2 + (irow >> 4)) * 16
+                              + (irow & 15);
+            float dhv[8];
+#pragma unroll
+            for (int hh = 0; hh < 2; ++hh) {
+                const int chunk = ((hc >> 2) + hh) ^ ((irow & 3) << 2);
+                const f32x4 v = *(const f32x4*)(img + irow * 256 + chunk * 4);
+#pragma unroll
+                for (int jj = 0; jj < 4; ++jj) dhv[hh * 4 + jj] = v[jj];
+            }
+            const long hcol = n_col0 + hc;
+            if (dh_up) {
+                const bf16x8 uv = *(const bf16x8*)(
+                    dh_up + grow * up_stride + hcol);
+#pragma unroll
+                for (int jj = 0; jj < 8; ++jj) dhv[jj] += (float)uv[jj];
+            }
+            const f32x4 cc0 = *(const f32x4*)(c_cur + grow * cc_stride + hcol);
+            const f32x4 cc1 = *(const f32x4*)(c_cur + grow * cc_stride + hcol + 4);
+            const f32x4 cp0 = *(const f32x4*)(c_prev + grow * cp_stride + hcol);
+            const f32x4 cp1 = *(const f32x4*)(c_prev + grow * cp_stride + hcol + 4);
+            const f32x4 dv0 = *(const f32x4*)(dc + grow * dc_stride + hcol);
+            const f32x4 dv1 = *(const f32x4*)(dc + grow * dc_stride + hcol + 4);
+            const __hip_bfloat16* ap = acts + grow * act_stride + hcol;
+            const bf16x8 vi = *(const bf16x8*)(ap);
+            const bf16x8 vf = *(const bf16x8*)(ap + (long)H);
+            const bf16x8 vg = *(const bf16x8*)(ap + (long)2 * H);
+            const bf16x8 vo = *(const bf16x8*)(ap + (long)3 * H);
+            bf16x8 out_i, out_f, out_g, out_o;
+            f32x4 dc0, dc1;
+#pragma unroll
+            for (int jj = 0; jj < 8; ++jj) {
+                const float i = (float)vi[jj], f = (float)vf[jj];
+                const float g = (float)vg[jj], o = (float)vo[jj];
+                const float ccv = jj < 4 ? cc0[jj] : cc1[jj - 4];
+                const float cpv = jj < 4 ? cp0[jj] : cp1[jj - 4];
+                const float dcv = jj < 4 ? dv0[jj] : dv1[jj - 4];
+                const float tc = tanhf_(ccv);
+                const float d_o = dhv[jj] * tc;
+                float d_c = dcv + dhv[jj] * o * (1.0f - tc * tc);
+                const float d_i = d_c * g;
+                const float d_g = d_c * i;
+                const float d_f = d_c * cpv;
+                const float dco = d_c * f;
+                if (jj < 4) dc0[jj] = dco; else dc1[jj - 4] = dco;
+                out_i[jj] = (__bf16)(d_i * i * (1.0f - i));
+                out_f[jj] = (__bf16)(d_f * f * (1.0f - f));
+                out_g[jj] = (__bf16)(d_g * (1.0f - g * g));
+                out_o[jj] = (__bf16)(d_o * o * (1.0f - o));
+            }
+            __hip_bfloat16* dgp = dgates_out + grow * dgo_stride + hcol;
+            *(bf16x8*)(dgp)               = out_i;
+            *(bf16x8*)(dgp + (long)H)     = out_f;
+            *(bf16x8*)(dgp + (long)2 * H) = out_g;
+            *(bf16x8*)(dgp + (long)3 * H) = out_o;
+            *(f32x4*)(dc + grow * dc_stride + hcol) = dc0;
+            *(f32x4*)(dc + grow * dc_stride + hcol + 4) = dc1;
+        }
+    }
+}
+
+extern "C" int launch_lstm_dh_pw_bwd_ov(
+    const void* dg_next, long dgn_stride, const void* w_h, long wh_stride,
+    const void* dh_up, long up_stride, const void* acts, long act_stride,
+    const float* c_cur, long cc_stride, const float* c_prev, long cp_stride,
+    void* dgates_out, long dgo_stride, float* dc, long dc_stride,
+    void* dh_out, long dho_stride, int B, int H, int raw, int xcd_swz,
+    hipStream_t stream)
+{
+    if (B % 128 != 0 || H % 256 != 0) return -1;
+    const size_t smem = 3 * (128 * 64 + 256 * 64);
+    dim3 grid(B / 128, H / 256), block(256);
+#define DHOVLAUNCH(RAW_, SW_)                                                 \
+    do {                                                                      \
+        hipFuncSetAttribute((const void*)(k_lstm_dh_pw_bwd_ov<RAW_, SW_>),    \
+                            hipFuncAttributeMaxDynamicSharedMemorySize,       \
+                            (int)smem);                                       \
+        hipLaunchKernelGGL((k_lstm_dh_pw_bwd_ov<RAW_, SW_>), grid, block,     \
+                           smem, stream,                                      \
+            (const __hip_bfloat16*)dg_next, dgn_stride,                       \
+            (const __hip_bfloat16*)w_h, wh_stride,                            \
+            (const __hip_bfloat16*)dh_up, up_stride,                          \
+            (const __hip_bfloat16*)acts, act_stride, c_cur, cc_stride,        \
+            c_prev, cp_stride, (__hip_bfloat16*)dgates_out, dgo_stride,       \
+            dc, dc_stride, (__hip_bfloat16*)dh_out, dho_stride, B, H);        \
+    } while (0)
+    if (raw) {
+        if (xcd_swz) DHOVLAUNCH(true, true); else DHOVLAUNCH(true, false);
+    } else {
+        if (xcd_swz) DHOVLAUNCH(false, true); else DHOVLAUNCH(false, false);
+    }
+#undef DHOVLAUNCH
+    return 0;
+}
+
 extern "C" int launch_lstm_dh_pw_bwd(
     const void* dg_next, long dgn_stride, const void* w_h, long wh_stride,
     const void* dh_up, long up_stride, const void* acts, long act_stride,
@@ -2961,8 +3207,12 @@ void launch_lstm_rec_fwd(
     hipStream_t stream)
 {
     int v = g_fwd_variant;
-    if (v < 0) v = (B % 256 == 0 && (B / 256) * (H / 64) >= 256)
-                     ? 1027                        // 2ph TUNE3 (kbench r2)
+    // auto: the 2-blocks/CU overlap kernel where its grid fills the chip
+    // and K=H is in the measured-win regime (see bindings.cpp fill2ph);
+    // the 512-thread 2ph (1027) stays available for A/B.
+    if (v < 0) v = (B % 128 == 0 && H <= 1536
+                    && (B / 128) * (H / 64) >= 512)
+                     ? 4096                        // overlap kernel (kbench r2)
                  : (B >= 2048) ? 10 : 14;
     launch_lstm_rec_fwd_v(h_prev, hp_stride, w_t, pre, pre_stride, c_prev,
                           cp_stride, h_out, ho_stride, c_out, co_stride, acts,
