@@ -333,25 +333,17 @@ std::vector<Tensor> lstm_seq_bwd(c10::optional<Tensor> dh_seq_opt,
         if (fused && !first) {
             // dgates of the step handled in the PREVIOUS iteration
             const long t_nx = reverse ? (t - 1) : (t + 1);
-            // overlap (2 blocks/CU) kernel first; 512-thread kernel as
-            // the fallback for shapes it cannot tile (B % 128 != 0)
-            int rc = launch_lstm_dh_pw_bwd_ov(
+            // 512-thread 256x256 kernel (measured: ties split and beats
+            // the 2-blocks/CU ov variant, whose 128-row tiles double the
+            // W-panel traffic at K=4H -- 248 vs 236 us/step at headline)
+            int rc = launch_lstm_dh_pw_bwd(
                 dg_p + 2 * t_nx * B * H4, H4, w_hc.data_ptr(), H4,
                 has_up ? up_p + 2 * t * B * H : nullptr, H,
                 act_p + 2 * t * B * H4, H4,
                 cs_p + t * B * H, H, cprev_p, H,
                 dg_p + 2 * t * B * H4, H4,
                 (float*)dc.data_ptr(), H,
-                nullptr, 0, (int)B, (int)H, /*raw=*/0, /*swz=*/0, stream);
-            if (rc != 0)
-                rc = launch_lstm_dh_pw_bwd(
-                    dg_p + 2 * t_nx * B * H4, H4, w_hc.data_ptr(), H4,
-                    has_up ? up_p + 2 * t * B * H : nullptr, H,
-                    act_p + 2 * t * B * H4, H4,
-                    cs_p + t * B * H, H, cprev_p, H,
-                    dg_p + 2 * t * B * H4, H4,
-                    (float*)dc.data_ptr(), H,
-                    nullptr, 0, (int)B, (int)H, /*raw=*/0, /*swz=*/1, stream);
+                nullptr, 0, (int)B, (int)H, /*raw=*/0, /*swz=*/1, stream);
             TORCH_CHECK(rc == 0, "fused bwd launch failed");
             continue;
         }
@@ -380,19 +372,12 @@ std::vector<Tensor> lstm_seq_bwd(c10::optional<Tensor> dh_seq_opt,
     if (fused) {
         // chain-end dh0 = dgates_{t_last} @ W_h^T via the RAW fused kernel
         const long t0 = reverse ? (T - 1) : 0;
-        int rc = launch_lstm_dh_pw_bwd_ov(
+        int rc = launch_lstm_dh_pw_bwd(
             dg_p + 2 * t0 * B * H4, H4, w_hc.data_ptr(), H4,
             nullptr, 0, nullptr, 0, nullptr, 0, nullptr, 0,
             nullptr, 0, nullptr, 0,
-            dh_buf.data_ptr(), H, (int)B, (int)H, /*raw=*/1, /*swz=*/0,
+            dh_buf.data_ptr(), H, (int)B, (int)H, /*raw=*/1, /*swz=*/1,
             stream);
-        if (rc != 0)
-            rc = launch_lstm_dh_pw_bwd(
-                dg_p + 2 * t0 * B * H4, H4, w_hc.data_ptr(), H4,
-                nullptr, 0, nullptr, 0, nullptr, 0, nullptr, 0,
-                nullptr, 0, nullptr, 0,
-                dh_buf.data_ptr(), H, (int)B, (int)H, /*raw=*/1, /*swz=*/1,
-                stream);
         TORCH_CHECK(rc == 0, "fused bwd raw launch failed");
     }
     return {dgates, dh_buf, dc};  // (dgates, dh0, dc0)
