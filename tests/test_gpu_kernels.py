@@ -63,7 +63,7 @@ def test_rec_fwd_step_full(ext, B, H):
     assert (acts.float() - acts_ref).abs().max().item() < 0.02
 
 
-@pytest.mark.parametrize("variant", [512, 520])
+@pytest.mark.parametrize("variant", [512, 520, 1027, 4096])
 @pytest.mark.parametrize("B,H", [(256, 64), (256, 128), (256, 256),
                                  (512, 1024), (1024, 1024)])
 def test_rec_fwd_8phase_variants(ext, variant, B, H):
