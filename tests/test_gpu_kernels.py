@@ -174,6 +174,29 @@ def test_seq_bwd_fused_vs_ref(ext, reverse, B, T, H):
     assert (dh0.float() - dh0_s.float()).abs().max().item() < 0.05
 
 
+def test_seq_bwd_fused_none_dh_up(ext):
+    """Fused backward with NO upstream per-step grads (loss on hT only):
+    dh_up is null inside k_lstm_dh_pw_bwd for every step."""
+    from lstm_tensorspark_amd.ops import functional
+
+    B, T, H = 256, 4, 256
+    pre, w_h, h0, c0 = seq_inputs(B, T, H, seed=7)
+    h_r, c_r, a_r, _, _ = functional.ref_seq_fwd(pre, w_h, h0, c0, False)
+    d_hT = (torch.randn(B, H) * 0.1).to("cuda").bfloat16()
+    prev = ext.get_fused_bwd()
+    try:
+        ext.set_fused_bwd(1)
+        dg, dh0, dc0 = ext.lstm_seq_bwd(None, d_hT, None, w_h, a_r, c_r,
+                                        c0, False)
+    finally:
+        ext.set_fused_bwd(prev)
+    dg_r, dh0_r, dc0_r = functional.ref_seq_bwd(None, d_hT, None, w_h, a_r,
+                                                c_r, c0, False)
+    assert (dg.float() - dg_r.float()).abs().max().item() < 0.05
+    assert (dh0.float() - dh0_r.float()).abs().max().item() < 0.05
+    assert (dc0 - dc0_r).abs().max().item() < 0.05
+
+
 def test_seq_bwd_none_grads(ext):
     from lstm_tensorspark_amd.ops import functional
 
