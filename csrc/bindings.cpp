@@ -75,8 +75,10 @@ bool g_persistent_enabled = false;  // per-step glds path is faster (kbench6)
 // fused wins to B=512 (18.7 vs 22.0 us), ~tie at B=1024, split wins from
 // B=2048; at H=2048 split wins from B=1024: 1107 vs 1209 ms/step).
 int g_split_fwd = -1;
-// Fused backward step: -1/1 = on where eligible (B%256==0, H%256==0),
-// 0 = off (split hipBLASLt dh GEMM + k_lstm_pw_bwd).
+// Fused backward step: -1 auto (on where eligible AND H in the measured
+// regime), 1 = force the 512-thread kernel on any eligible shape,
+// 2 = force the 2-blocks/CU ov variant (A/B only: measured slower --
+// docs/KERNELS.md), 0 = off (split hipBLASLt dh GEMM + k_lstm_pw_bwd).
 int g_fused_bwd = -1;
 // FP8 recurrent GEMM (experimental, MI355X 2x-bf16 MFMA rate): quantize
 // h (tanh-bounded, FIXED scale 448) and W_h (per-call amax scale) to OCP
@@ -336,7 +338,18 @@ std::vector<Tensor> lstm_seq_bwd(c10::optional<Tensor> dh_seq_opt,
             // 512-thread 256x256 kernel (measured: ties split and beats
             // the 2-blocks/CU ov variant, whose 128-row tiles double the
             // W-panel traffic at K=4H -- 248 vs 236 us/step at headline)
-            int rc = launch_lstm_dh_pw_bwd(
+            int rc = -1;
+            if (g_fused_bwd == 2)
+                rc = launch_lstm_dh_pw_bwd_ov(
+                    dg_p + 2 * t_nx * B * H4, H4, w_hc.data_ptr(), H4,
+                    has_up ? up_p + 2 * t * B * H : nullptr, H,
+                    act_p + 2 * t * B * H4, H4,
+                    cs_p + t * B * H, H, cprev_p, H,
+                    dg_p + 2 * t * B * H4, H4,
+                    (float*)dc.data_ptr(), H,
+                    nullptr, 0, (int)B, (int)H, /*raw=*/0, /*swz=*/0,
+                    stream);
+            if (rc != 0) rc = launch_lstm_dh_pw_bwd(
                 dg_p + 2 * t_nx * B * H4, H4, w_hc.data_ptr(), H4,
                 has_up ? up_p + 2 * t * B * H : nullptr, H,
                 act_p + 2 * t * B * H4, H4,
