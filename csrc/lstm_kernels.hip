@@ -1416,6 +1416,10 @@ __global__ __launch_bounds__(256, 2) void k_lstm_rec_fwd_ov(
         const int buf = j % 3;
         const char* ab = smem + buf * BUFB;
         const char* bb = ab + ABYTES;
+        // glds first (TUNE2 lesson from the 2ph kernel: issuing the DMA
+        // before the ds_reads measured -11% together with dropping the
+        // partial lgkm wait); 3 buffers make the order trivially safe
+        if (j + 2 < nsteps) stage((j + 2) % 3, (j + 2) * BK);
         bf16x8 wfr[4], hfr[8];
 #pragma unroll
         for (int g = 0; g < 4; ++g) {
@@ -1427,7 +1431,6 @@ __global__ __launch_bounds__(256, 2) void k_lstm_rec_fwd_ov(
             const int row = mi * 16 + lr;
             hfr[mi] = *(const bf16x8*)(ab + row * 64 + ((lk ^ swz64(row)) * 16));
         }
-        if (j + 2 < nsteps) stage((j + 2) % 3, (j + 2) * BK);
         __builtin_amdgcn_s_barrier();
         asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
 #pragma unroll
@@ -1886,6 +1889,10 @@ __global__ __launch_bounds__(256, 2) void k_lstm_dh_pw_bwd_ov(
         const int buf = j % 3;
         const char* ab = smem + buf * BUFB;
         const char* bb = ab + ABYTES;
+        // glds first (TUNE2 lesson from the 2ph kernel: issuing the DMA
+        // before the ds_reads measured -11% together with dropping the
+        // partial lgkm wait); 3 buffers make the order trivially safe
+        if (j + 2 < nsteps) stage((j + 2) % 3, (j + 2) * BK);
         bf16x8 wfr[4], hfr[8];
 #pragma unroll
         for (int g = 0; g < 4; ++g) {
@@ -1897,7 +1904,6 @@ __global__ __launch_bounds__(256, 2) void k_lstm_dh_pw_bwd_ov(
             const int row = mi * 16 + lr;
             hfr[mi] = *(const bf16x8*)(ab + row * 64 + ((lk ^ swz64(row)) * 16));
         }
-        if (j + 2 < nsteps) stage((j + 2) % 3, (j + 2) * BK);
         __builtin_amdgcn_s_barrier();
         asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
 #pragma unroll
