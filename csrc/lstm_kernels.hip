@@ -1335,7 +1335,7 @@ __global__ __launch_bounds__(512, 2) void k_lstm_rec_fwd_2ph(
 // per K-tile: {12 ds_reads; stage tile j+2 (6 glds, 3rd buffer);
 // s_barrier; lgkmcnt(0); 32 MFMA; vmcnt(6) boundary; s_barrier}.
 // ---------------------------------------------------------------------------
-template <bool RAW_DEBUG, bool XCD_SWZ>
+template <bool RAW_DEBUG, bool XCD_SWZ, int TUNE = 0>
 __global__ __launch_bounds__(256, 2) void k_lstm_rec_fwd_ov(
     const __hip_bfloat16* __restrict__ h_prev, long hp_stride,
     const __hip_bfloat16* __restrict__ w_t,     // (4H, H) row-major = W_h^T
@@ -1416,10 +1416,9 @@ __global__ __launch_bounds__(256, 2) void k_lstm_rec_fwd_ov(
         const int buf = j % 3;
         const char* ab = smem + buf * BUFB;
         const char* bb = ab + ABYTES;
-        // glds first (TUNE2 lesson from the 2ph kernel: issuing the DMA
-        // before the ds_reads measured -11% together with dropping the
-        // partial lgkm wait); 3 buffers make the order trivially safe
-        if (j + 2 < nsteps) stage((j + 2) % 3, (j + 2) * BK);
+        // glds first by default (TUNE2 lesson from the 2ph kernel);
+        // TUNE&1 = reads-first for A/B.  3 buffers make either order safe.
+        if (!(TUNE & 1) && j + 2 < nsteps) stage((j + 2) % 3, (j + 2) * BK);
         bf16x8 wfr[4], hfr[8];
 #pragma unroll
         for (int g = 0; g < 4; ++g) {
@@ -1431,6 +1430,7 @@ __global__ __launch_bounds__(256, 2) void k_lstm_rec_fwd_ov(
             const int row = mi * 16 + lr;
             hfr[mi] = *(const bf16x8*)(ab + row * 64 + ((lk ^ swz64(row)) * 16));
         }
+        if ((TUNE & 1) && j + 2 < nsteps) stage((j + 2) % 3, (j + 2) * BK);
         __builtin_amdgcn_s_barrier();
         asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
 #pragma unroll
@@ -1806,7 +1806,7 @@ __global__ __launch_bounds__(512, 2) void k_lstm_dh_pw_bwd(
 // 256 threads, 3 LDS buffers (72 KiB) -> 2 blocks/CU so the pointwise
 // epilogue's HBM traffic hides under the partner block's MFMA segments.
 // Same operand mapping as k_lstm_dh_pw_bwd.
-template <bool RAW_DEBUG, bool XCD_SWZ>
+template <bool RAW_DEBUG, bool XCD_SWZ, int TUNE = 0>
 __global__ __launch_bounds__(256, 2) void k_lstm_dh_pw_bwd_ov(
     const __hip_bfloat16* __restrict__ dg_next, long dgn_stride,  // (B,4H)
     const __hip_bfloat16* __restrict__ w_h, long wh_stride,       // (H,4H)
@@ -1889,10 +1889,9 @@ __global__ __launch_bounds__(256, 2) void k_lstm_dh_pw_bwd_ov(
         const int buf = j % 3;
         const char* ab = smem + buf * BUFB;
         const char* bb = ab + ABYTES;
-        // glds first (TUNE2 lesson from the 2ph kernel: issuing the DMA
-        // before the ds_reads measured -11% together with dropping the
-        // partial lgkm wait); 3 buffers make the order trivially safe
-        if (j + 2 < nsteps) stage((j + 2) % 3, (j + 2) * BK);
+        // glds first by default (TUNE2 lesson from the 2ph kernel);
+        // TUNE&1 = reads-first for A/B.  3 buffers make either order safe.
+        if (!(TUNE & 1) && j + 2 < nsteps) stage((j + 2) % 3, (j + 2) * BK);
         bf16x8 wfr[4], hfr[8];
 #pragma unroll
         for (int g = 0; g < 4; ++g) {
@@ -2940,17 +2939,22 @@ void launch_lstm_rec_fwd_v(
     if ((variant & 4096) && (B % 128 == 0)) {
         const size_t smem = 3 * (128 * 64 + 256 * 64);  // 72 KiB
         dim3 grid(B / 128, H / 64), block(256);
-#define OVLAUNCH(RAW_, SW_)                                                   \
+#define OVLAUNCH_T(RAW_, SW_, TN_)                                            \
     do {                                                                      \
-        hipFuncSetAttribute((const void*)(k_lstm_rec_fwd_ov<RAW_, SW_>),      \
+        hipFuncSetAttribute((const void*)(k_lstm_rec_fwd_ov<RAW_, SW_, TN_>), \
                             hipFuncAttributeMaxDynamicSharedMemorySize,       \
                             (int)smem);                                       \
-        hipLaunchKernelGGL((k_lstm_rec_fwd_ov<RAW_, SW_>), grid, block,       \
+        hipLaunchKernelGGL((k_lstm_rec_fwd_ov<RAW_, SW_, TN_>), grid, block,  \
                            smem, stream,                                      \
             (const __hip_bfloat16*)h_prev, hp_stride,                         \
             (const __hip_bfloat16*)w_t, (const __hip_bfloat16*)pre,           \
             pre_stride, c_prev, cp_stride, (__hip_bfloat16*)h_out, ho_stride, \
             c_out, co_stride, (__hip_bfloat16*)acts, act_stride, B, H);       \
+    } while (0)
+#define OVLAUNCH(RAW_, SW_)                                                   \
+    do {                                                                      \
+        if (variant & 1) OVLAUNCH_T(RAW_, SW_, 1);                            \
+        else OVLAUNCH_T(RAW_, SW_, 0);                                        \
     } while (0)
         if (raw_debug) {
             if (variant & 8) OVLAUNCH(true, true); else OVLAUNCH(true, false);
@@ -2958,6 +2962,7 @@ void launch_lstm_rec_fwd_v(
             if (variant & 8) OVLAUNCH(false, true); else OVLAUNCH(false, false);
         }
 #undef OVLAUNCH
+#undef OVLAUNCH_T
         return;
     }
     // 2-phase merged path (bit10): BMx256 BK=64, 32-MFMA segments;
