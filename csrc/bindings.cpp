@@ -128,8 +128,8 @@ std::vector<Tensor> lstm_seq_fwd(Tensor pre, Tensor w_h, Tensor h0, Tensor c0,
     // chip AND (b) K=H is small enough that hipBLASLt's deep-K advantage
     // doesn't apply (measured: fused wins H=1024 for B=4096..32768, ties
     // H=2048 at B=2048, LOSES H=8192 -- library GEMM ~1 PF/s at K>=2048).
-    const bool fill2ph = (B % 128 == 0) && (H % 64 == 0) && (H <= 1536)
-                         && (B / 128) * (H / 64) >= 512;
+    const bool fill2ph = (B % 256 == 0) && (H % 64 == 0) && (H <= 1536)
+                         && (B / 256) * (H / 64) >= 256;
     // fp8 mode's e4m3 recurrent GEMM lives on the split path -- keep it
     // there (measured faster than the bf16 fused kernel at 2x MFMA rate)
     const bool split_auto = (B * H >= 2097152) && (!fill2ph || g_fp8_fwd);

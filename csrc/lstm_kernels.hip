@@ -3218,12 +3218,14 @@ void launch_lstm_rec_fwd(
     hipStream_t stream)
 {
     int v = g_fwd_variant;
-    // auto: the 2-blocks/CU overlap kernel where its grid fills the chip
-    // and K=H is in the measured-win regime (see bindings.cpp fill2ph);
-    // the 512-thread 2ph (1027) stays available for A/B.
-    if (v < 0) v = (B % 128 == 0 && H <= 1536
-                    && (B / 128) * (H / 64) >= 512)
-                     ? 4096                        // overlap kernel (kbench r2)
+    // auto: the 2ph kernel where its grid fills the chip and K=H is in
+    // the measured-win regime (see bindings.cpp fill2ph).  The 4-round
+    // interleaved A/B picked 2ph over the 2-blocks/CU ov variant
+    // (median 200.0 vs 208.6 us/step at the headline shape); ov stays
+    // as variants 4096/4097 for A/B.
+    if (v < 0) v = (B % 256 == 0 && H <= 1536
+                    && (B / 256) * (H / 64) >= 256)
+                     ? 1027                        // 2ph TUNE3 (kbench r2)
                  : (B >= 2048) ? 10 : 14;
     launch_lstm_rec_fwd_v(h_prev, hp_stride, w_t, pre, pre_stride, c_prev,
                           cp_stride, h_out, ho_stride, c_out, co_stride, acts,
