@@ -339,6 +339,7 @@ std::vector<Tensor> lstm_seq_bwd(c10::optional<Tensor> dh_seq_opt,
             // the 2-blocks/CU ov variant, whose 128-row tiles double the
             // W-panel traffic at K=4H -- 248 vs 236 us/step at headline)
             int rc = -1;
+            const int bswz = (g_fused_bwd == 3) ? 0 : 1;  // 3 = no-swizzle A/B
             if (g_fused_bwd == 2)
                 rc = launch_lstm_dh_pw_bwd_ov(
                     dg_p + 2 * t_nx * B * H4, H4, w_hc.data_ptr(), H4,
@@ -356,7 +357,7 @@ std::vector<Tensor> lstm_seq_bwd(c10::optional<Tensor> dh_seq_opt,
                 cs_p + t * B * H, H, cprev_p, H,
                 dg_p + 2 * t * B * H4, H4,
                 (float*)dc.data_ptr(), H,
-                nullptr, 0, (int)B, (int)H, /*raw=*/0, /*swz=*/1, stream);
+                nullptr, 0, (int)B, (int)H, /*raw=*/0, bswz, stream);
             TORCH_CHECK(rc == 0, "fused bwd launch failed");
             continue;
         }
