@@ -339,7 +339,10 @@ std::vector<Tensor> lstm_seq_bwd(c10::optional<Tensor> dh_seq_opt,
             // the 2-blocks/CU ov variant, whose 128-row tiles double the
             // W-panel traffic at K=4H -- 248 vs 236 us/step at headline)
             int rc = -1;
-            const int bswz = (g_fused_bwd == 3) ? 0 : 1;  // 3 = no-swizzle A/B
+            // interleaved A/B (4 rounds): swz-off 206.7 vs swz-on 212.3
+            // us/step median -> no-swizzle is the default; mode 3 kept as
+            // the swz-on A/B arm
+            const int bswz = (g_fused_bwd == 3) ? 1 : 0;
             if (g_fused_bwd == 2)
                 rc = launch_lstm_dh_pw_bwd_ov(
                     dg_p + 2 * t_nx * B * H4, H4, w_hc.data_ptr(), H4,
@@ -390,7 +393,7 @@ std::vector<Tensor> lstm_seq_bwd(c10::optional<Tensor> dh_seq_opt,
             dg_p + 2 * t0 * B * H4, H4, w_hc.data_ptr(), H4,
             nullptr, 0, nullptr, 0, nullptr, 0, nullptr, 0,
             nullptr, 0, nullptr, 0,
-            dh_buf.data_ptr(), H, (int)B, (int)H, /*raw=*/1, /*swz=*/1,
+            dh_buf.data_ptr(), H, (int)B, (int)H, /*raw=*/1, /*swz=*/0,
             stream);
         TORCH_CHECK(rc == 0, "fused bwd raw launch failed");
     }
