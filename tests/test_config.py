@@ -88,3 +88,22 @@ def test_fp8_dtype_plumbing_cpu_fallback():
     t = Trainer(cfg, device="cpu")
     info = t.train_step(t.data.batch(0))
     assert info["loss"] == info["loss"]
+
+
+def test_fp8_head_field_roundtrip(tmp_path):
+    """The r2 train.fp8_head knob survives dict/YAML round trips and the
+    CLI --set override path."""
+    import yaml
+
+    from lstm_tensorspark_amd.config import Config, ModelConfig, TrainConfig, DataConfig
+
+    c = Config(model=ModelConfig(), train=TrainConfig(fp8_head=True),
+               data=DataConfig())
+    d = c.to_dict()
+    assert d["train"]["fp8_head"] is True
+    c2 = Config.from_dict(d)
+    assert c2.train.fp8_head is True
+    p = tmp_path / "c.yaml"
+    p.write_text(yaml.safe_dump(d))
+    c3 = Config.from_dict(yaml.safe_load(p.read_text()))
+    assert c3.train.fp8_head is True
