@@ -67,13 +67,11 @@ namespace {
 using at::Tensor;
 
 bool g_persistent_enabled = false;  // per-step glds path is faster (kbench6)
-// Split forward: hipBLASLt recurrent GEMM + k_lstm_pw_fwd pointwise.  The
-// library GEMM runs the (B,H)x(H,4H) step shape ~1.9x faster than the best
-// fused hand kernel at B=4096 (908 vs ~475 TF/s), which more than pays for
-// the raw-gates round trip.  -1 = auto: split when the per-step GEMM is
-// big enough for the library to win, B*H >= 2M (measured at H=1024:
-// fused wins to B=512 (18.7 vs 22.0 us), ~tie at B=1024, split wins from
-// B=2048; at H=2048 split wins from B=1024: 1107 vs 1209 ms/step).
+// Split forward: hipBLASLt recurrent GEMM + k_lstm_pw_fwd pointwise.
+// -1 = auto (r2): the FUSED 2ph hand kernel where its grid fills the chip
+// and H is in its measured-win regime (fill2ph in lstm_seq_fwd); split
+// where B*H >= 2M otherwise; small-tile fused kernels below that.
+// 0 forces fused, 1 forces split (tests/autotune).
 int g_split_fwd = -1;
 // Fused backward step: -1 auto (on where eligible AND H in the measured
 // regime), 1 = force the 512-thread kernel on any eligible shape,

@@ -7,15 +7,17 @@ timestep replayed by Session.run, the whole-sequence forward is
   1. ONE library GEMM (hipBLASLt via torch.addmm, or _scaled_mm in fp8
      mode) for the input projection of ALL timesteps:
      pre = x.view(T*B, X) @ W_x + b                              [K1a]
-  2. a C++-driven time loop, one of two per-step paths chosen by shape
-     (csrc/bindings.cpp): the SPLIT path (hipBLASLt recurrent GEMM +
-     k_lstm_pw_fwd vectorized cell update; default at B*H >= 2M) or the
-     FUSED hand-written CDNA4 MFMA kernel (recurrent GEMM + gate
-     activations + cell update in one launch; default below)     [K1b+K2]
+  2. a C++-driven time loop, per-step path chosen by shape
+     (csrc/bindings.cpp): the FUSED hand-written CDNA4 MFMA kernel
+     (recurrent GEMM + gate activations + cell update in one launch;
+     r2 default at chip-filling grids with H <= 1536, e.g. the headline
+     shape) or the SPLIT path (hipBLASLt recurrent GEMM + k_lstm_pw_fwd
+     vectorized cell update; mid/deep-K shapes)                  [K1b+K2]
 
 and the backward is
-  3. a C++ time loop of a fused pointwise-backward kernel + recurrent
-     dgates @ W_h^T MFMA kernel                                  [K4+K5a]
+  3. a C++ time loop of the FUSED dh-GEMM + pointwise-backward MFMA
+     kernel (k_lstm_dh_pw_bwd; split hipBLASLt + k_lstm_pw_bwd
+     fallback at non-filling shapes)                             [K4+K5a]
   4. THREE library GEMMs for the weight/input gradients:
      dW_h = hprev^T @ dgates, dW_x = x^T @ dgates, dx = dgates @ W_x^T
      (batched over all timesteps -- no per-step dW round trips)  [K5b]
