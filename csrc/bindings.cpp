@@ -54,7 +54,8 @@ int launch_lstm_seq_fwd_persistent(void*, const void*, const void*,
 int launch_lstm_dh_pw_bwd(const void*, long, const void*, long, const void*,
                           long, const void*, long, const float*, long,
                           const float*, long, void*, long, float*, long,
-                          void*, long, int, int, int, int, hipStream_t);
+                          void*, long, float*, int, int, int, int,
+                          hipStream_t);
 int launch_lstm_dh_pw_bwd_ov(const void*, long, const void*, long,
                              const void*, long, const void*, long,
                              const float*, long, const float*, long, void*,
@@ -305,6 +306,16 @@ std::vector<Tensor> lstm_seq_bwd(c10::optional<Tensor> dh_seq_opt,
         && (g_fused_bwd > 0
             || (g_fused_bwd < 0 && H <= 1536
                 && (B / 256) * (H / 256) >= 192));
+    // bias-grad partials: (B/256, 4H) fp32, accumulated by the fused
+    // kernels at dgates-write time (saves the separate 17 GB/layer colsum
+    // read).  Only the 512-thread kernel implements it; the ov A/B arm
+    // (mode 2) skips it and the caller falls back to colsum.
+    Tensor db_partial;
+    const bool want_db = fused && g_fused_bwd != 2 && T > 1;
+    if (want_db)
+        db_partial = at::zeros({B / 256, H4},
+                               acts.options().dtype(at::kFloat));
+    float* dbp = want_db ? (float*)db_partial.data_ptr() : nullptr;
     Tensor w_h_t;
     if (!fused) {
         // split fallback: W_h^T as a CONTIGUOUS copy (one 8 MB transpose
@@ -358,7 +369,7 @@ std::vector<Tensor> lstm_seq_bwd(c10::optional<Tensor> dh_seq_opt,
                 cs_p + t * B * H, H, cprev_p, H,
                 dg_p + 2 * t * B * H4, H4,
                 (float*)dc.data_ptr(), H,
-                nullptr, 0, (int)B, (int)H, /*raw=*/0, bswz, stream);
+                nullptr, 0, dbp, (int)B, (int)H, /*raw=*/0, bswz, stream);
             TORCH_CHECK(rc == 0, "fused bwd launch failed");
             continue;
         }
@@ -391,11 +402,16 @@ std::vector<Tensor> lstm_seq_bwd(c10::optional<Tensor> dh_seq_opt,
             dg_p + 2 * t0 * B * H4, H4, w_hc.data_ptr(), H4,
             nullptr, 0, nullptr, 0, nullptr, 0, nullptr, 0,
             nullptr, 0, nullptr, 0,
-            dh_buf.data_ptr(), H, (int)B, (int)H, /*raw=*/1, /*swz=*/0,
-            stream);
+            dh_buf.data_ptr(), H, nullptr, (int)B, (int)H, /*raw=*/1,
+            /*swz=*/0, stream);
         TORCH_CHECK(rc == 0, "fused bwd raw launch failed");
     }
-    return {dgates, dh_buf, dc};  // (dgates, dh0, dc0)
+    if (!want_db)
+        db_partial = at::empty({0, H4}, acts.options().dtype(at::kFloat));
+    // (dgates, dh0, dc0, bias-grad partials [0 rows when unavailable --
+    //  covers all timesteps EXCEPT the first processed step, which ran
+    //  through k_lstm_pw_bwd; the python side adds that step's colsum])
+    return {dgates, dh_buf, dc, db_partial};
 }
 
 // ---------------------------------------------------------------------

@@ -1563,6 +1563,10 @@ __global__ __launch_bounds__(512, 2) void k_lstm_dh_pw_bwd(
     __hip_bfloat16* __restrict__ dgates_out, long dgo_stride,     // (B,4H)
     float* __restrict__ dc, long dc_stride,                       // in-place
     __hip_bfloat16* __restrict__ dh_out, long dho_stride,         // RAW only
+    float* __restrict__ db_partial,   // nullable: (B/256, 4H) fp32, this
+                                      // block += its column sums (bias grad
+                                      // accumulated at write -- saves the
+                                      // separate 17 GB/layer colsum pass)
     int B, int H)
 {
     constexpr int BK = 64;
@@ -1721,6 +1725,12 @@ __global__ __launch_bounds__(512, 2) void k_lstm_dh_pw_bwd(
         return;
     }
     float* const img = (float*)smem;        // [64 rows][256 dh-cols] fp32
+    // bias-grad accumulation: this thread's fixed 8 dh-cols x 4 gates
+    float acc_db[4][8];
+#pragma unroll
+    for (int g = 0; g < 4; ++g)
+#pragma unroll
+        for (int jj = 0; jj < 8; ++jj) acc_db[g][jj] = 0.f;
 #pragma unroll
     for (int rd = 0; rd < 4; ++rd) {
         if (rd) __syncthreads();
@@ -1798,6 +1808,42 @@ __global__ __launch_bounds__(512, 2) void k_lstm_dh_pw_bwd(
             *(bf16x8*)(dgp + (long)3 * H) = out_o;
             *(f32x4*)(dc + grow * dc_stride + hcol) = dc0;
             *(f32x4*)(dc + grow * dc_stride + hcol + 4) = dc1;
+            if (db_partial) {
+#pragma unroll
+                for (int jj = 0; jj < 8; ++jj) {
+                    acc_db[0][jj] += (float)out_i[jj];
+                    acc_db[1][jj] += (float)out_f[jj];
+                    acc_db[2][jj] += (float)out_g[jj];
+                    acc_db[3][jj] += (float)out_o[jj];
+                }
+            }
+        }
+    }
+    if (db_partial) {
+        // deterministic block reduce: 16 ordered contributor groups into
+        // a 4-KiB LDS psum (placed after the 64-KiB img), then one
+        // read-modify-write of this block's own partials row (launches
+        // are stream-serialized across timesteps, so += is race-free and
+        // run-to-run reproducible).
+        float* const psum = (float*)(smem + 65536);   // [4 gates][256 cols]
+        for (int i = tid; i < 1024; i += 512) psum[i] = 0.f;
+        __syncthreads();
+        for (int c = 0; c < 16; ++c) {
+            if ((tid >> 5) == c) {
+                const int hc = (tid & 31) * 8;
+#pragma unroll
+                for (int g = 0; g < 4; ++g)
+#pragma unroll
+                    for (int jj = 0; jj < 8; ++jj)
+                        psum[g * 256 + hc + jj] += acc_db[g][jj];
+            }
+            __syncthreads();
+        }
+        for (int i = tid; i < 1024; i += 512) {
+            const int g = i >> 8, col = i & 255;
+            float* p = db_partial + (long)m_blk * (4 * H) + (long)g * H
+                       + n_col0 + col;
+            *p += psum[i];
         }
     }
 }
@@ -1817,6 +1863,10 @@ __global__ __launch_bounds__(256, 2) void k_lstm_dh_pw_bwd_ov(
     __hip_bfloat16* __restrict__ dgates_out, long dgo_stride,     // (B,4H)
     float* __restrict__ dc, long dc_stride,                       // in-place
     __hip_bfloat16* __restrict__ dh_out, long dho_stride,         // RAW only
+    float* __restrict__ db_partial,   // nullable: (B/256, 4H) fp32, this
+                                      // block += its column sums (bias grad
+                                      // accumulated at write -- saves the
+                                      // separate 17 GB/layer colsum pass)
     int B, int H)
 {
     constexpr int BK = 32;
@@ -1936,6 +1986,11 @@ __global__ __launch_bounds__(256, 2) void k_lstm_dh_pw_bwd_ov(
         return;
     }
     float* const img = (float*)smem;        // [32 rows][256 dh-cols] fp32
+    float acc_db[4][8];
+#pragma unroll
+    for (int g = 0; g < 4; ++g)
+#pragma unroll
+        for (int jj = 0; jj < 8; ++jj) acc_db[g][jj] = 0.f;
 #pragma unroll
     for (int rd = 0; rd < 4; ++rd) {
         if (rd) __syncthreads();
@@ -2012,6 +2067,42 @@ __global__ __launch_bounds__(256, 2) void k_lstm_dh_pw_bwd_ov(
             *(bf16x8*)(dgp + (long)3 * H) = out_o;
             *(f32x4*)(dc + grow * dc_stride + hcol) = dc0;
             *(f32x4*)(dc + grow * dc_stride + hcol + 4) = dc1;
+            if (db_partial) {
+#pragma unroll
+                for (int jj = 0; jj < 8; ++jj) {
+                    acc_db[0][jj] += (float)out_i[jj];
+                    acc_db[1][jj] += (float)out_f[jj];
+                    acc_db[2][jj] += (float)out_g[jj];
+                    acc_db[3][jj] += (float)out_o[jj];
+                }
+            }
+        }
+    }
+    if (db_partial) {
+        // deterministic block reduce: 16 ordered contributor groups into
+        // a 4-KiB LDS psum (placed after the 64-KiB img), then one
+        // read-modify-write of this block's own partials row (launches
+        // are stream-serialized across timesteps, so += is race-free and
+        // run-to-run reproducible).
+        float* const psum = (float*)(smem + 65536);   // [4 gates][256 cols]
+        for (int i = tid; i < 1024; i += 512) psum[i] = 0.f;
+        __syncthreads();
+        for (int c = 0; c < 16; ++c) {
+            if ((tid >> 5) == c) {
+                const int hc = (tid & 31) * 8;
+#pragma unroll
+                for (int g = 0; g < 4; ++g)
+#pragma unroll
+                    for (int jj = 0; jj < 8; ++jj)
+                        psum[g * 256 + hc + jj] += acc_db[g][jj];
+            }
+            __syncthreads();
+        }
+        for (int i = tid; i < 1024; i += 512) {
+            const int g = i >> 8, col = i & 255;
+            float* p = db_partial + (long)m_blk * (4 * H) + (long)g * H
+                       + n_col0 + col;
+            *p += psum[i];
         }
     }
 }
@@ -2039,7 +2130,8 @@ extern "C" int launch_lstm_dh_pw_bwd_ov(
             (const __hip_bfloat16*)dh_up, up_stride,                          \
             (const __hip_bfloat16*)acts, act_stride, c_cur, cc_stride,        \
             c_prev, cp_stride, (__hip_bfloat16*)dgates_out, dgo_stride,       \
-            dc, dc_stride, (__hip_bfloat16*)dh_out, dho_stride, B, H);        \
+            dc, dc_stride, (__hip_bfloat16*)dh_out, dho_stride,               \
+            (float*)nullptr, B, H);                                           \
     } while (0)
     if (raw) {
         if (xcd_swz) DHOVLAUNCH(true, true); else DHOVLAUNCH(true, false);
@@ -2055,8 +2147,8 @@ extern "C" int launch_lstm_dh_pw_bwd(
     const void* dh_up, long up_stride, const void* acts, long act_stride,
     const float* c_cur, long cc_stride, const float* c_prev, long cp_stride,
     void* dgates_out, long dgo_stride, float* dc, long dc_stride,
-    void* dh_out, long dho_stride, int B, int H, int raw, int xcd_swz,
-    hipStream_t stream)
+    void* dh_out, long dho_stride, float* db_partial, int B, int H, int raw,
+    int xcd_swz, hipStream_t stream)
 {
     if (B % 256 != 0 || H % 64 != 0) return -1;
     const size_t smem = 131072;
@@ -2074,7 +2166,8 @@ extern "C" int launch_lstm_dh_pw_bwd(
             (const __hip_bfloat16*)dh_up, up_stride,                          \
             (const __hip_bfloat16*)acts, act_stride, c_cur, cc_stride,        \
             c_prev, cp_stride, (__hip_bfloat16*)dgates_out, dgo_stride,       \
-            dc, dc_stride, (__hip_bfloat16*)dh_out, dho_stride, B, H);        \
+            dc, dc_stride, (__hip_bfloat16*)dh_out, dho_stride, db_partial,   \
+            B, H);                                                            \
     } while (0)
     if (raw) {
         if (xcd_swz) DHLAUNCH(true, true); else DHLAUNCH(true, false);

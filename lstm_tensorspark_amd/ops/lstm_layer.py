@@ -103,10 +103,17 @@ def _seq_fwd(pre, w_h, h0, c0, reverse, use_hip):
 
 
 def _seq_bwd(d_hseq, d_hT, d_cT, w_h, acts, c_seq, c0, reverse, use_hip):
+    """Returns (dgates, dh0, dc0, db_partial) -- db_partial is the fused
+    backward's write-time bias-grad partials ((B/256, 4H) fp32 covering
+    every step except the first processed one), or a 0-row tensor when
+    the fused path did not run (CPU reference / split shapes)."""
     if use_hip:
         ext = extension.get_extension()
-        return ext.lstm_seq_bwd(d_hseq, d_hT, d_cT, w_h, acts, c_seq, c0, reverse)
-    return functional.ref_seq_bwd(d_hseq, d_hT, d_cT, w_h, acts, c_seq, c0, reverse)
+        return ext.lstm_seq_bwd(d_hseq, d_hT, d_cT, w_h, acts, c_seq, c0,
+                                reverse)
+    out = functional.ref_seq_bwd(d_hseq, d_hT, d_cT, w_h, acts, c_seq, c0,
+                                 reverse)
+    return (*out, torch.empty(0, w_h.shape[1]))
 
 
 def _fp8_addmm(bias, a, w):
@@ -154,7 +161,7 @@ class LSTMLayerFn(torch.autograd.Function):
         H = w_h.shape[0]
         # set_materialize_grads(False): unused outputs arrive as None and the
         # backward kernels treat null pointers as zero.
-        dgates, dh0, dc0 = _seq_bwd(
+        dgates, dh0, dc0, db_part = _seq_bwd(
             d_hseq.contiguous() if d_hseq is not None else None,
             d_hT.contiguous() if d_hT is not None else None,
             d_cT.contiguous() if d_cT is not None else None,
@@ -243,7 +250,17 @@ class LSTMLayerFn(torch.autograd.Function):
             if ng[3]:
                 dw_x = (x_flat.t() @ dg_flat).to(w_x.dtype)
             if ng[5]:
-                if dg_flat.is_cuda and dg_flat.dtype == torch.bfloat16 \
+                if db_part.numel() > 0:
+                    # bias grad accumulated at dgates-WRITE time by the
+                    # fused backward kernel (r2): partials cover every
+                    # step except the first processed one -- add that
+                    # step's column sum (one (B,4H) slice, ~0.1% of the
+                    # old full colsum's 17 GB re-read)
+                    T_first = (dgates.shape[0] - 1) if not ctx.reverse else 0
+                    first = extension.get_extension().colsum_bf16(
+                        dgates.select(0, T_first))
+                    db = (db_part.sum(dim=0) + first).to(w_x.dtype)
+                elif dg_flat.is_cuda and dg_flat.dtype == torch.bfloat16 \
                         and dg_flat.shape[1] % 8 == 0:
                     # single-pass fused column sum (K8): torch's generic
                     # two-stage reduce measured ~3x roofline on this shape

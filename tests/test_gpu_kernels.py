@@ -135,7 +135,7 @@ def test_seq_bwd_vs_ref(ext, reverse):
     d_hT = (torch.randn(B, H, generator=g) * 0.1).to("cuda").bfloat16()
     d_cT = (torch.randn(B, H, generator=g) * 0.1).to("cuda")
 
-    dg, dh0, dc0 = ext.lstm_seq_bwd(d_hseq, d_hT, d_cT, w_h, a_r, c_r, c0, reverse)
+    dg, dh0, dc0 = ext.lstm_seq_bwd(d_hseq, d_hT, d_cT, w_h, a_r, c_r, c0, reverse)[:3]
     dg_r, dh0_r, dc0_r = functional.ref_seq_bwd(d_hseq, d_hT, d_cT, w_h, a_r,
                                                 c_r, c0, reverse)
     assert (dg.float() - dg_r.float()).abs().max().item() < 0.05
@@ -158,10 +158,10 @@ def test_seq_bwd_fused_vs_ref(ext, reverse, B, T, H):
     try:
         ext.set_fused_bwd(1)
         dg, dh0, dc0 = ext.lstm_seq_bwd(d_hseq, None, None, w_h, a_r, c_r,
-                                        c0, reverse)
+                                        c0, reverse)[:3]
         ext.set_fused_bwd(0)
         dg_s, dh0_s, dc0_s = ext.lstm_seq_bwd(d_hseq, None, None, w_h, a_r,
-                                              c_r, c0, reverse)
+                                              c_r, c0, reverse)[:3]
     finally:
         ext.set_fused_bwd(prev)
     dg_r, dh0_r, dc0_r = functional.ref_seq_bwd(d_hseq, None, None, w_h, a_r,
@@ -172,6 +172,31 @@ def test_seq_bwd_fused_vs_ref(ext, reverse, B, T, H):
     # fused vs split agree to bf16 rounding (fused keeps the dh carry fp32)
     assert (dg.float() - dg_s.float()).abs().max().item() < 0.05
     assert (dh0.float() - dh0_s.float()).abs().max().item() < 0.05
+
+
+@pytest.mark.parametrize("reverse", [False, True])
+def test_fused_bwd_write_time_bias_grad(ext, reverse):
+    """The r2 write-time bias-grad accumulation (db_partial) must equal
+    the full colsum of dgates to fp32-ordering tolerance."""
+    B, T, H = 256, 5, 256
+    pre, w_h, h0, c0 = seq_inputs(B, T, H, seed=9)
+    from lstm_tensorspark_amd.ops import functional
+    h_r, c_r, a_r, _, _ = functional.ref_seq_fwd(pre, w_h, h0, c0, reverse)
+    g = torch.Generator(device="cpu").manual_seed(10)
+    d_hseq = (torch.randn(T, B, H, generator=g) * 0.1).to("cuda").bfloat16()
+    prev = ext.get_fused_bwd()
+    try:
+        ext.set_fused_bwd(1)
+        dg, dh0, dc0, db_p = ext.lstm_seq_bwd(d_hseq, None, None, w_h, a_r,
+                                              c_r, c0, reverse)
+    finally:
+        ext.set_fused_bwd(prev)
+    assert db_p.numel() > 0, "partials missing on an eligible fused shape"
+    t_first = (T - 1) if not reverse else 0
+    db = db_p.sum(dim=0) + ext.colsum_bf16(dg.select(0, t_first))
+    ref = dg.float().sum(dim=(0, 1))
+    scale = max(1.0, ref.abs().max().item())
+    assert (db - ref).abs().max().item() < 1e-2 * scale
 
 
 def test_seq_bwd_fused_none_dh_up(ext):
@@ -204,7 +229,7 @@ def test_seq_bwd_none_grads(ext):
     pre, w_h, h0, c0 = seq_inputs(B, T, H, seed=4)
     h_r, c_r, a_r, _, _ = functional.ref_seq_fwd(pre, w_h, h0, c0, False)
     d_hT = (torch.randn(B, H) * 0.1).to("cuda").bfloat16()
-    dg, dh0, dc0 = ext.lstm_seq_bwd(None, d_hT, None, w_h, a_r, c_r, c0, False)
+    dg, dh0, dc0 = ext.lstm_seq_bwd(None, d_hT, None, w_h, a_r, c_r, c0, False)[:3]
     dg_r, dh0_r, dc0_r = functional.ref_seq_bwd(None, d_hT, None, w_h, a_r,
                                                 c_r, c0, False)
     assert (dg.float() - dg_r.float()).abs().max().item() < 0.05
