@@ -2079,15 +2079,14 @@ __global__ __launch_bounds__(256, 2) void k_lstm_dh_pw_bwd_ov(
         }
     }
     if (db_partial) {
-        // deterministic block reduce: 16 ordered contributor groups into
-        // a 4-KiB LDS psum (placed after the 64-KiB img), then one
-        // read-modify-write of this block's own partials row (launches
-        // are stream-serialized across timesteps, so += is race-free and
-        // run-to-run reproducible).
+        // 256-thread variant of the 512-thread kernel's deterministic
+        // reduce (8 contributor groups); NOTE the partials row index is
+        // in B/128 units here -- a caller enabling this must allocate
+        // (B/128, 4H) partials (the binding currently passes null).
         float* const psum = (float*)(smem + 65536);   // [4 gates][256 cols]
-        for (int i = tid; i < 1024; i += 512) psum[i] = 0.f;
+        for (int i = tid; i < 1024; i += 256) psum[i] = 0.f;
         __syncthreads();
-        for (int c = 0; c < 16; ++c) {
+        for (int c = 0; c < 8; ++c) {
             if ((tid >> 5) == c) {
                 const int hc = (tid & 31) * 8;
 #pragma unroll
@@ -2098,7 +2097,7 @@ __global__ __launch_bounds__(256, 2) void k_lstm_dh_pw_bwd_ov(
             }
             __syncthreads();
         }
-        for (int i = tid; i < 1024; i += 512) {
+        for (int i = tid; i < 1024; i += 256) {
             const int g = i >> 8, col = i & 255;
             float* p = db_partial + (long)m_blk * (4 * H) + (long)g * H
                        + n_col0 + col;
