@@ -118,14 +118,23 @@ def main():
     ext.set_persistent_fwd(True)
     ms = ms
 
-    # sequence backward
+    # sequence backward: split vs fused (512-thr) vs fused-ov A/B
     h_all, c_seq, acts, hT, cT = ext.lstm_seq_fwd(pre, w_h, h0, c0, False)
     d_hseq = torch.randn(T, B, H, generator=g, device=dev).bfloat16()
-    ms = time_op(lambda: ext.lstm_seq_bwd(d_hseq, None, None, w_h, acts,
-                                          c_seq, c0, False),
-                 max(args.iters // 10, 3))
-    report("seq_bwd", ms, 2 * rec_flops * T, B=B, H=H, T=T,
-           us_per_step=round(1e3 * ms / T, 2))
+    bwd_default = ext.get_fused_bwd()
+    for mode, tag in ((0, "split"), (-1, "auto"), (1, "fused512"),
+                      (3, "fused512-swz"), (2, "fused-ov")):
+        ext.set_fused_bwd(mode)
+        try:
+            ms = time_op(lambda: ext.lstm_seq_bwd(d_hseq, None, None, w_h,
+                                                  acts, c_seq, c0, False),
+                         max(args.iters // 10, 3))
+        except Exception as e:
+            report(f"seq_bwd {tag}", 0.0, note=f"skipped: {e}")
+            continue
+        report(f"seq_bwd {tag}", ms, 2 * rec_flops * T, B=B, H=H, T=T,
+               us_per_step=round(1e3 * ms / T, 2))
+    ext.set_fused_bwd(bwd_default)
 
     # components: library GEMMs
     dg1 = torch.randn(B, 4 * H, generator=g, device=dev).bfloat16()
