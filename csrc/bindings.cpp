@@ -38,7 +38,7 @@ void launch_quant_fp8_dyn(const void*, void*, long, const float*, hipStream_t);
 int launch_gemv_bf16(const void*, const void*, const float*, void*, float*,
                      int, long, int, hipStream_t);
 int launch_colsum_bf16(const void*, float*, float*, long, long, hipStream_t);
-int colsum_gy(long);
+int colsum_gy(long, long);
 void launch_quant_fp8_dyn_T(const void*, void*, long, long, const float*,
                             hipStream_t);
 int launch_quant_fp8_pack(const void*, void*, void*, float*, long, long,
@@ -469,7 +469,7 @@ Tensor colsum_bf16(Tensor t) {
     TORCH_CHECK(t.dim() == 2, "colsum wants (R,C)");
     t = t.contiguous();
     const long R = t.size(0), C = t.size(1);
-    auto partials = at::empty({(long)colsum_gy(R), C},
+    auto partials = at::empty({(long)colsum_gy(R, C), C},
                               t.options().dtype(at::kFloat));
     auto out = at::empty({C}, t.options().dtype(at::kFloat));
     int rc = launch_colsum_bf16(t.data_ptr(), (float*)partials.data_ptr(),

@@ -2493,8 +2493,17 @@ __global__ __launch_bounds__(256) void k_colsum_final(
     out[c] = s;
 }
 
-// partials must hold colsum_gy(R) * C floats (allocated by the caller).
-extern "C" int colsum_gy(long R) { return (int)((R + 4095) / 4096); }
+// Row-chunking: pick gy so grid.x * gy fills the chip (the fixed
+// 4096-row chunk of r1 left a (16384, 4096) input at 8 blocks -> 1.8 ms
+// for a 134 MB pass).  partials must hold colsum_gy(R, C) * C floats.
+extern "C" int colsum_gy(long R, long C) {
+    const long gx = (C / 8 + 255) / 256;
+    long gy = 1024 / (gx < 1 ? 1 : gx);
+    const long gy_max = (R + 63) / 64;   // >= 64 rows per chunk
+    if (gy > gy_max) gy = gy_max;
+    if (gy < 1) gy = 1;
+    return (int)gy;
+}
 
 extern "C" void launch_colsum_final(const float* partials, float* out, long C,
                                     int gy, hipStream_t stream)
@@ -2508,8 +2517,8 @@ extern "C" int launch_colsum_bf16(const void* in, float* partials, float* out,
                                   long R, long C, hipStream_t stream)
 {
     if (C % 8 != 0) return -1;
-    const int chunk = 4096;
-    const int gy = colsum_gy(R);
+    const int gy = colsum_gy(R, C);
+    const int chunk = (int)((R + gy - 1) / gy);
     dim3 block(256);
     dim3 grid((unsigned)((C / 8 + 255) / 256), (unsigned)gy);
     hipLaunchKernelGGL(k_colsum_bf16, grid, block, 0, stream,
