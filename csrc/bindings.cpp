@@ -37,6 +37,7 @@ void launch_quant_fp8_fixed(const void*, void*, long, float, hipStream_t);
 void launch_quant_fp8_dyn(const void*, void*, long, const float*, hipStream_t);
 int launch_gemv_bf16(const void*, const void*, const float*, void*, float*,
                      int, long, int, hipStream_t);
+void set_gemv_nt(int);
 int launch_colsum_bf16(const void*, float*, float*, long, long, hipStream_t);
 int colsum_gy(long, long);
 void launch_quant_fp8_dyn_T(const void*, void*, long, long, const float*,
@@ -650,6 +651,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("decode_gemv", &decode_gemv, "small-B serving GEMV",
           py::arg("A_t"), py::arg("x"), py::arg("bias") = py::none());
     m.def("decode_gemv_ok", &decode_gemv_ok);
+    m.def("set_gemv_nt", &set_gemv_nt,
+          "non-temporal weight loads in the decode GEMV (1 default)");
     m.def("quant_fp8", &quant_fp8, "bf16 -> e4m3, device-scalar multiplier");
     m.def("quant_fp8_T", &quant_fp8_T, "bf16 (R,C) -> e4m3 (C,R) transposed");
     m.def("quant_fp8_pack", &quant_fp8_pack,

@@ -2537,7 +2537,10 @@ extern "C" int launch_colsum_bf16(const void* in, float* partials, float* out,
 // (x is LDS-resident), cross-lane shfl reduce.  A is the PRE-TRANSPOSED
 // weight (serving layout, built once per model by the decode cache).
 // ---------------------------------------------------------------------------
-template <int BMAX>
+// NT: non-temporal weight loads (guide "nt-weights": a decode stream is
+// read ONCE by one CU per token -- nt loads landed 18% faster there; the
+// prefill-reuse caveat does not apply to this kernel's weight rows).
+template <int BMAX, bool NT = false>
 __global__ __launch_bounds__(256) void k_gemv_bf16(
     const __hip_bfloat16* __restrict__ A,    // (N, K) row-major
     const __hip_bfloat16* __restrict__ x,    // (B, K) row-major
@@ -2568,8 +2571,14 @@ __global__ __launch_bounds__(256) void k_gemv_bf16(
         // a single outstanding 16B load per lane leaves the row stream
         // latency-exposed)
         for (int k0 = lane * 16; k0 < K; k0 += 64 * 16) {
-            const bf16x8 av0 = *(const bf16x8*)(ar + k0);
-            const bf16x8 av1 = *(const bf16x8*)(ar + k0 + 8);
+            bf16x8 av0, av1;
+            if (NT) {
+                av0 = __builtin_nontemporal_load((const bf16x8*)(ar + k0));
+                av1 = __builtin_nontemporal_load((const bf16x8*)(ar + k0 + 8));
+            } else {
+                av0 = *(const bf16x8*)(ar + k0);
+                av1 = *(const bf16x8*)(ar + k0 + 8);
+            }
 #pragma unroll
             for (int b = 0; b < BMAX; ++b) {
                 if (b < B) {
@@ -2600,6 +2609,9 @@ __global__ __launch_bounds__(256) void k_gemv_bf16(
     }
 }
 
+static int g_gemv_nt = 1;  // nt weight loads on by default (A/B'd r2)
+extern "C" void set_gemv_nt(int v) { g_gemv_nt = v; }
+
 // Returns 0 on launch, -1 when the shape is outside the fast-path
 // envelope (the kernel template caps B at 16, K must satisfy the
 // lane-stride contract, and x must fit LDS) -- the caller must NOT read
@@ -2617,7 +2629,19 @@ extern "C" int launch_gemv_bf16(const void* A, const void* x,
     if (B < 1 || B > 16 || K % 16 != 0 || smem > 160 * 1024 || N < 1)
         return -1;
     dim3 block(256), grid((unsigned)((N + 3) / 4));
-#define GEMV_LAUNCH(BM)                                                           hipLaunchKernelGGL(k_gemv_bf16<BM>, grid, block, smem, stream,                    (const __hip_bfloat16*)A, (const __hip_bfloat16*)x, bias,                     (__hip_bfloat16*)out, out_f32, B, N, K)
+#define GEMV_LAUNCH(BM)                                                       \
+    do {                                                                      \
+        if (g_gemv_nt)                                                        \
+            hipLaunchKernelGGL((k_gemv_bf16<BM, true>), grid, block, smem,    \
+                               stream, (const __hip_bfloat16*)A,              \
+                               (const __hip_bfloat16*)x, bias,                \
+                               (__hip_bfloat16*)out, out_f32, B, N, K);       \
+        else                                                                  \
+            hipLaunchKernelGGL((k_gemv_bf16<BM, false>), grid, block, smem,   \
+                               stream, (const __hip_bfloat16*)A,              \
+                               (const __hip_bfloat16*)x, bias,                \
+                               (__hip_bfloat16*)out, out_f32, B, N, K);       \
+    } while (0)
     if (B <= 1) GEMV_LAUNCH(1);
     else if (B <= 4) GEMV_LAUNCH(4);
     else if (B <= 8) GEMV_LAUNCH(8);
