@@ -115,6 +115,7 @@ class LSTMModel(nn.Module):
         H = cfg.hidden_size
         D = cfg.num_directions
         self._params: dict[str, nn.Parameter] = {}
+        self._zero_state: dict = {}   # (B, device, dtype) -> (h0, c0)
 
         if cfg.head == "lm":
             assert cfg.vocab_size > 0
@@ -177,8 +178,19 @@ class LSTMModel(nn.Module):
         w_h = self.weight(_pname(layer, rev, "W_h"))
         b = self.weight(_pname(layer, rev, "b"))
         if state is None:
-            h0 = torch.zeros(B, H, dtype=x.dtype, device=x.device)
-            c0 = torch.zeros(B, H, dtype=torch.float32, device=x.device)
+            # cached zero states: the per-call torch.zeros pair cost
+            # ~2 ms/step at the headline shape (200 MB/step of zero-fill
+            # for 2 layers); these are read-only inputs everywhere (the
+            # kernels never write h0/c0 and autograd saves, not mutates),
+            # so one cached pair per (B, device, dtype) is safe -- and
+            # address-stable, which hipGraph capture likes.
+            key = (B, x.device, x.dtype)
+            zs = self._zero_state.get(key)
+            if zs is None:
+                zs = (torch.zeros(B, H, dtype=x.dtype, device=x.device),
+                      torch.zeros(B, H, dtype=torch.float32, device=x.device))
+                self._zero_state[key] = zs
+            h0, c0 = zs
         else:
             h0, c0 = state
         return lstm_layer(x, h0, c0, w_x, w_h, b, reverse=rev)
