@@ -187,3 +187,25 @@ def test_overlap_probe_tool_degenerate(tmp_path):
     r = main(["--batch", "2", "--steps", "3", "--out", out])
     assert r["per_step"]["buckets"], "no bucket-ready timestamps recorded"
     assert r["per_step"]["bwd_ms"] > 0
+
+
+def test_zero_state_cache_not_mutated():
+    """The r2 cached zero (h0, c0) must stay zero across steps (they are
+    shared read-only inputs; a kernel or autograd write would corrupt
+    every later step)."""
+    cfg = Config(
+        model=ModelConfig(input_size=8, hidden_size=8, num_layers=2),
+        train=TrainConfig(batch_size=3, seq_len=6, lr=0.1, steps=3,
+                          log_interval=0),
+        data=DataConfig(task="random"),
+    )
+    t = Trainer(cfg)
+    for i in range(3):
+        t.train_step(t.data.batch(i))
+    for (B, dev, dt), (h0, c0) in t.model._zero_state.items():
+        assert torch.all(h0 == 0) and torch.all(c0 == 0)
+    # cache keyed per batch size: a different B makes a new entry
+    cfg2 = cfg
+    x = torch.randn(6, 5, 8)
+    t.model(x.to(t.model.weight("layer0/W_x").dtype))
+    assert any(k[0] == 5 for k in t.model._zero_state)
