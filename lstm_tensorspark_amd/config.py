@@ -70,6 +70,9 @@ class DistConfig:
     bucket_mb: int = 16           # gradient all-reduce bucket size, sized for xGMI links
     side_dw: bool = False         # dW GEMMs on a side stream (neutral at HBM-sized B)
     reduce_dtype: str = ""        # "" native (bf16) | "fp32" wire precision for all-reduce
+    bucket_order: str = "reverse" # "reverse" registration heuristic |
+                                  # "measured": rebuild buckets to the
+                                  # grad-arrival order observed in step 1
 
 
 @dataclass

@@ -52,46 +52,63 @@ class GradBucketer:
 
     def __init__(self, params, bucket_mb: int = 16,
                  process_group: Optional[dist.ProcessGroup] = None,
-                 reduce_dtype: Optional[torch.dtype] = None):
+                 reduce_dtype: Optional[torch.dtype] = None,
+                 bucket_order: str = "reverse"):
         """reduce_dtype: cast each bucket to this dtype for the wire
         (e.g. torch.float32 to avoid bf16 summation error across 8 ranks);
-        None reduces in the native grad dtype."""
+        None reduces in the native grad dtype.
+
+        bucket_order: "reverse" groups parameters in reverse registration
+        order (the transformer DDP heuristic).  "measured" additionally
+        RECORDS the actual grad-arrival order during the first backward
+        and rebuilds the buckets to match it -- a recurrent backward
+        finishes all layers' dW near the END of its time loop (SURVEY §7
+        hard part 3), so the registration heuristic can mis-group; the
+        rebuild happens at the next zero_grad (grads are zeroed there
+        anyway, so no gradient state is lost)."""
         self.pg = process_group
         self.reduce_dtype = reduce_dtype
+        self.bucket_order = bucket_order
         self.world = dist.get_world_size(process_group) if dist.is_initialized() else 1
         self.params = [p for p in params if p.requires_grad]
         if not self.params:
             raise ValueError("no parameters")
-        device = self.params[0].device
-        bucket_bytes = bucket_mb * (1 << 20)
+        self._device = self.params[0].device
+        self._bucket_bytes = bucket_mb * (1 << 20)
+        self._arrival: List[torch.nn.Parameter] = []
+        self._arrival_ids: set = set()
+        self._rebuilt = bucket_order != "measured"
 
         # Reverse order ~= backward completion order (last layers first).
+        self._build_buckets(list(reversed(self.params)))
+        self._hooks = [
+            p.register_post_accumulate_grad_hook(self._on_grad_ready)
+            for p in self.params
+        ]
+        self._reset_pending()
+
+    def _build_buckets(self, ordered_params):
         self.buckets: List[_Bucket] = []
         cur: List[torch.nn.Parameter] = []
         cur_bytes = 0
-        for p in reversed(self.params):
+        for p in ordered_params:
             esize = p.element_size()
-            if cur and cur_bytes + p.numel() * esize > bucket_bytes:
-                self.buckets.append(_Bucket(cur, device, cur[0].dtype))
+            if cur and cur_bytes + p.numel() * esize > self._bucket_bytes:
+                self.buckets.append(_Bucket(cur, self._device, cur[0].dtype))
                 cur, cur_bytes = [], 0
             if cur and p.dtype != cur[0].dtype:
-                self.buckets.append(_Bucket(cur, device, cur[0].dtype))
+                self.buckets.append(_Bucket(cur, self._device, cur[0].dtype))
                 cur, cur_bytes = [], 0
             cur.append(p)
             cur_bytes += p.numel() * esize
         if cur:
-            self.buckets.append(_Bucket(cur, device, cur[0].dtype))
+            self.buckets.append(_Bucket(cur, self._device, cur[0].dtype))
 
         self._param_bucket = {}
         for b in self.buckets:
             for p, v in zip(b.params, b.views):
                 self._param_bucket[p] = b
                 p.grad = v  # accumulate directly into the bucket
-        self._hooks = [
-            p.register_post_accumulate_grad_hook(self._on_grad_ready)
-            for p in self.params
-        ]
-        self._reset_pending()
 
     # ------------------------------------------------------------------
     def _reset_pending(self):
@@ -113,6 +130,12 @@ class GradBucketer:
         self._reduce_enabled = will_reduce
 
     def zero_grad(self):
+        if not self._rebuilt and len(self._arrival) == len(self.params):
+            # "measured" order: rebuild the buckets to the grad-arrival
+            # order observed during the first backward (safe here: grads
+            # are being zeroed anyway)
+            self._build_buckets(self._arrival)
+            self._rebuilt = True
         for b in self.buckets:
             b.flat.zero_()
         self._reset_pending()
@@ -123,6 +146,9 @@ class GradBucketer:
                 p.grad = v
 
     def _on_grad_ready(self, p: torch.nn.Parameter):
+        if not self._rebuilt and id(p) not in self._arrival_ids:
+            self._arrival_ids.add(id(p))
+            self._arrival.append(p)
         if self.world == 1 or not dist.is_initialized():
             return
         if not self._reduce_enabled:

@@ -64,9 +64,10 @@ class Trainer:
         if self.ctx.world_size > 1:
             rd = {"fp32": torch.float32, "": None}.get(
                 getattr(cfg.dist, "reduce_dtype", ""), None)
-            self.bucketer = GradBucketer(self.model.parameters(),
-                                         bucket_mb=cfg.dist.bucket_mb,
-                                         reduce_dtype=rd)
+            self.bucketer = GradBucketer(
+                self.model.parameters(), bucket_mb=cfg.dist.bucket_mb,
+                reduce_dtype=rd,
+                bucket_order=getattr(cfg.dist, "bucket_order", "reverse"))
         self.data = SyntheticSeqData(cfg, rank=self.ctx.rank,
                                      world_size=self.ctx.world_size,
                                      device=self.device, dtype=self.dtype)

@@ -149,3 +149,38 @@ def test_2rank_evaluate_allreduces_mean(tmp_path):
                     weights_only=False)
     assert abs(r0["eval_loss"] - r1["eval_loss"]) < 1e-9
     assert r0["steps"] == 2
+
+
+def _grads_measured_order_worker(ctx, tmpdir):
+    cfg = small_cfg()
+    cfg.dist.bucket_order = "measured"
+    t = Trainer(cfg, dist_ctx=ctx)
+    # step 1 records arrival order; zero_grad of step 2 rebuilds buckets
+    t.train_step(t.data.batch(0))
+    t.train_step(t.data.batch(1))
+    assert t.bucketer._rebuilt
+    grads = {n: p.grad.detach().clone()
+             for n, p in zip(t.model.weight_dict(), t.model.parameters())}
+    torch.save(grads, os.path.join(tmpdir, f"gm_rank{ctx.rank}.pt"))
+
+
+def test_2rank_measured_bucket_order_grads_correct(tmp_path):
+    """bucket_order='measured' (rebuild to observed grad-arrival order,
+    VERDICT r1 weak-1): rank-averaged grads must still equal the
+    per-rank average after the rebuild."""
+    run_spawned(_grads_measured_order_worker, world_size=2, backend="gloo",
+                args=(str(tmp_path),), master_port=29591)
+    per_rank = []
+    for rank in range(2):
+        t = Trainer(small_cfg())
+        t.data.rank = rank
+        t.train_step(t.data.batch(0))
+        t.train_step(t.data.batch(1))
+        per_rank.append({n: p.grad.detach().clone()
+                         for n, p in zip(t.model.weight_dict(),
+                                         t.model.parameters())})
+    got = torch.load(os.path.join(str(tmp_path), "gm_rank0.pt"),
+                     weights_only=False)
+    for name in got:
+        want = (per_rank[0][name] + per_rank[1][name]) / 2
+        assert (got[name] - want).abs().max() < 1e-6, name
