@@ -1,0 +1,107 @@
+#!/usr/bin/env python3
+"""Minimal production-style serving front end for the LM decode path.
+
+Wraps `LSTMModel.generate()` (hipGraph-captured, streaming-GEMV decode --
+see README "Serving / decode") behind a FastAPI endpoint:
+
+    python examples/serve.py --ckpt ckpt_final.pt --port 8000
+    curl -s localhost:8000/generate -X POST -H 'content-type: application/json' \
+         -d '{"tokens": [1, 2, 3], "steps": 16}'
+
+With --demo (no checkpoint) it serves a random-init model -- useful for
+latency/throughput testing of the serving stack itself.  Single-GPU,
+single-worker: the decode step is already graph-replayed, so the Python
+layer only pays tensor marshalling per request.
+
+(No `from __future__ import annotations` here: FastAPI resolves the
+request-model type hints at runtime, and postponed annotations of a
+function-local pydantic model are unresolvable -- the endpoint would
+silently degrade to a query parameter.)
+"""
+import argparse
+import os
+import sys
+
+import torch
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+from lstm_tensorspark_amd.config import Config, ModelConfig, TrainConfig, DataConfig
+from lstm_tensorspark_amd.models.lstm import LSTMModel
+
+
+def build_app(model: LSTMModel, device: torch.device):
+    from fastapi import FastAPI, HTTPException
+    from pydantic import BaseModel
+
+    class GenRequest(BaseModel):
+        tokens: list[int]
+        steps: int = 32
+        temperature: float = 0.0
+
+    app = FastAPI(title="lstm_tensorspark_amd serving")
+
+    @app.get("/health")
+    def health():
+        return {"ok": True, "device": str(device),
+                "vocab": model.cfg.vocab_size}
+
+    @app.post("/generate")
+    def generate(req: GenRequest):
+        V = model.cfg.vocab_size
+        if not req.tokens or any(t < 0 or t >= V for t in req.tokens):
+            raise HTTPException(400, f"tokens must be in [0, {V})")
+        if req.steps < 1 or req.steps > 4096:
+            raise HTTPException(400, "steps must be in [1, 4096]")
+        prompt = torch.tensor(req.tokens, dtype=torch.long,
+                              device=device).unsqueeze(1)  # (Tp, B=1)
+        with torch.no_grad():
+            out = model.generate(prompt, steps=req.steps,
+                                 temperature=req.temperature)
+        return {"tokens": out[:, 0].tolist(),
+                "generated": out[len(req.tokens):, 0].tolist()}
+
+    return app
+
+
+def load_model(args) -> tuple[LSTMModel, torch.device]:
+    device = torch.device("cuda:0" if torch.cuda.is_available() else "cpu")
+    if args.demo:
+        cfg = ModelConfig(input_size=args.hidden, hidden_size=args.hidden,
+                          num_layers=2, head="lm", vocab_size=args.vocab)
+        model = LSTMModel(cfg)
+    else:
+        payload = torch.load(args.ckpt, map_location="cpu",
+                             weights_only=False)
+        cfg = Config.from_dict(payload["config"]).model
+        model = LSTMModel(cfg)
+        for name, p in model.weight_dict().items():
+            p.data.copy_(payload["weights"][name])
+    model = model.to(device)
+    if device.type == "cuda":
+        model = model.bfloat16()
+    model.eval()
+    return model, device
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--ckpt", default="", help="checkpoint (.pt) to serve")
+    ap.add_argument("--demo", action="store_true",
+                    help="serve a random-init model (stack testing)")
+    ap.add_argument("--hidden", type=int, default=1024)
+    ap.add_argument("--vocab", type=int, default=50000)
+    ap.add_argument("--host", default="127.0.0.1")
+    ap.add_argument("--port", type=int, default=8000)
+    args = ap.parse_args()
+    if not args.demo and not args.ckpt:
+        print("need --ckpt or --demo", file=sys.stderr)
+        sys.exit(2)
+    model, device = load_model(args)
+    app = build_app(model, device)
+    import uvicorn
+    uvicorn.run(app, host=args.host, port=args.port, log_level="info")
+
+
+if __name__ == "__main__":
+    main()

@@ -209,3 +209,38 @@ def test_zero_state_cache_not_mutated():
     x = torch.randn(6, 5, 8)
     t.model(x.to(t.model.weight("layer0/W_x").dtype))
     assert any(k[0] == 5 for k in t.model._zero_state)
+
+
+def test_serving_example_endpoint():
+    """examples/serve.py: the FastAPI generate endpoint works end-to-end
+    on CPU with a tiny demo model (TestClient, no network)."""
+    import sys as _sys
+    sys_path_added = False
+    import os as _os
+    repo = _os.path.dirname(_os.path.dirname(_os.path.abspath(__file__)))
+    ex = _os.path.join(repo, "examples")
+    if ex not in _sys.path:
+        _sys.path.insert(0, ex)
+        sys_path_added = True
+    try:
+        import importlib
+        serve = importlib.import_module("serve")
+        from starlette.testclient import TestClient
+        from lstm_tensorspark_amd.config import ModelConfig
+        from lstm_tensorspark_amd.models.lstm import LSTMModel
+
+        cfg = ModelConfig(input_size=16, hidden_size=16, num_layers=1,
+                          head="lm", vocab_size=32)
+        m = LSTMModel(cfg).eval()
+        app = serve.build_app(m, torch.device("cpu"))
+        c = TestClient(app)
+        assert c.get("/health").json()["ok"]
+        r = c.post("/generate", json={"tokens": [1, 2, 3], "steps": 4})
+        assert r.status_code == 200, r.text
+        body = r.json()
+        assert len(body["generated"]) == 4
+        assert all(0 <= t < 32 for t in body["tokens"])
+        assert c.post("/generate", json={"tokens": [99], "steps": 4}).status_code == 400
+    finally:
+        if sys_path_added:
+            _sys.path.remove(ex)
