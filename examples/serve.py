@@ -70,6 +70,21 @@ def load_model(args) -> tuple[LSTMModel, torch.device]:
         cfg = ModelConfig(input_size=args.hidden, hidden_size=args.hidden,
                           num_layers=2, head="lm", vocab_size=args.vocab)
         model = LSTMModel(cfg)
+    elif args.ckpt.endswith(".safetensors"):
+        # weights-only deployment artifact: model config travels in the
+        # file's metadata (json) if the exporter put it there, else from CLI
+        import json
+        from safetensors import safe_open
+        from lstm_tensorspark_amd.checkpoint import load_weights_safetensors
+        with safe_open(args.ckpt, framework="pt", device="cpu") as f:
+            meta = dict(f.metadata() or {})
+        if "config" in meta:
+            cfg = ModelConfig(**json.loads(meta["config"]))
+        else:
+            cfg = ModelConfig(input_size=args.hidden, hidden_size=args.hidden,
+                              num_layers=2, head="lm", vocab_size=args.vocab)
+        model = LSTMModel(cfg)
+        load_weights_safetensors(args.ckpt, model)
     else:
         payload = torch.load(args.ckpt, map_location="cpu",
                              weights_only=False)

@@ -55,6 +55,48 @@ def save_checkpoint(path: str, model, optimizer=None, step: int = 0,
             os.unlink(tmp)
 
 
+def export_weights_safetensors(path: str, model, metadata: Optional[dict] = None):
+    """Weights-only export in safetensors format (zero-copy mmap load,
+    no pickle) -- the interchange format for serving/deployment.  The full
+    training checkpoint stays torch.save (optimizer + RNG state are
+    arbitrary python objects); this exports just the flat named-weight
+    dict {layer_i[_rev]/{W_x,W_h,b}, embed/W, head/{W,b}}.
+    """
+    from safetensors.torch import save_file
+    weights = {name: p.detach().cpu().contiguous()
+               for name, p in model.weight_dict().items()}
+    meta = {"format": "lstm_tensorspark_amd", "version": str(FORMAT_VERSION)}
+    for k, v in (metadata or {}).items():
+        meta[str(k)] = str(v)  # safetensors metadata must be str->str
+    d = os.path.dirname(os.path.abspath(path))
+    os.makedirs(d, exist_ok=True)
+    fd, tmp = tempfile.mkstemp(dir=d, suffix=".tmp")
+    os.close(fd)
+    try:
+        save_file(weights, tmp, metadata=meta)
+        os.replace(tmp, path)
+    finally:
+        if os.path.exists(tmp):
+            os.unlink(tmp)
+
+
+def load_weights_safetensors(path: str, model) -> dict:
+    """Load a safetensors weight export into `model` (strict name match).
+    Returns the file's metadata dict."""
+    from safetensors import safe_open
+    wd = model.weight_dict()
+    with safe_open(path, framework="pt", device="cpu") as f:
+        names = set(f.keys())
+        missing = names ^ set(wd)
+        if missing:
+            raise ValueError(
+                f"safetensors/model weight-name mismatch: {sorted(missing)}")
+        with torch.no_grad():
+            for name, p in wd.items():
+                p.copy_(f.get_tensor(name).to(p.device, p.dtype))
+        return dict(f.metadata() or {})
+
+
 def load_checkpoint(path: str, model=None, optimizer=None,
                     restore_rng: bool = True, map_location="cpu") -> dict:
     payload = torch.load(path, map_location=map_location, weights_only=False)

@@ -114,3 +114,31 @@ def test_fault_injection_and_auto_resume(tmp_path):
     tC.fit(6 - tC.step)
     for k, vA in tA.model.weight_dict().items():
         assert torch.equal(vA, tC.model.weight_dict()[k]), k
+
+
+def test_safetensors_export_roundtrip(tmp_path):
+    from lstm_tensorspark_amd.checkpoint import (
+        export_weights_safetensors, load_weights_safetensors)
+    cfg = small_cfg(tmp_path)
+    tA = Trainer(cfg)
+    tA.fit(2)
+    path = os.path.join(str(tmp_path), "w.safetensors")
+    export_weights_safetensors(path, tA.model, metadata={"step": 2})
+    tB = Trainer(small_cfg(tmp_path))
+    meta = load_weights_safetensors(path, tB.model)
+    assert meta["step"] == "2" and meta["format"] == "lstm_tensorspark_amd"
+    for k, vA in tA.model.weight_dict().items():
+        assert torch.equal(vA, tB.model.weight_dict()[k]), k
+    # strict name check: wrong architecture is rejected
+    bad = Trainer(Config(
+        model=ModelConfig(input_size=12, hidden_size=10, num_layers=3,
+                          num_classes=7),
+        train=TrainConfig(batch_size=4, seq_len=8, steps=1, log_interval=0,
+                          ckpt_dir=str(tmp_path)),
+        data=DataConfig(task="copy"),
+    ))
+    try:
+        load_weights_safetensors(path, bad.model)
+        assert False, "expected mismatch error"
+    except ValueError as e:
+        assert "mismatch" in str(e)
