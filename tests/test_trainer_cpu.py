@@ -244,3 +244,28 @@ def test_serving_example_endpoint():
     finally:
         if sys_path_added:
             _sys.path.remove(ex)
+
+
+def test_serving_safetensors_load(tmp_path):
+    """serve.load_model accepts a .safetensors deployment artifact with
+    the model config riding in the file metadata."""
+    import argparse
+    import json
+    import examples.serve as serve
+    from lstm_tensorspark_amd.checkpoint import export_weights_safetensors
+    from lstm_tensorspark_amd.config import ModelConfig
+    from lstm_tensorspark_amd.models.lstm import LSTMModel
+
+    cfg = ModelConfig(input_size=32, hidden_size=32, num_layers=2,
+                      head="lm", vocab_size=64)
+    src = LSTMModel(cfg)
+    path = str(tmp_path / "w.safetensors")
+    export_weights_safetensors(path, src, metadata={
+        "config": json.dumps({"input_size": 32, "hidden_size": 32,
+                              "num_layers": 2, "head": "lm",
+                              "vocab_size": 64})})
+    args = argparse.Namespace(demo=False, ckpt=path, hidden=999, vocab=999)
+    model, device = serve.load_model(args)
+    assert model.cfg.vocab_size == 64          # config came from metadata
+    for k, v in src.weight_dict().items():
+        assert torch.equal(v.to(device), model.weight_dict()[k])
