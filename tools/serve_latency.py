@@ -22,14 +22,14 @@ sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 from examples.serve import build_app, load_model  # noqa: E402
 
 
-def main():
+def main(argv=None):
     ap = argparse.ArgumentParser()
     ap.add_argument("--hidden", type=int, default=1024)
     ap.add_argument("--vocab", type=int, default=50000)
     ap.add_argument("--steps", type=int, default=64)
     ap.add_argument("--reps", type=int, default=20)
     ap.add_argument("--warmup", type=int, default=3)
-    args = ap.parse_args()
+    args = ap.parse_args(argv)
     args.demo, args.ckpt = True, ""
     model, device = load_model(args)
     app = build_app(model, device)
@@ -47,13 +47,15 @@ def main():
         assert r.status_code == 200 and len(r.json()["generated"]) == args.steps
     lat.sort()
     p50 = statistics.median(lat)
-    print(json.dumps({
+    result = {
         "metric": "serve_latency", "device": str(device),
         "hidden": args.hidden, "vocab": args.vocab, "steps": args.steps,
         "reps": args.reps, "p50_ms": round(p50, 3),
         "p95_ms": round(lat[int(0.95 * len(lat)) - 1], 3),
         "tok_per_s": round(args.steps / (p50 / 1e3), 1),
-    }))
+    }
+    print(json.dumps(result))
+    return result
 
 
 if __name__ == "__main__":
