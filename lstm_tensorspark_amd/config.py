@@ -77,8 +77,11 @@ class DistConfig:
 
 @dataclass
 class DataConfig:
-    task: str = "random"          # "random" | "copy" | "lm" | "lm_assoc"
+    task: str = "random"          # "random" | "copy" | "lm" | "lm_assoc" | "lm_file"
     copy_delay: int = 4           # for the learnable copy task
+    path: str = ""                # lm_file: token file (.bin uint16/uint32
+                                  # memmap, or raw text read byte-level)
+    file_dtype: str = "uint16"    # lm_file .bin element type
 
 
 @dataclass

@@ -115,3 +115,86 @@ def test_eval_stream_disjoint_from_train():
     xt, _ = d.batch(5, stream=0)
     xe, _ = d.batch(5, stream=1)
     assert not torch.equal(xt, xe)
+
+
+# ---------------------------------------------------------------- lm_file
+
+
+def _file_cfg(path, vocab=300, file_dtype="uint16", B=4, T=8):
+    from lstm_tensorspark_amd.config import (Config, ModelConfig,
+                                             TrainConfig, DataConfig)
+    return Config(
+        model=ModelConfig(input_size=16, hidden_size=16, num_layers=1,
+                          head="lm", vocab_size=vocab),
+        train=TrainConfig(batch_size=B, seq_len=T, seed=11, steps=2,
+                          log_interval=0),
+        data=DataConfig(task="lm_file", path=path, file_dtype=file_dtype),
+    )
+
+
+def test_lm_file_bin_deterministic_and_shifted(tmp_path):
+    import numpy as np
+    from lstm_tensorspark_amd.data import SyntheticSeqData
+    toks = np.random.default_rng(0).integers(0, 300, 4096).astype(np.uint16)
+    p = str(tmp_path / "corpus.bin")
+    toks.tofile(p)
+    d = SyntheticSeqData(_file_cfg(p))
+    x, y = d.batch(3)
+    assert x.shape == (8, 4) and y.shape == (8, 4) and x.dtype == torch.long
+    assert torch.equal(x[1:], y[:-1])          # next-token shift
+    x2, y2 = d.batch(3)
+    assert torch.equal(x, x2) and torch.equal(y, y2)   # resumable
+    # windows really come from the file
+    flat = torch.from_numpy(toks.astype("int64"))
+    col = torch.cat([x[:, 0], y[-1:, 0]])
+    starts = (flat.unfold(0, 9, 1) == col).all(-1)
+    assert bool(starts.any())
+
+
+def test_lm_file_ranks_disjoint_streams(tmp_path):
+    import numpy as np
+    from lstm_tensorspark_amd.data import SyntheticSeqData
+    np.random.default_rng(1).integers(0, 300, 4096).astype(
+        np.uint16).tofile(str(tmp_path / "c.bin"))
+    cfg = _file_cfg(str(tmp_path / "c.bin"))
+    d0 = SyntheticSeqData(cfg, rank=0, world_size=2)
+    d1 = SyntheticSeqData(cfg, rank=1, world_size=2)
+    x0, _ = d0.batch(0)
+    x1, _ = d1.batch(0)
+    assert not torch.equal(x0, x1)
+
+
+def test_lm_file_text_byte_level(tmp_path):
+    from lstm_tensorspark_amd.data import SyntheticSeqData
+    p = str(tmp_path / "corpus.txt")
+    open(p, "w").write("the quick brown fox jumps over the lazy dog " * 40)
+    d = SyntheticSeqData(_file_cfg(p, vocab=256))
+    x, y = d.batch(0)
+    assert x.max() < 256 and torch.equal(x[1:], y[:-1])
+
+
+def test_lm_file_trains(tmp_path):
+    import numpy as np
+    from lstm_tensorspark_amd.trainer import Trainer
+    np.random.default_rng(2).integers(0, 300, 2048).astype(
+        np.uint16).tofile(str(tmp_path / "c.bin"))
+    t = Trainer(_file_cfg(str(tmp_path / "c.bin")))
+    t.fit(2)
+    assert t.step == 2
+
+
+def test_lm_file_errors(tmp_path):
+    import numpy as np
+    import pytest
+    from lstm_tensorspark_amd.data import SyntheticSeqData
+    # out-of-range token vs vocab
+    np.array([5, 999], dtype=np.uint16).tofile(str(tmp_path / "bad.bin"))
+    with pytest.raises(ValueError, match="out of range"):
+        SyntheticSeqData(_file_cfg(str(tmp_path / "bad.bin"),
+                                   vocab=100)).batch(0)
+    # corpus shorter than a window
+    np.arange(4, dtype=np.uint16).tofile(str(tmp_path / "short.bin"))
+    with pytest.raises(ValueError, match="tokens"):
+        SyntheticSeqData(_file_cfg(str(tmp_path / "short.bin"))).batch(0)
+    with pytest.raises(ValueError, match="requires data.path"):
+        SyntheticSeqData(_file_cfg("")).batch(0)
