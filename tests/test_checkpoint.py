@@ -142,3 +142,31 @@ def test_safetensors_export_roundtrip(tmp_path):
         assert False, "expected mismatch error"
     except ValueError as e:
         assert "mismatch" in str(e)
+
+
+def test_ckpt_to_safetensors_tool(tmp_path):
+    """End-to-end deployment pipeline: train -> .pt checkpoint ->
+    converter tool -> .safetensors -> serve.load_model."""
+    import argparse
+    from tools.ckpt_to_safetensors import convert
+    import examples.serve as serve
+
+    cfg = Config(
+        model=ModelConfig(input_size=16, hidden_size=16, num_layers=1,
+                          head="lm", vocab_size=32),
+        train=TrainConfig(batch_size=4, seq_len=8, steps=2, log_interval=0,
+                          ckpt_dir=str(tmp_path)),
+        data=DataConfig(task="lm"),
+    )
+    t = Trainer(cfg)
+    t.fit(2)
+    src = os.path.join(str(tmp_path), "c.pt")
+    save_checkpoint(src, t.model, t.optimizer, step=2, config=cfg.to_dict())
+    dst = os.path.join(str(tmp_path), "m.safetensors")
+    meta = convert(src, dst)
+    assert meta["step"] == "2"
+    args = argparse.Namespace(demo=False, ckpt=dst, hidden=1, vocab=1)
+    model, _ = serve.load_model(args)
+    assert model.cfg.vocab_size == 32
+    for k, v in t.model.weight_dict().items():
+        assert torch.equal(v, model.weight_dict()[k]), k
