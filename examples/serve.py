@@ -31,6 +31,8 @@ from lstm_tensorspark_amd.models.lstm import LSTMModel
 
 
 def build_app(model: LSTMModel, device: torch.device):
+    import time
+
     from fastapi import FastAPI, HTTPException
     from pydantic import BaseModel
 
@@ -41,6 +43,31 @@ def build_app(model: LSTMModel, device: torch.device):
 
     app = FastAPI(title="lstm_tensorspark_amd serving")
 
+    # Prometheus metrics (optional: /metrics 404s if the client lib is
+    # absent; the serving path itself never depends on it)
+    try:
+        from prometheus_client import (CONTENT_TYPE_LATEST,
+                                       CollectorRegistry, Counter,
+                                       Histogram, generate_latest)
+        from starlette.responses import Response
+        reg = CollectorRegistry()  # per-app: the default registry is
+        # process-global and double-registers if build_app runs twice
+        m_reqs = Counter("serve_generate_requests_total",
+                         "generate requests", ["status"], registry=reg)
+        m_toks = Counter("serve_generated_tokens_total", "tokens generated",
+                         registry=reg)
+        m_lat = Histogram("serve_generate_seconds",
+                          "end-to-end /generate latency",
+                          buckets=(.002, .005, .01, .025, .05, .1, .25,
+                                   .5, 1., 2.5), registry=reg)
+
+        @app.get("/metrics")
+        def metrics():
+            return Response(generate_latest(reg),
+                            media_type=CONTENT_TYPE_LATEST)
+    except ImportError:  # pragma: no cover
+        m_reqs = m_toks = m_lat = None
+
     @app.get("/health")
     def health():
         return {"ok": True, "device": str(device),
@@ -48,16 +75,25 @@ def build_app(model: LSTMModel, device: torch.device):
 
     @app.post("/generate")
     def generate(req: GenRequest):
+        t0 = time.perf_counter()
         V = model.cfg.vocab_size
         if not req.tokens or any(t < 0 or t >= V for t in req.tokens):
+            if m_reqs:
+                m_reqs.labels(status="bad_request").inc()
             raise HTTPException(400, f"tokens must be in [0, {V})")
         if req.steps < 1 or req.steps > 4096:
+            if m_reqs:
+                m_reqs.labels(status="bad_request").inc()
             raise HTTPException(400, "steps must be in [1, 4096]")
         prompt = torch.tensor(req.tokens, dtype=torch.long,
                               device=device).unsqueeze(1)  # (Tp, B=1)
         with torch.no_grad():
             out = model.generate(prompt, steps=req.steps,
                                  temperature=req.temperature)
+        if m_reqs:
+            m_reqs.labels(status="ok").inc()
+            m_toks.inc(req.steps)
+            m_lat.observe(time.perf_counter() - t0)
         return {"tokens": out[:, 0].tolist(),
                 "generated": out[len(req.tokens):, 0].tolist()}
 

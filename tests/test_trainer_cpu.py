@@ -278,3 +278,27 @@ def test_serve_latency_tool_degenerate():
     r = main(["--hidden", "32", "--vocab", "64", "--steps", "4",
               "--reps", "2", "--warmup", "1"])
     assert r["p50_ms"] > 0 and r["tok_per_s"] > 0 and r["device"] == "cpu"
+
+
+def test_serving_prometheus_metrics():
+    """/metrics exposes request count / token count / latency histogram
+    after a successful and a rejected request."""
+    import torch as _t
+    from fastapi.testclient import TestClient
+    from examples.serve import build_app
+    from lstm_tensorspark_amd.config import ModelConfig
+    from lstm_tensorspark_amd.models.lstm import LSTMModel
+
+    cfg = ModelConfig(input_size=16, hidden_size=16, num_layers=1,
+                      head="lm", vocab_size=32)
+    app = build_app(LSTMModel(cfg), _t.device("cpu"))
+    client = TestClient(app)
+    ok = client.post("/generate", json={"tokens": [1, 2], "steps": 3})
+    assert ok.status_code == 200
+    bad = client.post("/generate", json={"tokens": [999], "steps": 3})
+    assert bad.status_code == 400
+    text = client.get("/metrics").text
+    assert 'serve_generate_requests_total{status="ok"} 1.0' in text
+    assert 'serve_generate_requests_total{status="bad_request"} 1.0' in text
+    assert "serve_generated_tokens_total 3.0" in text
+    assert "serve_generate_seconds_bucket" in text
