@@ -40,6 +40,8 @@ def build_app(model: LSTMModel, device: torch.device):
         tokens: list[int]
         steps: int = 32
         temperature: float = 0.0
+        top_k: int = 0
+        top_p: float = 0.0
 
     app = FastAPI(title="lstm_tensorspark_amd serving")
 
@@ -89,7 +91,8 @@ def build_app(model: LSTMModel, device: torch.device):
                               device=device).unsqueeze(1)  # (Tp, B=1)
         with torch.no_grad():
             out = model.generate(prompt, steps=req.steps,
-                                 temperature=req.temperature)
+                                 temperature=req.temperature,
+                                 top_k=req.top_k, top_p=req.top_p)
         if m_reqs:
             m_reqs.labels(status="ok").inc()
             m_toks.inc(req.steps)

@@ -300,13 +300,16 @@ class LSTMModel(nn.Module):
                  temperature: float = 0.0,
                  generator: "Optional[torch.Generator]" = None,
                  use_graph: Optional[bool] = None,
-                 use_gemv: Optional[bool] = None) -> torch.Tensor:
+                 use_gemv: Optional[bool] = None,
+                 top_k: int = 0, top_p: float = 0.0) -> torch.Tensor:
         """Autoregressive decode for the LM head (the serving path).
 
         prompt: (Tp, B) long tokens (time-major).  Returns (Tp+steps, B):
         the prompt followed by `steps` generated tokens.  temperature 0 =
         greedy argmax; >0 = softmax sampling.  The recurrent (h, c) state
         is carried across steps, so each decode step is one T=1 pass.
+        top_k / top_p restrict sampling to the k most likely tokens /
+        the smallest nucleus with mass >= top_p (sampling path only).
 
         use_graph (default: auto = greedy + CUDA + steps >= 16): capture
         ONE decode step as a hipGraph whose body copies its outputs back
@@ -321,9 +324,11 @@ class LSTMModel(nn.Module):
         self.eval()
         if use_graph is None:
             use_graph = (temperature <= 0 and prompt.is_cuda and steps >= 16)
+        assert not (use_graph and temperature > 0), \
+            "the graphed decode step is greedy-only"
         try:
             out, state = self.forward(prompt)          # (Tp, B, V)
-            cur = self._sample(out[-1], temperature, generator)
+            cur = self._sample(out[-1], temperature, generator, top_k, top_p)
             if use_graph:
                 return torch.cat(
                     [prompt, self._graph_decode(cur, state, steps, use_gemv)],
@@ -332,7 +337,8 @@ class LSTMModel(nn.Module):
             for _ in range(steps - 1):
                 toks.append(cur.unsqueeze(0))
                 out, state = self.forward(cur.unsqueeze(0), state)
-                cur = self._sample(out[-1], temperature, generator)
+                cur = self._sample(out[-1], temperature, generator,
+                                   top_k, top_p)
             toks.append(cur.unsqueeze(0))
             return torch.cat(toks, dim=0)
         finally:
@@ -456,8 +462,21 @@ class LSTMModel(nn.Module):
         return outs[0] if len(outs) == 1 else torch.cat(outs, dim=0)
 
     @staticmethod
-    def _sample(logits, temperature, generator):
+    def _sample(logits, temperature, generator, top_k=0, top_p=0.0):
         if temperature <= 0:
             return logits.argmax(dim=-1)
-        probs = torch.softmax(logits.float() / temperature, dim=-1)
+        logits = logits.float() / temperature
+        if top_k and top_k > 0 and top_k < logits.shape[-1]:
+            kth = logits.topk(top_k, dim=-1).values[..., -1:]
+            logits = logits.masked_fill(logits < kth, float("-inf"))
+        if 0.0 < top_p < 1.0:
+            # nucleus: keep the smallest prefix of the sorted distribution
+            # whose mass reaches top_p (always at least the top token)
+            srt, idx = logits.sort(dim=-1, descending=True)
+            cum = torch.softmax(srt, dim=-1).cumsum(dim=-1)
+            drop = cum - torch.softmax(srt, dim=-1) >= top_p
+            srt = srt.masked_fill(drop, float("-inf"))
+            logits = torch.full_like(logits, float("-inf")).scatter_(
+                -1, idx, srt)
+        probs = torch.softmax(logits, dim=-1)
         return torch.multinomial(probs, 1, generator=generator).squeeze(-1)

@@ -302,3 +302,43 @@ def test_serving_prometheus_metrics():
     assert 'serve_generate_requests_total{status="bad_request"} 1.0' in text
     assert "serve_generated_tokens_total 3.0" in text
     assert "serve_generate_seconds_bucket" in text
+
+
+def test_generate_top_k_top_p_cpu():
+    """top_k restricts samples to the k highest-logit tokens; top_p=tiny
+    collapses to greedy; both compose with seeded reproducibility."""
+    from lstm_tensorspark_amd.config import ModelConfig
+    from lstm_tensorspark_amd.models.lstm import LSTMModel
+
+    torch.manual_seed(6)
+    cfg = ModelConfig(input_size=16, hidden_size=16, num_layers=1,
+                      head="lm", vocab_size=32)
+    m = LSTMModel(cfg)
+    prompt = torch.randint(0, 32, (4, 3))
+
+    # top_k=1 must equal greedy regardless of temperature
+    g = torch.Generator().manual_seed(1)
+    a = m.generate(prompt, steps=6, temperature=2.0, top_k=1, generator=g)
+    assert torch.equal(a, m.generate(prompt, steps=6, temperature=0.0))
+
+    # a minuscule nucleus also collapses to greedy
+    g = torch.Generator().manual_seed(1)
+    b = m.generate(prompt, steps=6, temperature=2.0, top_p=1e-9, generator=g)
+    assert torch.equal(b, m.generate(prompt, steps=6, temperature=0.0))
+
+    # top_k=5: every generated token is within that step's top-5 logits
+    g = torch.Generator().manual_seed(2)
+    c = m.generate(prompt, steps=8, temperature=1.5, top_k=5, generator=g)
+    logits, _ = m(c)
+    for t in range(4, 12):
+        top5 = logits[t - 1].topk(5, dim=-1).indices
+        assert (c[t].unsqueeze(-1) == top5).any(-1).all()
+
+    # seeded reproducibility holds with both filters on
+    g1 = torch.Generator().manual_seed(3)
+    g2 = torch.Generator().manual_seed(3)
+    d1 = m.generate(prompt, steps=6, temperature=1.0, top_k=8, top_p=0.9,
+                    generator=g1)
+    d2 = m.generate(prompt, steps=6, temperature=1.0, top_k=8, top_p=0.9,
+                    generator=g2)
+    assert torch.equal(d1, d2)
