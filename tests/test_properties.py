@@ -1,0 +1,102 @@
+"""Property-based tests (hypothesis) for invariant-heavy CPU paths:
+the RNG stream mixer, the sampling filters, and config serialization.
+These complement the example-based suites with randomized coverage of
+the full input envelopes.
+"""
+import torch
+from hypothesis import given, settings
+from hypothesis import strategies as st
+
+from lstm_tensorspark_amd.data import mix64
+
+_SMALL = dict(max_examples=50, deadline=None)
+
+
+@given(seed=st.integers(0, 2**31 - 1), stream=st.integers(0, 3),
+       rank=st.integers(0, 63), step=st.integers(0, 10**7))
+@settings(**_SMALL)
+def test_mix64_in_u64_range(seed, stream, rank, step):
+    v = mix64(seed, stream, rank, step)
+    assert 0 <= v < 2**64
+
+
+@given(seed=st.integers(0, 2**20), rank=st.integers(0, 7),
+       step=st.integers(0, 10**6), d_rank=st.integers(1, 7),
+       d_step=st.integers(1, 10**6))
+@settings(**_SMALL)
+def test_mix64_rank_step_disjoint(seed, rank, step, d_rank, d_step):
+    """No (rank, step) pair may alias another's stream -- the exact
+    failure mode of the r1 linear mix (rank 0 step 7919 == rank 1 step 0)."""
+    base = mix64(seed, 0, rank, step)
+    assert base != mix64(seed, 0, rank + d_rank, step)
+    assert base != mix64(seed, 0, rank, step + d_step)
+    assert base != mix64(seed, 1, rank, step)  # train vs eval stream
+
+
+@given(order=st.permutations([11, 22, 33]))
+@settings(**_SMALL)
+def test_mix64_order_sensitive(order):
+    """The mix must absorb position, not just values (a commutative mix
+    would alias (rank=a, step=b) with (rank=b, step=a))."""
+    if tuple(order) != (11, 22, 33):
+        assert mix64(*order) != mix64(11, 22, 33)
+
+
+@given(vocab=st.integers(4, 200), k=st.integers(1, 200),
+       seed=st.integers(0, 1000))
+@settings(**_SMALL)
+def test_sample_top_k_support(vocab, k, seed):
+    """Sampled tokens always lie inside the top-k logit set; top_k >= V
+    degrades to plain sampling without error."""
+    from lstm_tensorspark_amd.models.lstm import LSTMModel
+    g = torch.Generator().manual_seed(seed)
+    logits = torch.randn(3, vocab, generator=g)
+    tok = LSTMModel._sample(logits, temperature=1.0, generator=g, top_k=k)
+    assert tok.shape == (3,)
+    kk = min(k, vocab)
+    topk = logits.topk(kk, dim=-1).indices
+    assert (tok.unsqueeze(-1) == topk).any(-1).all()
+
+
+@given(vocab=st.integers(4, 200), p=st.floats(0.05, 0.999),
+       seed=st.integers(0, 1000))
+@settings(**_SMALL)
+def test_sample_top_p_support(vocab, p, seed):
+    """Nucleus sampling only emits tokens from the smallest prefix of the
+    sorted distribution whose mass reaches p (plus the boundary token)."""
+    from lstm_tensorspark_amd.models.lstm import LSTMModel
+    g = torch.Generator().manual_seed(seed)
+    logits = torch.randn(2, vocab, generator=g)
+    tok = LSTMModel._sample(logits, temperature=1.0, generator=g, top_p=p)
+    probs = torch.softmax(logits, dim=-1)
+    srt, idx = probs.sort(dim=-1, descending=True)
+    cum = srt.cumsum(-1)
+    for r in range(2):
+        nucleus_n = int((cum[r] - srt[r] < p).sum())  # tokens kept
+        assert (tok[r] == idx[r, :nucleus_n]).any()
+
+
+@given(seed=st.integers(0, 2**31 - 1))
+@settings(max_examples=20, deadline=None)
+def test_lm_assoc_multiplier_always_bijective(seed):
+    """For any config seed the lm_assoc affine map must be a bijection on
+    Z_V (gcd(a, V) == 1) for awkward V too."""
+    import math
+    for V in (50000, 65536, 99, 30030):
+        a = 2 * (seed % 1000) + 1
+        while math.gcd(a, V) != 1:
+            a += 2
+        assert math.gcd(a, V) == 1
+        assert len({(a * t) % V for t in range(min(V, 512))}) == min(V, 512)
+
+
+@given(hidden=st.integers(8, 64), layers=st.integers(1, 3),
+       bptt=st.integers(0, 16), clip=st.floats(0.0, 10.0))
+@settings(max_examples=15, deadline=None)
+def test_config_dict_roundtrip(hidden, layers, bptt, clip):
+    from lstm_tensorspark_amd.config import Config, ModelConfig, TrainConfig
+    cfg = Config(model=ModelConfig(input_size=8, hidden_size=hidden,
+                                   num_layers=layers),
+                 train=TrainConfig(bptt=bptt, grad_clip=clip))
+    back = Config.from_dict(cfg.to_dict())
+    assert back.to_dict() == cfg.to_dict()
