@@ -198,3 +198,16 @@ def test_lm_file_errors(tmp_path):
         SyntheticSeqData(_file_cfg(str(tmp_path / "short.bin"))).batch(0)
     with pytest.raises(ValueError, match="requires data.path"):
         SyntheticSeqData(_file_cfg("")).batch(0)
+
+
+def test_lm_file_eval_stream_disjoint(tmp_path):
+    """evaluate()'s stream=1 draws different windows than training
+    stream=0 at the same step for file-backed data too."""
+    import numpy as np
+    from lstm_tensorspark_amd.data import SyntheticSeqData
+    np.random.default_rng(3).integers(0, 300, 8192).astype(
+        np.uint16).tofile(str(tmp_path / "c.bin"))
+    d = SyntheticSeqData(_file_cfg(str(tmp_path / "c.bin")))
+    x_tr, _ = d.batch(5, stream=0)
+    x_ev, _ = d.batch(5, stream=1)
+    assert not torch.equal(x_tr, x_ev)
