@@ -26,7 +26,7 @@ import torch
 
 sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
-from lstm_tensorspark_amd.config import Config, ModelConfig, TrainConfig, DataConfig
+from lstm_tensorspark_amd.config import Config, ModelConfig
 from lstm_tensorspark_amd.models.lstm import LSTMModel
 
 

@@ -15,7 +15,6 @@ import statistics
 import sys
 import time
 
-import torch
 
 sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
