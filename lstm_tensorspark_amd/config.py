@@ -45,6 +45,13 @@ class TrainConfig:
                                   # bf16 backward); quality-verified with bf16
                                   # recurrence -- the lm-50k speed knob
     steps: int = 100
+    accum_steps: int = 1          # micro-batches accumulated per optimizer
+                                  # step (loss scaled 1/accum; effective
+                                  # batch = batch_size*accum_steps*world)
+    lr_schedule: str = ""         # "" constant | "cosine" | "linear" decay
+                                  # from lr to min_lr over train.steps
+    warmup_steps: int = 0         # linear LR warmup 0 -> lr first
+    min_lr: float = 0.0           # decay floor for cosine/linear
     log_interval: int = 10
     ckpt_interval: int = 0        # 0 => only at end if ckpt_dir set
     ckpt_dir: str = ""

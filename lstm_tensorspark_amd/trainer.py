@@ -90,9 +90,12 @@ class Trainer:
                     if self.ctx.is_main:
                         self.log.info(f"autotune fwd path: {r}")
         single_window = cfg.train.bptt <= 0 or cfg.train.bptt >= cfg.train.seq_len
+        no_accum = getattr(cfg.train, "accum_steps", 1) <= 1
+        # side_dw is only safe when autograd STORES weight grads; both BPTT
+        # windows and micro-batch accumulation make it ACCUMULATE them
         enable_side_dw(getattr(cfg.dist, "side_dw", False)
                        and self.bucketer is None and single_window
-                       and self.device.type == "cuda")
+                       and no_accum and self.device.type == "cuda")
         self.step = 0
         self.history: list[dict] = []
         # hipGraph capture state (train.cuda_graph)
@@ -102,7 +105,11 @@ class Trainer:
             cfg.train.cuda_graph and self.device.type == "cuda"
             and self.ctx.world_size == 1 and self.bucketer is None
             and cfg.train.optimizer == "sgd" and single_window
-            and cfg.model.dropout == 0.0)
+            and cfg.model.dropout == 0.0
+            # lr is baked into the captured kernels; accumulation needs
+            # multiple backwards per step -- both fall back to eager
+            and no_accum and not cfg.train.lr_schedule
+            and cfg.train.warmup_steps <= 0)
 
     # ------------------------------------------------------------------
     def _bptt_windows(self, x, y):
@@ -119,34 +126,65 @@ class Trainer:
             wins.append((x[s:e], y[s:e] if per_step_y else y))
         return wins
 
+    def _lr_at(self, step: int) -> float:
+        """LR schedule as a pure function of the optimizer step (so resume
+        lands on the same value): linear warmup 0 -> lr over warmup_steps,
+        then constant / cosine / linear decay to min_lr at train.steps."""
+        t = self.cfg.train
+        if t.warmup_steps > 0 and step < t.warmup_steps:
+            return t.lr * (step + 1) / t.warmup_steps
+        if not t.lr_schedule:
+            return t.lr
+        total = max(t.steps, t.warmup_steps + 1)
+        frac = min(1.0, (step - t.warmup_steps)
+                   / max(1, total - t.warmup_steps))
+        if t.lr_schedule == "cosine":
+            import math
+            c = 0.5 * (1.0 + math.cos(math.pi * frac))
+        elif t.lr_schedule == "linear":
+            c = 1.0 - frac
+        else:
+            raise ValueError(f"unknown lr_schedule {t.lr_schedule!r}")
+        return t.min_lr + (t.lr - t.min_lr) * c
+
     def train_step(self, batch) -> dict:
+        """One optimizer step.  `batch` is one (x, y) pair, or a list of
+        them for gradient accumulation (grads are summed across
+        micro-batches with the loss scaled 1/n, so the update equals one
+        step on the concatenated batch)."""
         cfg = self.cfg
         if cfg.train.fault_step == self.step and cfg.train.fault_rank == self.ctx.rank:
             raise RuntimeError(
                 f"injected fault on rank {self.ctx.rank} at step {self.step}")
-        x, y = batch
+        batches = batch if isinstance(batch, list) else [batch]
         if self._graph_eligible:
-            return self._graphed_step(x, y)
+            return self._graphed_step(*batches[0])
         if self.bucketer is not None:
             self.bucketer.zero_grad()
         else:
             self.optimizer.zero_grad()
 
         total_loss = 0.0
-        wins = self._bptt_windows(x, y)
-        state = None
-        for wi, (wx, wy) in enumerate(wins):
-            out, state = self.model(wx, state)
-            state = self.model.detach_state(state)
-            loss = self.model.loss(out, wy)
-            if len(wins) > 1:
-                loss = loss / len(wins)
-            if self.bucketer is not None:
-                # grads accumulate across windows into the bucket buffers;
-                # only the LAST window's backward launches the all-reduces.
-                self.bucketer.start_backward(will_reduce=(wi == len(wins) - 1))
-            loss.backward()
-            total_loss += float(loss.detach())
+        n_micro = len(batches)
+        for bi, (x, y) in enumerate(batches):
+            wins = self._bptt_windows(x, y)
+            state = None
+            for wi, (wx, wy) in enumerate(wins):
+                out, state = self.model(wx, state)
+                state = self.model.detach_state(state)
+                loss = self.model.loss(out, wy)
+                scale = len(wins) * n_micro
+                if scale > 1:
+                    loss = loss / scale
+                if self.bucketer is not None:
+                    # grads accumulate across windows/micro-batches into the
+                    # bucket buffers; only the LAST backward launches the
+                    # all-reduces.
+                    self.bucketer.start_backward(
+                        will_reduce=(bi == n_micro - 1
+                                     and wi == len(wins) - 1))
+                loss.backward()
+                total_loss += float(loss.detach())
         if self.bucketer is not None:
             self.bucketer.finalize()
         # grads may still be in flight on the dW side stream
@@ -155,6 +193,8 @@ class Trainer:
         norm = None
         if cfg.train.grad_clip > 0:
             norm = self.optimizer.clip_grad_norm(cfg.train.grad_clip)
+        if cfg.train.lr_schedule or cfg.train.warmup_steps > 0:
+            self.optimizer.lr = self._lr_at(self.step)
         self.optimizer.step()
         # invalidate serving-weight caches (model.generate's gemv layout)
         self.model._weights_version = getattr(
@@ -265,11 +305,19 @@ class Trainer:
         t_last = time.perf_counter()
         seqs_since = 0
         start_step = self.step
+        accum = getattr(cfg.train, "accum_steps", 1)
         while self.step < start_step + steps:
-            batch = self.data.batch(self.step)
+            if accum > 1:
+                # micro-batch data streams: step s consumes data steps
+                # [s*accum, (s+1)*accum) so every micro-batch is distinct
+                # and the mapping is resume-deterministic
+                batch = [self.data.batch(self.step * accum + i)
+                         for i in range(accum)]
+            else:
+                batch = self.data.batch(self.step)
             info = self.train_step(batch)
             self.step += 1
-            seqs_since += cfg.train.batch_size * self.ctx.world_size
+            seqs_since += cfg.train.batch_size * self.ctx.world_size * accum
             if cfg.train.log_interval and self.step % cfg.train.log_interval == 0:
                 if self.device.type == "cuda":
                     torch.cuda.synchronize()
