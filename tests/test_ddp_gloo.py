@@ -199,3 +199,40 @@ def test_2rank_measured_bucket_order_grads_correct(tmp_path):
     for name in got:
         want = (per_rank[0][name] + per_rank[1][name]) / 2
         assert (got[name] - want).abs().max() < 1e-6, name
+
+
+# ------------------------------------------------- gradient accumulation
+def _grads_accum_worker(ctx, tmpdir):
+    cfg = small_cfg(momentum=0.0)
+    cfg.train.accum_steps = 2
+    t = Trainer(cfg, dist_ctx=ctx)
+    t.train_step([t.data.batch(0), t.data.batch(1)])
+    grads = {n: p.grad.detach().clone()
+             for n, p in zip(t.model.weight_dict(), t.model.parameters())}
+    torch.save(grads, os.path.join(tmpdir, f"ag_rank{ctx.rank}.pt"))
+
+
+def test_2rank_accumulation_grads_correct(tmp_path):
+    """accum_steps=2 at world 2: only the LAST micro-batch's backward may
+    launch the all-reduces (earlier reduction would double-count), and the
+    final grad must equal the rank-average of each rank's two-micro-batch
+    mean gradient."""
+    run_spawned(_grads_accum_worker, world_size=2, backend="gloo",
+                args=(str(tmp_path),), master_port=29541)
+    per_rank = []
+    for rank in range(2):
+        cfg = small_cfg(momentum=0.0)
+        cfg.train.accum_steps = 2
+        t = Trainer(cfg)
+        t.data.rank = rank
+        t.train_step([t.data.batch(0), t.data.batch(1)])
+        per_rank.append({n: p.grad.detach().clone()
+                         for n, p in zip(t.model.weight_dict(),
+                                         t.model.parameters())})
+    for rank in range(2):
+        got = torch.load(os.path.join(str(tmp_path), f"ag_rank{rank}.pt"),
+                         weights_only=False)
+        for name in got:
+            want = (per_rank[0][name] + per_rank[1][name]) / 2
+            err = (got[name] - want).abs().max()
+            assert err < 1e-6, f"{name}: {err}"
