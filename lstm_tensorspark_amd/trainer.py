@@ -158,6 +158,11 @@ class Trainer:
                 f"injected fault on rank {self.ctx.rank} at step {self.step}")
         batches = batch if isinstance(batch, list) else [batch]
         if self._graph_eligible:
+            if len(batches) != 1:
+                raise ValueError(
+                    "micro-batch lists need accum_steps > 1 in the config "
+                    "(graph capture is single-backward; it would silently "
+                    "drop the extra micro-batches)")
             return self._graphed_step(*batches[0])
         if self.bucketer is not None:
             self.bucketer.zero_grad()

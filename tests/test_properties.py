@@ -100,3 +100,36 @@ def test_config_dict_roundtrip(hidden, layers, bptt, clip):
                  train=TrainConfig(bptt=bptt, grad_clip=clip))
     back = Config.from_dict(cfg.to_dict())
     assert back.to_dict() == cfg.to_dict()
+
+
+@given(n=st.integers(10, 5000), t=st.integers(1, 64), b=st.integers(1, 8),
+       step=st.integers(0, 1000), rank=st.integers(0, 7))
+@settings(max_examples=25, deadline=None)
+def test_lm_file_windows_always_in_bounds(tmp_path_factory, n, t, b, step,
+                                          rank):
+    """Any corpus length >= T+1 yields valid next-token windows for any
+    (rank, step); shorter corpora raise cleanly."""
+    import numpy as np
+    from lstm_tensorspark_amd.config import (Config, ModelConfig,
+                                             TrainConfig, DataConfig)
+    from lstm_tensorspark_amd.data import SyntheticSeqData
+    d = tmp_path_factory.mktemp("lmf")
+    path = str(d / "c.bin")
+    np.random.default_rng(n).integers(0, 250, n).astype(
+        np.uint16).tofile(path)
+    cfg = Config(
+        model=ModelConfig(input_size=8, hidden_size=8, num_layers=1,
+                          head="lm", vocab_size=256),
+        train=TrainConfig(batch_size=b, seq_len=t, seed=7),
+        data=DataConfig(task="lm_file", path=path),
+    )
+    src = SyntheticSeqData(cfg, rank=rank)
+    if n < t + 1:
+        import pytest
+        with pytest.raises(ValueError):
+            src.batch(step)
+        return
+    x, y = src.batch(step)
+    assert x.shape == (t, b) and y.shape == (t, b)
+    assert 0 <= int(x.min()) and int(x.max()) < 256
+    assert torch.equal(x[1:], y[:-1])
