@@ -89,6 +89,9 @@ class DataConfig:
     path: str = ""                # lm_file: token file (.bin uint16/uint32
                                   # memmap, or raw text read byte-level)
     file_dtype: str = "uint16"    # lm_file .bin element type
+    eval_frac: float = 0.0        # lm_file: reserve the corpus tail for
+                                  # eval (stream=1) -- a true held-out
+                                  # split; 0 = eval samples everywhere
 
 
 @dataclass

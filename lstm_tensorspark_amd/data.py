@@ -150,13 +150,23 @@ class SyntheticSeqData:
             # partitioning; no driver round-trip, no data shipping).
             arr = self._file_tokens()
             N = arr.shape[0]
-            if N < T + 1:
+            # held-out split: train windows start in [0, split-T), eval
+            # (stream=1) windows in [split, N-T) -- windows cannot cross
+            # the boundary, so the eval tail is never trained on
+            f = getattr(self.cfg.data, "eval_frac", 0.0)
+            lo, hi = 0, N - T
+            if 0.0 < f < 1.0:
+                split = int(N * (1.0 - f))
+                lo, hi = (split, N - T) if stream == 1 else (0, split - T)
+            if hi <= lo:
                 raise ValueError(
-                    f"lm_file corpus has {N} tokens < seq_len+1={T + 1}")
+                    f"lm_file corpus has {N} tokens: stream {stream} split "
+                    f"[{lo}, {hi}) has no room for seq_len+1={T + 1} "
+                    f"windows (eval_frac={f})")
             # offsets on CPU (memmap gather is a host op), batch to device
             gc = torch.Generator()
             gc.manual_seed(mix64(self.base_seed, stream, self.rank, step))
-            off = torch.randint(0, N - T, (B,), generator=gc)
+            off = torch.randint(lo, hi, (B,), generator=gc)
             idx = (off.unsqueeze(0)
                    + torch.arange(T + 1).unsqueeze(1)).numpy()  # (T+1, B)
             import numpy as np
