@@ -53,6 +53,8 @@ class TrainConfig:
     warmup_steps: int = 0         # linear LR warmup 0 -> lr first
     min_lr: float = 0.0           # decay floor for cosine/linear
     log_interval: int = 10
+    eval_interval: int = 0        # run evaluate() every N steps in fit()
+    eval_steps: int = 10          # batches per periodic eval
     ckpt_interval: int = 0        # 0 => only at end if ckpt_dir set
     ckpt_dir: str = ""
     seed: int = 1234

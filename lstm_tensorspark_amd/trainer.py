@@ -339,6 +339,17 @@ class Trainer:
                 if self.ctx.is_main:
                     self.log.info(json.dumps(rec))
                 t_last, seqs_since = now, 0
+            ei = getattr(cfg.train, "eval_interval", 0)
+            if ei and self.step % ei == 0:
+                # all ranks reach this at the same step (same counters), so
+                # evaluate()'s all-reduce stays collective-aligned
+                r = self.evaluate(steps=cfg.train.eval_steps)
+                rec = {"step": self.step,
+                       "eval_loss": round(r["eval_loss"], 6)}
+                self.history.append(rec)
+                if self.ctx.is_main:
+                    self.log.info(json.dumps(rec))
+                t_last = time.perf_counter()  # don't bill eval to seq/s
             if (cfg.train.ckpt_interval and cfg.train.ckpt_dir
                     and self.step % cfg.train.ckpt_interval == 0):
                 self.save(os.path.join(cfg.train.ckpt_dir,

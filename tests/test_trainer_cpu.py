@@ -430,3 +430,17 @@ def test_lr_schedule_applied_and_resume_consistent(tmp_path):
     t2.load(mid)
     t2.fit(1)
     assert abs(t.optimizer.lr - t2.optimizer.lr) < 1e-12
+
+
+def test_periodic_eval_in_fit():
+    """train.eval_interval logs eval_loss records into history at the
+    right steps without disturbing training."""
+    cfg = copy_cfg(steps=6, log_interval=0)
+    cfg.train.eval_interval = 2
+    cfg.train.eval_steps = 2
+    t = Trainer(cfg)
+    t.fit(6)
+    evals = [r for r in t.history if "eval_loss" in r]
+    assert [r["step"] for r in evals] == [2, 4, 6]
+    assert all(r["eval_loss"] == r["eval_loss"] for r in evals)
+    assert t.model.training  # back in train mode
