@@ -37,3 +37,20 @@ def test_train_cli_smoke(tmp_path):
         capture_output=True, text=True, timeout=300, cwd=REPO)
     assert out.returncode == 0, out.stderr[-2000:]
     assert os.path.exists(os.path.join(str(tmp_path), "ckpt_final.pt"))
+
+
+def test_train_cli_new_knobs(tmp_path):
+    """accum_steps / lr_schedule / eval_interval all plumb through the
+    generic --set CLI overrides into a real run."""
+    out = subprocess.run(
+        [sys.executable, os.path.join(REPO, "train.py"),
+         "--preset", "plumbing-cpu", "--set", "train.steps=4",
+         "--set", "train.batch_size=4", "--set", "train.seq_len=8",
+         "--set", "train.accum_steps=2", "--set", "train.lr_schedule=cosine",
+         "--set", "train.warmup_steps=2", "--set", "train.eval_interval=2",
+         "--set", "train.eval_steps=1", "--set", "train.log_interval=1",
+         "--set", f"train.ckpt_dir={tmp_path}"],
+        capture_output=True, text=True, timeout=300, cwd=REPO)
+    assert out.returncode == 0, out.stderr[-2000:]
+    assert "eval_loss" in out.stdout + out.stderr
+    assert os.path.exists(os.path.join(str(tmp_path), "ckpt_final.pt"))
