@@ -358,6 +358,22 @@ class Trainer:
             self.save(os.path.join(cfg.train.ckpt_dir, "ckpt_final.pt"))
         return self.history
 
+    def export_history(self, path: str):
+        """Write the step/eval history as CSV (pandas/spreadsheet-ready;
+        the TensorBoard package is not in this image, and the JSONL log +
+        this CSV cover the same observability need -- SURVEY.md §5
+        metrics row)."""
+        import csv
+        cols: list = []
+        for r in self.history:
+            for k in r:
+                if k not in cols:
+                    cols.append(k)
+        with open(path, "w", newline="") as f:
+            w = csv.DictWriter(f, fieldnames=cols)
+            w.writeheader()
+            w.writerows(self.history)
+
     # alias matching the reference's API surface
     train = fit
 

@@ -444,3 +444,19 @@ def test_periodic_eval_in_fit():
     assert [r["step"] for r in evals] == [2, 4, 6]
     assert all(r["eval_loss"] == r["eval_loss"] for r in evals)
     assert t.model.training  # back in train mode
+
+
+def test_export_history_csv(tmp_path):
+    cfg = copy_cfg(steps=4, log_interval=2)
+    cfg.train.eval_interval = 2
+    cfg.train.eval_steps = 1
+    t = Trainer(cfg)
+    t.fit(4)
+    p = str(tmp_path / "hist.csv")
+    t.export_history(p)
+    import csv
+    rows = list(csv.DictReader(open(p)))
+    assert len(rows) == len(t.history) >= 4
+    assert "loss" in rows[0] and "eval_loss" in rows[0]
+    steps = [int(r["step"]) for r in rows if r["step"]]
+    assert steps == sorted(steps)
