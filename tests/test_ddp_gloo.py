@@ -236,3 +236,24 @@ def test_2rank_accumulation_grads_correct(tmp_path):
             want = (per_rank[0][name] + per_rank[1][name]) / 2
             err = (got[name] - want).abs().max()
             assert err < 1e-6, f"{name}: {err}"
+
+
+def test_4rank_grads_equal_average_of_per_rank_grads(tmp_path):
+    """Same parity contract at world 4 (the bucketer's arming/finalize
+    logic is world-size independent; this guards the averaging factor)."""
+    run_spawned(_grads_worker, world_size=4, backend="gloo",
+                args=(str(tmp_path), 0), master_port=29551)
+    per_rank = []
+    for rank in range(4):
+        t = Trainer(small_cfg(bptt=0))
+        t.data.rank = rank
+        t.train_step(t.data.batch(0))
+        per_rank.append({n: p.grad.detach().clone()
+                         for n, p in zip(t.model.weight_dict(),
+                                         t.model.parameters())})
+    got = torch.load(os.path.join(str(tmp_path), "grads_rank0.pt"),
+                     weights_only=False)
+    for name in got:
+        want = sum(pr[name] for pr in per_rank) / 4
+        err = (got[name] - want).abs().max()
+        assert err < 1e-6, f"{name}: {err}"
