@@ -42,6 +42,7 @@ def build_app(model: LSTMModel, device: torch.device):
         temperature: float = 0.0
         top_k: int = 0
         top_p: float = 0.0
+        beam_size: int = 1
 
     app = FastAPI(title="lstm_tensorspark_amd serving")
 
@@ -89,16 +90,24 @@ def build_app(model: LSTMModel, device: torch.device):
             raise HTTPException(400, "steps must be in [1, 4096]")
         prompt = torch.tensor(req.tokens, dtype=torch.long,
                               device=device).unsqueeze(1)  # (Tp, B=1)
+        score = None
         with torch.no_grad():
-            out = model.generate(prompt, steps=req.steps,
-                                 temperature=req.temperature,
-                                 top_k=req.top_k, top_p=req.top_p)
+            if req.beam_size > 1:
+                out, score = model.generate_beam(prompt, steps=req.steps,
+                                                 beam_size=req.beam_size)
+            else:
+                out = model.generate(prompt, steps=req.steps,
+                                     temperature=req.temperature,
+                                     top_k=req.top_k, top_p=req.top_p)
         if m_reqs:
             m_reqs.labels(status="ok").inc()
             m_toks.inc(req.steps)
             m_lat.observe(time.perf_counter() - t0)
-        return {"tokens": out[:, 0].tolist(),
+        resp = {"tokens": out[:, 0].tolist(),
                 "generated": out[len(req.tokens):, 0].tolist()}
+        if score is not None:
+            resp["score"] = score
+        return resp
 
     return app
 

@@ -345,6 +345,57 @@ class LSTMModel(nn.Module):
             if was_training:
                 self.train()
 
+    @torch.no_grad()
+    def generate_beam(self, prompt: torch.Tensor, steps: int,
+                      beam_size: int = 4, eos_token: int = -1):
+        """Beam-search decode for a single prompt (B=1): the beams ride the
+        BATCH dimension, so every decode step is ONE batched forward (the
+        same kernels that serve batched greedy decode -- no per-beam loop).
+
+        Keeps the `beam_size` highest-sum-logprob prefixes each step.  If
+        `eos_token` >= 0, a beam that emits it is frozen (it keeps emitting
+        eos at no score change) so finished hypotheses compete with live
+        ones on equal footing.  Returns (tokens (Tp+steps, 1), score) --
+        the best beam and its total log-probability.
+        """
+        assert self.cfg.head == "lm", "beam search requires the LM head"
+        assert not self.cfg.bidirectional, "decode is causal: fwd-only"
+        assert prompt.dim() == 2 and prompt.shape[1] == 1, \
+            "generate_beam takes a single prompt (Tp, 1)"
+        assert beam_size >= 1
+        V = self.cfg.vocab_size
+        was_training = self.training
+        self.eval()
+        try:
+            out, state = self.forward(prompt)               # (Tp, 1, V)
+            logp = torch.log_softmax(out[-1, 0].float(), dim=-1)
+            k = min(beam_size, V)
+            scores, tok = logp.topk(k)                      # (k,)
+            state = [(h.expand(k, -1).contiguous(),
+                      c.expand(k, -1).contiguous()) for h, c in state]
+            seqs = tok.unsqueeze(0)                         # (t, k)
+            frozen = (tok == eos_token)
+            for _ in range(steps - 1):
+                out, state = self.forward(seqs[-1:].contiguous(), state)
+                lp = torch.log_softmax(out[-1].float(), dim=-1)  # (k, V)
+                if frozen.any():
+                    # frozen beams only extend with eos at zero cost
+                    lp[frozen] = float("-inf")
+                    lp[frozen, eos_token] = 0.0
+                cand = (scores.unsqueeze(1) + lp).reshape(-1)    # (k*V,)
+                scores, idx = cand.topk(k)
+                parent = torch.div(idx, V, rounding_mode="floor")
+                tok = idx % V
+                state = [(h[parent], c[parent]) for h, c in state]
+                seqs = torch.cat([seqs[:, parent], tok.unsqueeze(0)], dim=0)
+                frozen = frozen[parent] | (tok == eos_token)
+            best = int(scores.argmax())
+            seq = torch.cat([prompt, seqs[:, best:best + 1]], dim=0)
+            return seq, float(scores[best])
+        finally:
+            if was_training:
+                self.train()
+
     _DECODE_CAP = 8192  # tokens per replay series (out ring capacity)
 
     def _gemv_cache(self):

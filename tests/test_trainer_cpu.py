@@ -460,3 +460,58 @@ def test_export_history_csv(tmp_path):
     assert "loss" in rows[0] and "eval_loss" in rows[0]
     steps = [int(r["step"]) for r in rows if r["step"]]
     assert steps == sorted(steps)
+
+
+def test_generate_beam_cpu():
+    """Beam search: beam_size=1 == greedy; the returned score is the exact
+    teacher-forced log-probability of the returned continuation; eos
+    freezing keeps frozen beams emitting eos."""
+    from lstm_tensorspark_amd.config import ModelConfig
+    from lstm_tensorspark_amd.models.lstm import LSTMModel
+
+    torch.manual_seed(8)
+    cfg = ModelConfig(input_size=16, hidden_size=16, num_layers=2,
+                      head="lm", vocab_size=24)
+    m = LSTMModel(cfg)
+    prompt = torch.randint(0, 24, (5, 1))
+
+    # beam_size=1 reduces to greedy
+    seq1, s1 = m.generate_beam(prompt, steps=6, beam_size=1)
+    greedy = m.generate(prompt, steps=6, temperature=0.0)
+    assert torch.equal(seq1, greedy)
+
+    # score == sum of stepwise logprobs of the emitted tokens
+    seq4, s4 = m.generate_beam(prompt, steps=6, beam_size=4)
+    assert seq4.shape == (11, 1)
+    logits, _ = m(seq4)
+    lp = torch.log_softmax(logits.float(), dim=-1)
+    total = sum(float(lp[t - 1, 0, int(seq4[t, 0])]) for t in range(5, 11))
+    assert abs(total - s4) < 1e-4, (total, s4)
+    # a wider beam never scores worse than greedy's path
+    assert s4 >= s1 - 1e-6
+
+    # eos freezing: after the first eos in the best beam, only eos follows
+    eos = int(greedy[5, 0])  # force an early finish to be likely
+    seqE, _ = m.generate_beam(prompt, steps=8, beam_size=3, eos_token=eos)
+    tail = seqE[5:, 0].tolist()
+    if eos in tail:
+        i = tail.index(eos)
+        assert all(t == eos for t in tail[i:])
+
+
+def test_serving_beam_endpoint():
+    from fastapi.testclient import TestClient
+    from examples.serve import build_app
+    from lstm_tensorspark_amd.config import ModelConfig
+    from lstm_tensorspark_amd.models.lstm import LSTMModel
+
+    cfg = ModelConfig(input_size=16, hidden_size=16, num_layers=1,
+                      head="lm", vocab_size=32)
+    app = build_app(LSTMModel(cfg), torch.device("cpu"))
+    client = TestClient(app)
+    r = client.post("/generate", json={"tokens": [1, 2], "steps": 4,
+                                       "beam_size": 3})
+    assert r.status_code == 200
+    body = r.json()
+    assert len(body["generated"]) == 4 and "score" in body
+    assert body["score"] <= 0.0
