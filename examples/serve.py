@@ -30,7 +30,7 @@ from lstm_tensorspark_amd.config import Config, ModelConfig
 from lstm_tensorspark_amd.models.lstm import LSTMModel
 
 
-def build_app(model: LSTMModel, device: torch.device):
+def build_app(model: LSTMModel, device: torch.device, batcher=None):
     import time
 
     from fastapi import FastAPI, HTTPException
@@ -91,6 +91,17 @@ def build_app(model: LSTMModel, device: torch.device):
         prompt = torch.tensor(req.tokens, dtype=torch.long,
                               device=device).unsqueeze(1)  # (Tp, B=1)
         score = None
+        if (batcher is not None and req.beam_size <= 1
+                and req.temperature <= 0):
+            # continuous batching: greedy requests share batched decode
+            # steps with every other in-flight request
+            rid = batcher.submit(req.tokens, req.steps)
+            toks = batcher.result(rid, timeout=120.0)
+            if m_reqs:
+                m_reqs.labels(status="ok").inc()
+                m_toks.inc(req.steps)
+                m_lat.observe(time.perf_counter() - t0)
+            return {"tokens": toks, "generated": toks[len(req.tokens):]}
         with torch.no_grad():
             if req.beam_size > 1:
                 out, score = model.generate_beam(prompt, steps=req.steps,
@@ -154,6 +165,10 @@ def main():
                     help="serve a random-init model (stack testing)")
     ap.add_argument("--hidden", type=int, default=1024)
     ap.add_argument("--vocab", type=int, default=50000)
+    ap.add_argument("--continuous", action="store_true",
+                    help="continuous batching: concurrent greedy requests "
+                         "share batched decode steps")
+    ap.add_argument("--max-batch", type=int, default=64)
     ap.add_argument("--host", default="127.0.0.1")
     ap.add_argument("--port", type=int, default=8000)
     args = ap.parse_args()
@@ -161,7 +176,12 @@ def main():
         print("need --ckpt or --demo", file=sys.stderr)
         sys.exit(2)
     model, device = load_model(args)
-    app = build_app(model, device)
+    batcher = None
+    if args.continuous:
+        from lstm_tensorspark_amd.serving import ContinuousBatcher
+        batcher = ContinuousBatcher(model, max_batch=args.max_batch)
+        batcher.start()
+    app = build_app(model, device, batcher=batcher)
     import uvicorn
     uvicorn.run(app, host=args.host, port=args.port, log_level="info")
 
