@@ -43,6 +43,7 @@ def build_app(model: LSTMModel, device: torch.device, batcher=None):
         top_k: int = 0
         top_p: float = 0.0
         beam_size: int = 1
+        eos_token: int = -1
 
     app = FastAPI(title="lstm_tensorspark_amd serving")
 
@@ -95,7 +96,8 @@ def build_app(model: LSTMModel, device: torch.device, batcher=None):
                 and req.temperature <= 0):
             # continuous batching: greedy requests share batched decode
             # steps with every other in-flight request
-            rid = batcher.submit(req.tokens, req.steps)
+            rid = batcher.submit(req.tokens, req.steps,
+                                 eos_token=req.eos_token)
             toks = batcher.result(rid, timeout=120.0)
             if m_reqs:
                 m_reqs.labels(status="ok").inc()
@@ -105,11 +107,13 @@ def build_app(model: LSTMModel, device: torch.device, batcher=None):
         with torch.no_grad():
             if req.beam_size > 1:
                 out, score = model.generate_beam(prompt, steps=req.steps,
-                                                 beam_size=req.beam_size)
+                                                 beam_size=req.beam_size,
+                                                 eos_token=req.eos_token)
             else:
                 out = model.generate(prompt, steps=req.steps,
                                      temperature=req.temperature,
-                                     top_k=req.top_k, top_p=req.top_p)
+                                     top_k=req.top_k, top_p=req.top_p,
+                                     eos_token=req.eos_token)
         if m_reqs:
             m_reqs.labels(status="ok").inc()
             m_toks.inc(req.steps)

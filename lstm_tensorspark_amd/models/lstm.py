@@ -301,7 +301,8 @@ class LSTMModel(nn.Module):
                  generator: "Optional[torch.Generator]" = None,
                  use_graph: Optional[bool] = None,
                  use_gemv: Optional[bool] = None,
-                 top_k: int = 0, top_p: float = 0.0) -> torch.Tensor:
+                 top_k: int = 0, top_p: float = 0.0,
+                 eos_token: int = -1) -> torch.Tensor:
         """Autoregressive decode for the LM head (the serving path).
 
         prompt: (Tp, B) long tokens (time-major).  Returns (Tp+steps, B):
@@ -310,6 +311,9 @@ class LSTMModel(nn.Module):
         is carried across steps, so each decode step is one T=1 pass.
         top_k / top_p restrict sampling to the k most likely tokens /
         the smallest nucleus with mass >= top_p (sampling path only).
+        eos_token >= 0 stops a sequence at that token (a finished column
+        keeps emitting eos); decode ends early once EVERY column is done
+        (eager path only -- the graphed decode runs fixed steps).
 
         use_graph (default: auto = greedy + CUDA + steps >= 16): capture
         ONE decode step as a hipGraph whose body copies its outputs back
@@ -323,9 +327,12 @@ class LSTMModel(nn.Module):
         was_training = self.training
         self.eval()
         if use_graph is None:
-            use_graph = (temperature <= 0 and prompt.is_cuda and steps >= 16)
+            use_graph = (temperature <= 0 and prompt.is_cuda and steps >= 16
+                         and eos_token < 0)
         assert not (use_graph and temperature > 0), \
             "the graphed decode step is greedy-only"
+        assert not (use_graph and eos_token >= 0), \
+            "eos early-stop needs the eager path (use_graph=False)"
         try:
             out, state = self.forward(prompt)          # (Tp, B, V)
             cur = self._sample(out[-1], temperature, generator, top_k, top_p)
@@ -334,12 +341,20 @@ class LSTMModel(nn.Module):
                     [prompt, self._graph_decode(cur, state, steps, use_gemv)],
                     dim=0)
             toks = [prompt]
+            done = (cur == eos_token) if eos_token >= 0 else None
             for _ in range(steps - 1):
                 toks.append(cur.unsqueeze(0))
+                if done is not None and bool(done.all()):
+                    break
                 out, state = self.forward(cur.unsqueeze(0), state)
                 cur = self._sample(out[-1], temperature, generator,
                                    top_k, top_p)
-            toks.append(cur.unsqueeze(0))
+                if done is not None:
+                    cur = torch.where(done, torch.full_like(cur, eos_token),
+                                      cur)
+                    done = done | (cur == eos_token)
+            else:
+                toks.append(cur.unsqueeze(0))
             return torch.cat(toks, dim=0)
         finally:
             if was_training:

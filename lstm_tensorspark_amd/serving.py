@@ -36,13 +36,19 @@ import torch
 
 
 class _Slot:
-    __slots__ = ("rid", "tokens", "remaining", "prompt_len")
+    __slots__ = ("rid", "tokens", "remaining", "prompt_len", "eos")
 
-    def __init__(self, rid: int, tokens: List[int], remaining: int):
+    def __init__(self, rid: int, tokens: List[int], remaining: int,
+                 eos: int = -1):
         self.rid = rid
         self.tokens = list(tokens)
         self.prompt_len = len(tokens)
         self.remaining = remaining
+        self.eos = eos
+
+    def finished(self) -> bool:
+        return (self.remaining <= 0
+                or (self.eos >= 0 and self.tokens[-1] == self.eos))
 
 
 class ContinuousBatcher:
@@ -75,7 +81,10 @@ class ContinuousBatcher:
         self._running = False
 
     # ------------------------------------------------------------ requests
-    def submit(self, tokens: List[int], steps: int) -> int:
+    def submit(self, tokens: List[int], steps: int,
+               eos_token: int = -1) -> int:
+        """eos_token >= 0: the request finishes early when it emits that
+        token (same semantics as model.generate(eos_token=...))."""
         V = self.model.cfg.vocab_size
         if not tokens or any(t < 0 or t >= V for t in tokens):
             raise ValueError(f"tokens must be in [0, {V})")
@@ -83,7 +92,7 @@ class ContinuousBatcher:
             raise ValueError(f"steps must be in [1, {self.max_steps}]")
         with self._cv:
             rid = next(self._ids)
-            self._queue.append(_Slot(rid, tokens, steps))
+            self._queue.append(_Slot(rid, tokens, steps, eos_token))
             self._cv.notify_all()
         return rid
 
@@ -146,7 +155,7 @@ class ContinuousBatcher:
     def _retire_locked(self):
         keep, done = [], []
         for i, s in enumerate(self._slots):
-            (done if s.remaining <= 0 else keep).append(i)
+            (done if s.finished() else keep).append(i)
         if not done:
             return
         for i in done:

@@ -127,3 +127,22 @@ def test_serving_endpoint_with_batcher():
         assert r2.status_code == 200
     finally:
         eng.stop()
+
+
+def test_eos_early_stop_matches_generate():
+    """Per-request eos: the batcher stops a request at its eos token
+    exactly like model.generate(eos_token=...), and neighbors continue."""
+    m = _model(seed=17)
+    eng = ContinuousBatcher(m)
+    # pick an eos that the greedy path actually emits: the 2nd generated
+    # token of this prompt
+    probe = _solo(m, [1, 2], 6)
+    eos = probe[3]
+    prompt = torch.tensor([1, 2], dtype=torch.long).unsqueeze(1)
+    solo = m.generate(prompt, steps=6, eos_token=eos)[:, 0].tolist()
+    assert solo[-1] == eos and len(solo) < 8   # it really stopped early
+    ra = eng.submit([1, 2], 6, eos_token=eos)
+    rb = eng.submit([5, 6], 5)                 # no eos, full length
+    eng.drain()
+    assert eng.result(ra, timeout=0) == solo
+    assert eng.result(rb, timeout=0) == _solo(m, [5, 6], 5)
