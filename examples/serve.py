@@ -44,6 +44,7 @@ def build_app(model: LSTMModel, device: torch.device, batcher=None):
         top_p: float = 0.0
         beam_size: int = 1
         eos_token: int = -1
+        seed: int | None = None
 
     app = FastAPI(title="lstm_tensorspark_amd serving")
 
@@ -92,12 +93,14 @@ def build_app(model: LSTMModel, device: torch.device, batcher=None):
         prompt = torch.tensor(req.tokens, dtype=torch.long,
                               device=device).unsqueeze(1)  # (Tp, B=1)
         score = None
-        if (batcher is not None and req.beam_size <= 1
-                and req.temperature <= 0):
-            # continuous batching: greedy requests share batched decode
-            # steps with every other in-flight request
+        if batcher is not None and req.beam_size <= 1:
+            # continuous batching: greedy AND sampled requests share
+            # batched decode steps (sampled ones use their own generator)
             rid = batcher.submit(req.tokens, req.steps,
-                                 eos_token=req.eos_token)
+                                 eos_token=req.eos_token,
+                                 temperature=req.temperature,
+                                 top_k=req.top_k, top_p=req.top_p,
+                                 seed=req.seed)
             toks = batcher.result(rid, timeout=120.0)
             if m_reqs:
                 m_reqs.labels(status="ok").inc()

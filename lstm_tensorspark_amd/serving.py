@@ -36,15 +36,21 @@ import torch
 
 
 class _Slot:
-    __slots__ = ("rid", "tokens", "remaining", "prompt_len", "eos")
+    __slots__ = ("rid", "tokens", "remaining", "prompt_len", "eos",
+                 "temperature", "top_k", "top_p", "gen")
 
     def __init__(self, rid: int, tokens: List[int], remaining: int,
-                 eos: int = -1):
+                 eos: int = -1, temperature: float = 0.0, top_k: int = 0,
+                 top_p: float = 0.0, gen=None):
         self.rid = rid
         self.tokens = list(tokens)
         self.prompt_len = len(tokens)
         self.remaining = remaining
         self.eos = eos
+        self.temperature = temperature
+        self.top_k = top_k
+        self.top_p = top_p
+        self.gen = gen
 
     def finished(self) -> bool:
         return (self.remaining <= 0
@@ -82,17 +88,29 @@ class ContinuousBatcher:
 
     # ------------------------------------------------------------ requests
     def submit(self, tokens: List[int], steps: int,
-               eos_token: int = -1) -> int:
+               eos_token: int = -1, temperature: float = 0.0,
+               top_k: int = 0, top_p: float = 0.0,
+               seed: Optional[int] = None) -> int:
         """eos_token >= 0: the request finishes early when it emits that
-        token (same semantics as model.generate(eos_token=...))."""
+        token (same semantics as model.generate(eos_token=...)).
+
+        temperature > 0 samples with this request's OWN generator (seeded
+        with `seed` when given), so a sampled request is reproducible and
+        token-identical to a solo model.generate() with the same seed --
+        regardless of what else shares the batch."""
         V = self.model.cfg.vocab_size
         if not tokens or any(t < 0 or t >= V for t in tokens):
             raise ValueError(f"tokens must be in [0, {V})")
         if steps < 1 or steps > self.max_steps:
             raise ValueError(f"steps must be in [1, {self.max_steps}]")
+        gen = None
+        if temperature > 0 and seed is not None:
+            gen = torch.Generator(device=self.device)
+            gen.manual_seed(seed)
         with self._cv:
             rid = next(self._ids)
-            self._queue.append(_Slot(rid, tokens, steps, eos_token))
+            self._queue.append(_Slot(rid, tokens, steps, eos_token,
+                                     temperature, top_k, top_p, gen))
             self._cv.notify_all()
         return rid
 
@@ -125,7 +143,7 @@ class ContinuousBatcher:
             prompt = torch.tensor(slot.tokens, dtype=torch.long,
                                   device=self.device).unsqueeze(1)
             out, st = m.forward(prompt)              # (Tp, 1, V)
-            tok = int(out[-1, 0].argmax())
+            tok = int(self._pick(out[-1], slot)[0])
             slot.tokens.append(tok)
             slot.remaining -= 1
             if self._state is None or not self._slots:
@@ -146,11 +164,21 @@ class ContinuousBatcher:
         x = torch.tensor([[s.tokens[-1] for s in self._slots]],
                          dtype=torch.long, device=self.device)  # (1, B)
         out, self._state = m.forward(x, self._state)            # (1, B, V)
-        toks = out[0].argmax(dim=-1)                            # (B,)
+        logits = out[0]                                         # (B, V)
+        greedy = logits.argmax(dim=-1)                          # (B,)
         for i, s in enumerate(self._slots):
-            s.tokens.append(int(toks[i]))
+            if s.temperature > 0:
+                s.tokens.append(int(self._pick(logits[i:i + 1], s)[0]))
+            else:
+                s.tokens.append(int(greedy[i]))
             s.remaining -= 1
         self._retire_locked()
+
+    def _pick(self, logits, slot: _Slot):
+        """Next token for one slot from its (1, V) logits row -- the same
+        _sample used by model.generate, fed the slot's own generator."""
+        return self.model._sample(logits, slot.temperature, slot.gen,
+                                  slot.top_k, slot.top_p)
 
     def _retire_locked(self):
         keep, done = [], []

@@ -146,3 +146,20 @@ def test_eos_early_stop_matches_generate():
     eng.drain()
     assert eng.result(ra, timeout=0) == solo
     assert eng.result(rb, timeout=0) == _solo(m, [5, 6], 5)
+
+
+def test_sampled_requests_match_solo_seeded():
+    """A temperature request in the batch samples with its own seeded
+    generator and reproduces the solo seeded decode exactly -- even while
+    sharing steps with greedy neighbors."""
+    m = _model(seed=18)
+    eng = ContinuousBatcher(m)
+    prompt = torch.tensor([2, 3], dtype=torch.long).unsqueeze(1)
+    g = torch.Generator().manual_seed(77)
+    solo = m.generate(prompt, steps=6, temperature=0.8, top_k=8,
+                      generator=g)[:, 0].tolist()
+    ra = eng.submit([2, 3], 6, temperature=0.8, top_k=8, seed=77)
+    rb = eng.submit([4], 4)                      # greedy neighbor
+    eng.drain()
+    assert eng.result(ra, timeout=0) == solo
+    assert eng.result(rb, timeout=0) == _solo(m, [4], 4)
