@@ -73,6 +73,44 @@ def build_app(model: LSTMModel, device: torch.device, batcher=None):
     except ImportError:  # pragma: no cover
         m_reqs = m_toks = m_lat = None
 
+    @app.post("/generate_stream")
+    def generate_stream(req: GenRequest):
+        """Server-sent events: tokens stream out as the continuous
+        batcher produces them (requires --continuous; beam not streamed).
+        Each event is a JSON list of new tokens; the final event is
+        [DONE]."""
+        import json as _json
+        from starlette.responses import StreamingResponse
+        if batcher is None:
+            raise HTTPException(400, "streaming requires --continuous")
+        if req.beam_size > 1:
+            raise HTTPException(400, "beam search is not streamable")
+        V = model.cfg.vocab_size
+        if not req.tokens or any(t < 0 or t >= V for t in req.tokens):
+            raise HTTPException(400, f"tokens must be in [0, {V})")
+        if req.steps < 1 or req.steps > 4096:
+            raise HTTPException(400, "steps must be in [1, 4096]")
+        rid = batcher.submit(req.tokens, req.steps,
+                             eos_token=req.eos_token,
+                             temperature=req.temperature,
+                             top_k=req.top_k, top_p=req.top_p,
+                             seed=req.seed)
+
+        def events():
+            sent = len(req.tokens)
+            while True:
+                toks, done = batcher.peek(rid)
+                if len(toks) > sent:
+                    yield f"data: {_json.dumps(toks[sent:])}\n\n"
+                    sent = len(toks)
+                if done:
+                    batcher.try_result(rid)
+                    yield "data: [DONE]\n\n"
+                    return
+                time.sleep(0.003)
+
+        return StreamingResponse(events(), media_type="text/event-stream")
+
     @app.get("/health")
     def health():
         return {"ok": True, "device": str(device),

@@ -114,6 +114,20 @@ class ContinuousBatcher:
             self._cv.notify_all()
         return rid
 
+    def peek(self, rid: int):
+        """(tokens_so_far, done) without claiming the result -- the
+        polling primitive for streaming responses."""
+        with self._cv:
+            if rid in self._results:
+                return list(self._results[rid]), True
+            for s in self._slots:
+                if s.rid == rid:
+                    return list(s.tokens), False
+            for s in self._queue:
+                if s.rid == rid:
+                    return list(s.tokens), False
+        raise KeyError(f"unknown request id {rid}")
+
     def try_result(self, rid: int) -> Optional[List[int]]:
         with self._cv:
             return self._results.pop(rid, None)

@@ -163,3 +163,34 @@ def test_sampled_requests_match_solo_seeded():
     eng.drain()
     assert eng.result(ra, timeout=0) == solo
     assert eng.result(rb, timeout=0) == _solo(m, [4], 4)
+
+
+def test_streaming_endpoint_tokens_match_solo():
+    """SSE /generate_stream: concatenated streamed tokens == the solo
+    greedy continuation, with a [DONE] terminator."""
+    import json
+    from fastapi.testclient import TestClient
+    from examples.serve import build_app
+
+    m = _model(seed=19)
+    eng = ContinuousBatcher(m)
+    eng.start()
+    try:
+        app = build_app(m, torch.device("cpu"), batcher=eng)
+        client = TestClient(app)
+        got, done = [], False
+        with client.stream("POST", "/generate_stream",
+                           json={"tokens": [3, 1], "steps": 6}) as r:
+            assert r.status_code == 200
+            for line in r.iter_lines():
+                if not line.startswith("data: "):
+                    continue
+                payload = line[len("data: "):]
+                if payload == "[DONE]":
+                    done = True
+                    break
+                got.extend(json.loads(payload))
+        assert done
+        assert got == _solo(m, [3, 1], 6)[2:]
+    finally:
+        eng.stop()
