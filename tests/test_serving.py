@@ -194,3 +194,42 @@ def test_streaming_endpoint_tokens_match_solo():
         assert got == _solo(m, [3, 1], 6)[2:]
     finally:
         eng.stop()
+
+
+def test_batcher_random_schedules_property():
+    """Randomized schedules (hypothesis): arbitrary prompts, lengths and
+    join times -- every request must still match its solo greedy decode."""
+    from hypothesis import given, settings
+    from hypothesis import strategies as st
+
+    m = _model(seed=20)
+
+    @st.composite
+    def schedule(draw):
+        n = draw(st.integers(1, 6))
+        reqs = []
+        for _ in range(n):
+            plen = draw(st.integers(1, 6))
+            prompt = [draw(st.integers(0, 47)) for _ in range(plen)]
+            steps = draw(st.integers(1, 8))
+            join_after = draw(st.integers(0, 5))   # engine steps before submit
+            reqs.append((prompt, steps, join_after))
+        return reqs
+
+    @given(schedule())
+    @settings(max_examples=20, deadline=None)
+    def run(reqs):
+        eng = ContinuousBatcher(m, max_batch=3)
+        pending = sorted(enumerate(reqs), key=lambda kv: kv[1][2])
+        rids = {}
+        t = 0
+        while pending or eng.stats()["active"] or eng.stats()["queued"]:
+            while pending and pending[0][1][2] <= t:
+                i, (prompt, steps, _) = pending.pop(0)
+                rids[i] = eng.submit(prompt, steps)
+            eng.step()
+            t += 1
+        for i, (prompt, steps, _) in enumerate(reqs):
+            assert eng.result(rids[i], timeout=0) == _solo(m, prompt, steps)
+
+    run()
