@@ -233,3 +233,14 @@ def test_batcher_random_schedules_property():
             assert eng.result(rids[i], timeout=0) == _solo(m, prompt, steps)
 
     run()
+
+
+def test_batcher_bench_tool_degenerate():
+    """tools/batcher_bench.py runs on CPU (one GPU lease away from the r3
+    multi-tenant throughput measurement)."""
+    from tools.batcher_bench import main
+    r = main(["--hidden", "32", "--vocab", "64", "--layers", "1",
+              "--clients", "3", "--steps", "3", "--prompt", "2",
+              "--requests", "6", "--solo-baseline", "1"])
+    assert r["requests"] == 6 and r["req_per_s"] > 0
+    assert r["p50_ms"] > 0 and r["solo_ms"] > 0
