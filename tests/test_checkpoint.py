@@ -170,3 +170,27 @@ def test_ckpt_to_safetensors_tool(tmp_path):
     assert model.cfg.vocab_size == 32
     for k, v in t.model.weight_dict().items():
         assert torch.equal(v, model.weight_dict()[k]), k
+
+
+def test_load_checkpoint_from_older_config_schema(tmp_path):
+    """A checkpoint saved before accum/schedule/eval/data-file fields
+    existed must still load (dataclass defaults fill absent keys)."""
+    cfg = small_cfg(tmp_path)
+    t = Trainer(cfg)
+    t.fit(2)
+    path = os.path.join(str(tmp_path), "old.pt")
+    save_checkpoint(path, t.model, t.optimizer, step=2, config=cfg.to_dict())
+    payload = torch.load(path, map_location="cpu", weights_only=False)
+    for k in ("accum_steps", "lr_schedule", "warmup_steps", "min_lr",
+              "eval_interval", "eval_steps"):
+        payload["config"]["train"].pop(k, None)
+    for k in ("path", "file_dtype", "eval_frac"):
+        payload["config"]["data"].pop(k, None)
+    torch.save(payload, path)
+    t2 = Trainer(small_cfg(tmp_path))
+    t2.load(path)
+    assert t2.step == 2
+    from lstm_tensorspark_amd.config import Config
+    back = Config.from_dict(torch.load(path, map_location="cpu",
+                                       weights_only=False)["config"])
+    assert back.train.accum_steps == 1 and back.data.eval_frac == 0.0
