@@ -128,8 +128,6 @@ def build_app(model: LSTMModel, device: torch.device, batcher=None):
             if m_reqs:
                 m_reqs.labels(status="bad_request").inc()
             raise HTTPException(400, "steps must be in [1, 4096]")
-        prompt = torch.tensor(req.tokens, dtype=torch.long,
-                              device=device).unsqueeze(1)  # (Tp, B=1)
         score = None
         if batcher is not None and req.beam_size <= 1:
             # continuous batching: greedy AND sampled requests share
@@ -142,9 +140,11 @@ def build_app(model: LSTMModel, device: torch.device, batcher=None):
             toks = batcher.result(rid, timeout=120.0)
             if m_reqs:
                 m_reqs.labels(status="ok").inc()
-                m_toks.inc(req.steps)
+                m_toks.inc(len(toks) - len(req.tokens))  # EOS may end early
                 m_lat.observe(time.perf_counter() - t0)
             return {"tokens": toks, "generated": toks[len(req.tokens):]}
+        prompt = torch.tensor(req.tokens, dtype=torch.long,
+                              device=device).unsqueeze(1)  # (Tp, B=1)
         with torch.no_grad():
             if req.beam_size > 1:
                 out, score = model.generate_beam(prompt, steps=req.steps,
@@ -157,7 +157,7 @@ def build_app(model: LSTMModel, device: torch.device, batcher=None):
                                      eos_token=req.eos_token)
         if m_reqs:
             m_reqs.labels(status="ok").inc()
-            m_toks.inc(req.steps)
+            m_toks.inc(out.shape[0] - len(req.tokens))
             m_lat.observe(time.perf_counter() - t0)
         resp = {"tokens": out[:, 0].tolist(),
                 "generated": out[len(req.tokens):, 0].tolist()}
@@ -211,8 +211,8 @@ def main():
     ap.add_argument("--hidden", type=int, default=1024)
     ap.add_argument("--vocab", type=int, default=50000)
     ap.add_argument("--continuous", action="store_true",
-                    help="continuous batching: concurrent greedy requests "
-                         "share batched decode steps")
+                    help="continuous batching: concurrent requests (greedy "
+                         "and sampled) share batched decode steps")
     ap.add_argument("--max-batch", type=int, default=64)
     ap.add_argument("--host", default="127.0.0.1")
     ap.add_argument("--port", type=int, default=8000)

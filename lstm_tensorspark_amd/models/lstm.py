@@ -385,7 +385,14 @@ class LSTMModel(nn.Module):
             out, state = self.forward(prompt)               # (Tp, 1, V)
             logp = torch.log_softmax(out[-1, 0].float(), dim=-1)
             k = min(beam_size, V)
-            scores, tok = logp.topk(k)                      # (k,)
+            if k == 1:
+                # argmax guarantees first-index tie-breaking (topk does
+                # not), making beam_size=1 identical to greedy decode by
+                # construction -- including on bf16 logit ties
+                tok = logp.argmax().unsqueeze(0)
+                scores = logp[tok]
+            else:
+                scores, tok = logp.topk(k)                  # (k,)
             state = [(h.expand(k, -1).contiguous(),
                       c.expand(k, -1).contiguous()) for h, c in state]
             seqs = tok.unsqueeze(0)                         # (t, k)
@@ -398,7 +405,11 @@ class LSTMModel(nn.Module):
                     lp[frozen] = float("-inf")
                     lp[frozen, eos_token] = 0.0
                 cand = (scores.unsqueeze(1) + lp).reshape(-1)    # (k*V,)
-                scores, idx = cand.topk(k)
+                if k == 1:
+                    idx = cand.argmax().unsqueeze(0)
+                    scores = cand[idx]
+                else:
+                    scores, idx = cand.topk(k)
                 parent = torch.div(idx, V, rounding_mode="floor")
                 tok = idx % V
                 state = [(h[parent], c[parent]) for h, c in state]
