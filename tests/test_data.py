@@ -167,7 +167,8 @@ def test_lm_file_ranks_disjoint_streams(tmp_path):
 def test_lm_file_text_byte_level(tmp_path):
     from lstm_tensorspark_amd.data import SyntheticSeqData
     p = str(tmp_path / "corpus.txt")
-    open(p, "w").write("the quick brown fox jumps over the lazy dog " * 40)
+    with open(p, "w") as f:
+        f.write("the quick brown fox jumps over the lazy dog " * 40)
     d = SyntheticSeqData(_file_cfg(p, vocab=256))
     x, y = d.batch(0)
     assert x.max() < 256 and torch.equal(x[1:], y[:-1])
