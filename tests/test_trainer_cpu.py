@@ -175,7 +175,8 @@ def test_rccl_sweep_tool_degenerate(tmp_path):
     out = str(tmp_path / "sweep.jsonl")
     main(["--buckets", "4", "--batch", "2", "--steps", "1",
           "--hidden", "32", "--seq-len", "8", "--out", out])
-    rows = [json.loads(l) for l in open(out)]
+    with open(out) as f:
+        rows = [json.loads(l) for l in f]
     assert any(r["kind"] == "step" for r in rows)
 
 
@@ -455,7 +456,8 @@ def test_export_history_csv(tmp_path):
     p = str(tmp_path / "hist.csv")
     t.export_history(p)
     import csv
-    rows = list(csv.DictReader(open(p)))
+    with open(p) as f:
+        rows = list(csv.DictReader(f))
     assert len(rows) == len(t.history) >= 4
     assert "loss" in rows[0] and "eval_loss" in rows[0]
     steps = [int(r["step"]) for r in rows if r["step"]]
