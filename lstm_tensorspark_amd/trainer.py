@@ -58,6 +58,14 @@ class Trainer:
         if self.dtype != torch.float32:
             self.model = self.model.to(self.dtype)
         broadcast_parameters(self.model)  # belt-and-braces weight sync at init
+        if self.ctx.world_size > 1 and self.ctx.rank > 0:
+            # init above must be seed-identical on every rank, but the
+            # ONGOING rng stream (dropout masks) should not be: identical
+            # masks across ranks correlate the regularization noise.
+            # Diverge rank>0 only -- rank 0 and world-1 trajectories stay
+            # bitwise identical to the single-process run.
+            from .data import mix64
+            torch.manual_seed(mix64(cfg.train.seed, 3, self.ctx.rank))
 
         self.optimizer = build_optimizer(cfg.train, self.model.parameters())
         self.bucketer = None

@@ -257,3 +257,32 @@ def test_4rank_grads_equal_average_of_per_rank_grads(tmp_path):
         want = sum(pr[name] for pr in per_rank) / 4
         err = (got[name] - want).abs().max()
         assert err < 1e-6, f"{name}: {err}"
+
+
+# ------------------------------------------------------- dropout streams
+def _dropout_stream_worker(ctx, tmpdir):
+    cfg = small_cfg()
+    cfg.model.dropout = 0.5
+    cfg.model.num_layers = 2
+    t = Trainer(cfg, dist_ctx=ctx)
+    # sample the post-init rng stream (what dropout masks draw from)
+    sample = torch.rand(16)
+    t.fit(2)
+    w = {k: v.detach().clone() for k, v in t.model.weight_dict().items()}
+    torch.save({"sample": sample, "w": w},
+               os.path.join(tmpdir, f"ds_rank{ctx.rank}.pt"))
+
+
+def test_2rank_dropout_masks_decorrelated_weights_synced(tmp_path):
+    """Ranks must draw DIFFERENT dropout masks (identical masks correlate
+    the regularization noise) while the all-reduce still keeps weights
+    bitwise synchronized."""
+    run_spawned(_dropout_stream_worker, world_size=2, backend="gloo",
+                args=(str(tmp_path),), master_port=29561)
+    a = torch.load(os.path.join(str(tmp_path), "ds_rank0.pt"),
+                   weights_only=False)
+    b = torch.load(os.path.join(str(tmp_path), "ds_rank1.pt"),
+                   weights_only=False)
+    assert not torch.equal(a["sample"], b["sample"]), "rng streams collide"
+    for k in a["w"]:
+        assert torch.equal(a["w"][k], b["w"][k]), f"desync: {k}"
