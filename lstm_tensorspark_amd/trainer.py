@@ -425,6 +425,13 @@ class Trainer:
     def load(self, path: str):
         payload = load_checkpoint(path, self.model, self.optimizer)
         self.step = payload["step"]
+        if self.ctx.world_size > 1 and self.ctx.rank > 0:
+            # the checkpoint carries rank 0's rng state; re-diverge this
+            # rank's mask stream (mixing the step keeps the post-resume
+            # stream from replaying the pre-save one)
+            from .data import mix64
+            torch.manual_seed(mix64(self.cfg.train.seed, 3, self.ctx.rank,
+                                    self.step))
         self.model._weights_version = getattr(
             self.model, "_weights_version", 0) + 1
         self.log.info(f"checkpoint loaded: {path} (step {self.step})")
