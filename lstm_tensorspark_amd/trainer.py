@@ -391,24 +391,34 @@ class Trainer:
         counterpart of the reference's Session.run(logits) path, SURVEY.md
         §3.3).  Uses the structurally disjoint eval RNG stream (data.mix64
         stream=1) so eval batches never coincide with training batches.
-        All-reduces the mean loss across ranks."""
+        All-reduces the mean loss (and accuracy, for logit heads)
+        across ranks."""
         self.model.eval()
         src = data if data is not None else self.data
-        total, n = 0.0, 0
-        for i in range(steps):
-            x, y = src.batch(i, stream=1)
-            out, _ = self.model(x)
-            total += float(self.model.loss(out, y).detach())
-            n += 1
+        logit_head = self.cfg.model.head in ("classifier", "seq_classifier",
+                                             "lm")
+        total, acc, n = 0.0, 0.0, 0
+        with torch.no_grad():
+            for i in range(steps):
+                x, y = src.batch(i, stream=1)
+                out, _ = self.model(x)
+                total += float(self.model.loss(out, y))
+                if logit_head:
+                    acc += float((out.argmax(-1) == y).float().mean())
+                n += 1
         self.model.train()
         mean = total / max(n, 1)
+        mean_acc = acc / max(n, 1)
         if dist.is_initialized() and self.ctx.world_size > 1:
-            t = torch.tensor([mean], dtype=torch.float64)
+            t = torch.tensor([mean, mean_acc], dtype=torch.float64)
             if self.ctx.backend == "nccl":
                 t = t.to(self.device)
             dist.all_reduce(t, op=dist.ReduceOp.AVG)
-            mean = float(t.item())
-        return {"eval_loss": mean, "steps": n}
+            mean, mean_acc = float(t[0].item()), float(t[1].item())
+        r = {"eval_loss": mean, "steps": n}
+        if logit_head:
+            r["eval_acc"] = mean_acc
+        return r
 
     # ------------------------------------------------------------------
     def save(self, path: str):

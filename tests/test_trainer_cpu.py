@@ -565,3 +565,15 @@ def test_dropout_training_reproducible():
     tB.fit(3)
     for k, vA in tA.model.weight_dict().items():
         assert torch.equal(vA, tB.model.weight_dict()[k]), k
+
+
+def test_evaluate_reports_accuracy():
+    """Logit heads get eval_acc in [0, 1]; training the copy task drives
+    it up."""
+    cfg = copy_cfg(steps=60)
+    t = Trainer(cfg)
+    a0 = t.evaluate(steps=3)["eval_acc"]
+    t.fit(60)
+    a1 = t.evaluate(steps=3)["eval_acc"]
+    assert 0.0 <= a0 <= 1.0 and 0.0 <= a1 <= 1.0
+    assert a1 > a0
