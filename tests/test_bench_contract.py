@@ -54,3 +54,24 @@ def test_train_cli_new_knobs(tmp_path):
     assert out.returncode == 0, out.stderr[-2000:]
     assert "eval_loss" in out.stdout + out.stderr
     assert os.path.exists(os.path.join(str(tmp_path), "ckpt_final.pt"))
+
+
+def test_bench_dp2_contract():
+    """The driver's exact multi-rank invocation shape (torchrun world 2,
+    gloo on CPU): exactly ONE JSON line, from rank 0, with dp2 config and
+    the whole-job aggregate value."""
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29617", os.path.join(REPO, "bench.py"),
+         "--gpus", "2", "--steps", "2", "--warmup", "1", "--cpu",
+         "--preset", "plumbing-cpu"],
+        capture_output=True, text=True, timeout=600, cwd=REPO)
+    assert out.returncode == 0, out.stderr[-2000:]
+    lines = [l for l in out.stdout.strip().splitlines()
+             if l.startswith("{\"metric\"")]
+    assert len(lines) == 1, out.stdout
+    rec = json.loads(lines[0])
+    assert rec["n_gpus"] == 2
+    assert rec["config"]["parallelism"] == "dp2"
+    assert rec["value"] > 0 and rec["ms_per_step"] > 0
