@@ -185,7 +185,9 @@ def test_gpu_generate_beam_and_filters():
     greedy = m.generate(prompt, steps=8, use_graph=False)
     assert torch.equal(seq1, greedy)
     seq4, s4 = m.generate_beam(prompt, steps=8, beam_size=4)
-    assert seq4.shape == (14, 1) and s4 >= s1 - 1e-3
+    # NOTE: no s4 >= s1 assertion -- beam search may prune the greedy
+    # path, so that inequality is not guaranteed in general
+    assert seq4.shape == (14, 1) and s4 == s4
     assert (seq4 >= 0).all() and (seq4 < 512).all()
     seq4b, s4b = m.generate_beam(prompt, steps=8, beam_size=4)
     assert torch.equal(seq4, seq4b) and s4 == s4b
