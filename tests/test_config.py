@@ -107,3 +107,16 @@ def test_fp8_head_field_roundtrip(tmp_path):
     p.write_text(yaml.safe_dump(d))
     c3 = Config.from_dict(yaml.safe_load(p.read_text()))
     assert c3.train.fp8_head is True
+
+
+def test_example_yamls_load():
+    """Every checked-in example YAML must parse into a valid Config."""
+    import glob
+    import os
+    from lstm_tensorspark_amd.config import Config
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    paths = sorted(glob.glob(os.path.join(repo, "examples", "*.yaml")))
+    assert len(paths) >= 3
+    for p in paths:
+        cfg = Config.from_yaml(p)
+        assert cfg.model.hidden_size > 0, p
