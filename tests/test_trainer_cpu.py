@@ -577,3 +577,12 @@ def test_evaluate_reports_accuracy():
     a1 = t.evaluate(steps=3)["eval_acc"]
     assert 0.0 <= a0 <= 1.0 and 0.0 <= a1 <= 1.0
     assert a1 > a0
+
+
+def test_unknown_lr_schedule_raises():
+    cfg = copy_cfg(steps=2)
+    cfg.train.lr_schedule = "exponential"   # not implemented
+    t = Trainer(cfg)
+    import pytest
+    with pytest.raises(ValueError, match="lr_schedule"):
+        t.train_step(t.data.batch(0))
