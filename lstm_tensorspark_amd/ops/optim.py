@@ -30,6 +30,15 @@ class MasterOptimizer:
         self.masters = [p.detach().float().clone() for p in self.params]
         self.step_count = 0
 
+    def sync_masters(self):
+        """Re-derive the fp32 masters from the CURRENT parameter values
+        (required after loading weights from outside the optimizer, e.g.
+        a safetensors fine-tune init -- otherwise the first step snaps
+        the params back to the stale masters)."""
+        with torch.no_grad():
+            for m, p in zip(self.masters, self.params):
+                m.copy_(p.detach().float())
+
     # -- grad clip (K6) ----------------------------------------------------
     def global_grad_norm(self) -> torch.Tensor:
         grads = [p.grad for p in self.params if p.grad is not None]

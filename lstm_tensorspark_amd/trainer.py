@@ -432,6 +432,19 @@ class Trainer:
         if dist.is_initialized():
             dist.barrier()
 
+    def load_weights(self, path: str) -> dict:
+        """Weights-only init from a safetensors artifact (fine-tune /
+        continued-pretraining entry: fresh optimizer and step counter,
+        pretrained weights).  Returns the file's metadata."""
+        from .checkpoint import load_weights_safetensors
+        meta = load_weights_safetensors(path, self.model)
+        broadcast_parameters(self.model)   # keep ranks bitwise identical
+        self.optimizer.sync_masters()
+        self.model._weights_version = getattr(
+            self.model, "_weights_version", 0) + 1
+        self.log.info(f"weights loaded: {path}")
+        return meta
+
     def load(self, path: str):
         payload = load_checkpoint(path, self.model, self.optimizer)
         self.step = payload["step"]

@@ -194,3 +194,31 @@ def test_load_checkpoint_from_older_config_schema(tmp_path):
     back = Config.from_dict(torch.load(path, map_location="cpu",
                                        weights_only=False)["config"])
     assert back.train.accum_steps == 1 and back.data.eval_frac == 0.0
+
+
+def test_trainer_load_weights_finetune(tmp_path):
+    """Fine-tune entry: load a safetensors artifact into a fresh Trainer --
+    weights land, fp32 masters re-sync (first step must move FROM the
+    loaded weights, not snap back to stale masters), step stays 0."""
+    from lstm_tensorspark_amd.checkpoint import export_weights_safetensors
+
+    cfgA = small_cfg(tmp_path)
+    tA = Trainer(cfgA)
+    tA.fit(3)
+    path = os.path.join(str(tmp_path), "w.safetensors")
+    export_weights_safetensors(path, tA.model, metadata={"step": 3})
+
+    tB = Trainer(small_cfg(tmp_path))
+    meta = tB.load_weights(path)
+    assert meta["step"] == "3" and tB.step == 0
+    for k, v in tA.model.weight_dict().items():
+        assert torch.equal(v, tB.model.weight_dict()[k]), k
+    # masters were re-synced: one step from identical grads moves both
+    # trainers identically
+    wd0 = {k: v.detach().clone() for k, v in tB.model.weight_dict().items()}
+    tB.train_step(tB.data.batch(0))
+    moved = any(not torch.equal(wd0[k], v)
+                for k, v in tB.model.weight_dict().items())
+    assert moved
+    for m, p in zip(tB.optimizer.masters, tB.optimizer.params):
+        assert torch.allclose(m, p.detach().float(), atol=1e-6)
