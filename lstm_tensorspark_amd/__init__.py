@@ -12,3 +12,8 @@ __version__ = "0.1.0"
 from .config import Config, get_preset, PRESETS  # noqa: F401
 from .trainer import Trainer  # noqa: F401
 from .models.lstm import LSTMModel  # noqa: F401
+from .checkpoint import (  # noqa: F401
+    save_checkpoint, load_checkpoint,
+    export_weights_safetensors, load_weights_safetensors,
+)
+from .serving import ContinuousBatcher  # noqa: F401
