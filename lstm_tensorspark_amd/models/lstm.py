@@ -492,6 +492,13 @@ class LSTMModel(nn.Module):
         dev = cur.device
         if use_gemv is None:
             use_gemv = self._gemv_eligible(B)
+        if use_gemv:
+            # MUST run before the cache check: if the weights version
+            # moved, this rebuilds the serving-weight tensors AND drops
+            # the captured graph that holds the old tensors' addresses --
+            # otherwise a cached gemv graph would replay STALE weights
+            # after train -> generate -> train -> generate
+            self._gemv_cache()
         cache = getattr(self, "_dec_cache", None)
         if cache is None or cache["B"] != B or cache["gemv"] != use_gemv:
             g_cur = cur.clone()

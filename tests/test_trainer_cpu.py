@@ -586,3 +586,27 @@ def test_unknown_lr_schedule_raises():
     import pytest
     with pytest.raises(ValueError, match="lr_schedule"):
         t.train_step(t.data.batch(0))
+
+
+def test_gemv_cache_rebuild_drops_decode_graph():
+    """The serving-weight cache rebuild (on weights-version change) must
+    drop a captured gemv decode graph -- the graph holds raw addresses of
+    the OLD cache tensors and would otherwise replay stale weights."""
+    from lstm_tensorspark_amd.config import ModelConfig
+    from lstm_tensorspark_amd.models.lstm import LSTMModel
+
+    m = LSTMModel(ModelConfig(input_size=8, hidden_size=8, num_layers=1,
+                              head="lm", vocab_size=16))
+    m._weights_version = 1
+    m._gemv_cache()
+    m._dec_cache = {"B": 1, "gemv": True}      # simulate a captured graph
+    m._gemv_cache()                             # same version: cache kept
+    assert m._dec_cache is not None
+    m._weights_version = 2                      # optimizer stepped
+    m._gemv_cache()
+    assert m._dec_cache is None                 # graph dropped
+    # non-gemv graphs read parameters in place -> never stale, never drop
+    m._dec_cache = {"B": 1, "gemv": False}
+    m._weights_version = 3
+    m._gemv_cache()
+    assert m._dec_cache is not None
