@@ -85,6 +85,8 @@ class ContinuousBatcher:
         self._cv = threading.Condition()
         self._thread: Optional[threading.Thread] = None
         self._running = False
+        self._decode_steps = 0      # batched decode steps executed
+        self._slot_steps = 0        # sum of active slots over those steps
 
     # ------------------------------------------------------------ requests
     def submit(self, tokens: List[int], steps: int,
@@ -144,8 +146,12 @@ class ContinuousBatcher:
 
     def stats(self) -> dict:
         with self._cv:
+            occ = (self._slot_steps / self._decode_steps
+                   if self._decode_steps else 0.0)
             return {"active": len(self._slots), "queued": len(self._queue),
-                    "done_unclaimed": len(self._results)}
+                    "done_unclaimed": len(self._results),
+                    "decode_steps": self._decode_steps,
+                    "mean_occupancy": round(occ, 2)}
 
     # ------------------------------------------------------------- engine
     @torch.no_grad()
@@ -174,6 +180,8 @@ class ContinuousBatcher:
         """One batched decode step for every live slot."""
         if not self._slots:
             return
+        self._decode_steps += 1
+        self._slot_steps += len(self._slots)
         m = self.model
         x = torch.tensor([[s.tokens[-1] for s in self._slots]],
                          dtype=torch.long, device=self.device)  # (1, B)

@@ -257,4 +257,5 @@ def test_gpu_continuous_batcher():
         assert len(out) == len(toks) + steps
         assert out[:len(toks)] == toks
         assert all(0 <= t < 256 for t in out)
-    assert eng.stats() == {"active": 0, "queued": 0, "done_unclaimed": 0}
+    st = eng.stats()
+    assert (st["active"], st["queued"], st["done_unclaimed"]) == (0, 0, 0)

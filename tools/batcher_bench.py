@@ -91,6 +91,7 @@ def main(argv=None):
     for _ in threads:
         done.acquire()
     wall = time.perf_counter() - t0
+    occupancy = eng.stats()["mean_occupancy"]
     eng.stop()
     lat.sort()
     n = len(lat)
@@ -102,6 +103,7 @@ def main(argv=None):
         "tok_per_s": round(n * args.steps / wall, 1),
         "p50_ms": round(statistics.median(lat), 2),
         "p95_ms": round(lat[max(0, int(0.95 * n) - 1)], 2),
+        "mean_occupancy": occupancy,
         "solo_ms": round(solo_ms, 2) if solo_ms is not None else None,
         "speedup_vs_solo": (round(solo_ms * n / (wall * 1e3), 2)
                             if solo_ms is not None else None),
