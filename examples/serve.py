@@ -113,8 +113,15 @@ def build_app(model: LSTMModel, device: torch.device, batcher=None):
 
     @app.get("/health")
     def health():
-        return {"ok": True, "device": str(device),
-                "vocab": model.cfg.vocab_size}
+        r = {"ok": True, "device": str(device),
+             "vocab": model.cfg.vocab_size,
+             "hidden": model.cfg.hidden_size,
+             "layers": model.cfg.num_layers,
+             "weights_version": getattr(model, "_weights_version", 0),
+             "continuous_batching": batcher is not None}
+        if batcher is not None:
+            r["engine"] = batcher.stats()
+        return r
 
     @app.post("/generate")
     def generate(req: GenRequest):

@@ -120,6 +120,9 @@ def test_serving_endpoint_with_batcher():
     try:
         app = build_app(m, torch.device("cpu"), batcher=eng)
         client = TestClient(app)
+        h = client.get("/health").json()
+        assert h["continuous_batching"] and "engine" in h
+        assert h["hidden"] == 24 and h["layers"] == 2
         r = client.post("/generate", json={"tokens": [1, 2, 3], "steps": 5})
         assert r.status_code == 200
         assert r.json()["tokens"] == _solo(m, [1, 2, 3], 5)
