@@ -75,3 +75,25 @@ def test_bench_dp2_contract():
     assert rec["n_gpus"] == 2
     assert rec["config"]["parallelism"] == "dp2"
     assert rec["value"] > 0 and rec["ms_per_step"] > 0
+
+
+def test_train_cli_auto_resume(tmp_path):
+    """--auto-resume: a relaunched job picks up the newest checkpoint in
+    ckpt_dir and continues (the failure-recovery entry point)."""
+    base = ["--preset", "plumbing-cpu", "--set", "train.batch_size=4",
+            "--set", "train.seq_len=8", "--set", "train.log_interval=1",
+            "--set", f"train.ckpt_dir={tmp_path}"]
+    r1 = subprocess.run(
+        [sys.executable, os.path.join(REPO, "train.py"),
+         "--set", "train.steps=2"] + base,
+        capture_output=True, text=True, timeout=300, cwd=REPO)
+    assert r1.returncode == 0, r1.stderr[-2000:]
+    r2 = subprocess.run(
+        [sys.executable, os.path.join(REPO, "train.py"),
+         "--set", "train.steps=4", "--auto-resume"] + base,
+        capture_output=True, text=True, timeout=300, cwd=REPO)
+    assert r2.returncode == 0, r2.stderr[-2000:]
+    logs = r2.stdout + r2.stderr
+    assert "checkpoint loaded" in logs          # it resumed
+    assert '"step": 4' in logs                  # continued 3..4, not 1..2
+    assert '"step": 1"' not in logs

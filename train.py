@@ -23,6 +23,10 @@ def main():
     p.add_argument("--set", action="append", default=[], metavar="SEC.KEY=V",
                    help="config override, repeatable")
     p.add_argument("--resume", default=None, help="checkpoint to resume from")
+    p.add_argument("--auto-resume", action="store_true",
+                   help="resume from the newest checkpoint in "
+                        "train.ckpt_dir if one exists (the relaunch-after-"
+                        "failure entry point -- SURVEY.md §5 recovery row)")
     args = p.parse_args()
 
     if args.preset:
@@ -38,6 +42,8 @@ def main():
     trainer = Trainer(cfg, dist_ctx=ctx)
     if args.resume:
         trainer.load(args.resume)
+    elif args.auto_resume:
+        trainer.resume_latest()
     trainer.fit()
     cleanup()
 
