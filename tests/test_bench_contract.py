@@ -95,5 +95,6 @@ def test_train_cli_auto_resume(tmp_path):
     assert r2.returncode == 0, r2.stderr[-2000:]
     logs = r2.stdout + r2.stderr
     assert "checkpoint loaded" in logs          # it resumed
-    assert '"step": 4' in logs                  # continued 3..4, not 1..2
-    assert '"step": 1"' not in logs
+    assert '"step": 4' in logs                  # ran 3..4 (total budget 4)
+    assert '"step": 5' not in logs              # ...and STOPPED at 4
+    assert '"step": 1}' not in logs and '"step": 1,' not in logs

@@ -44,7 +44,10 @@ def main():
         trainer.load(args.resume)
     elif args.auto_resume:
         trainer.resume_latest()
-    trainer.fit()
+    # train.steps is the TOTAL budget (the LR schedule decays over it),
+    # so a resumed run finishes the remaining steps rather than adding
+    # a whole new budget on top
+    trainer.fit(max(0, cfg.train.steps - trainer.step))
     cleanup()
 
 
