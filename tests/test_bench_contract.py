@@ -98,3 +98,29 @@ def test_train_cli_auto_resume(tmp_path):
     assert '"step": 4' in logs                  # ran 3..4 (total budget 4)
     assert '"step": 5' not in logs              # ...and STOPPED at 4
     assert '"step": 1}' not in logs and '"step": 1,' not in logs
+
+
+def test_train_cli_crash_and_auto_resume_recovery(tmp_path):
+    """Full recovery story: a run crashes mid-training (injected fault),
+    the relaunch with --auto-resume picks up the last periodic checkpoint
+    and finishes the budget."""
+    base = ["--preset", "plumbing-cpu", "--set", "train.batch_size=4",
+            "--set", "train.seq_len=8", "--set", "train.log_interval=1",
+            "--set", "train.steps=6", "--set", "train.ckpt_interval=2",
+            "--set", f"train.ckpt_dir={tmp_path}"]
+    crash = subprocess.run(
+        [sys.executable, os.path.join(REPO, "train.py"),
+         "--set", "train.fault_step=4"] + base,
+        capture_output=True, text=True, timeout=300, cwd=REPO)
+    assert crash.returncode != 0                    # it really died
+    assert "injected fault" in crash.stderr
+    assert os.path.exists(os.path.join(str(tmp_path), "ckpt_step4.pt"))
+    relaunch = subprocess.run(
+        [sys.executable, os.path.join(REPO, "train.py"),
+         "--auto-resume"] + base,
+        capture_output=True, text=True, timeout=300, cwd=REPO)
+    assert relaunch.returncode == 0, relaunch.stderr[-2000:]
+    logs = relaunch.stdout + relaunch.stderr
+    assert "checkpoint loaded" in logs
+    assert '"step": 6' in logs                      # budget completed
+    assert os.path.exists(os.path.join(str(tmp_path), "ckpt_final.pt"))
