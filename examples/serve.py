@@ -98,16 +98,21 @@ def build_app(model: LSTMModel, device: torch.device, batcher=None):
 
         def events():
             sent = len(req.tokens)
-            while True:
-                toks, done = batcher.peek(rid)
-                if len(toks) > sent:
-                    yield f"data: {_json.dumps(toks[sent:])}\n\n"
-                    sent = len(toks)
-                if done:
-                    batcher.try_result(rid)
-                    yield "data: [DONE]\n\n"
-                    return
-                time.sleep(0.003)
+            try:
+                while True:
+                    toks, done = batcher.peek(rid)
+                    if len(toks) > sent:
+                        yield f"data: {_json.dumps(toks[sent:])}\n\n"
+                        sent = len(toks)
+                    if done:
+                        batcher.try_result(rid)
+                        yield "data: [DONE]\n\n"
+                        return
+                    time.sleep(0.003)
+            finally:
+                # client gone (GeneratorExit) or stream finished: free the
+                # batch slot either way
+                batcher.cancel(rid)
 
         return StreamingResponse(events(), media_type="text/event-stream")
 

@@ -87,6 +87,7 @@ class ContinuousBatcher:
         self._running = False
         self._decode_steps = 0      # batched decode steps executed
         self._slot_steps = 0        # sum of active slots over those steps
+        self._cancelled: set = set()  # rids retired by cancel()
 
     # ------------------------------------------------------------ requests
     def submit(self, tokens: List[int], steps: int,
@@ -129,6 +130,24 @@ class ContinuousBatcher:
                 if s.rid == rid:
                     return list(s.tokens), False
         raise KeyError(f"unknown request id {rid}")
+
+    def cancel(self, rid: int) -> bool:
+        """Abort a request: dequeue it, retire its live slot at the next
+        step boundary, or drop its unclaimed result.  Returns True if the
+        rid was known.  (The SSE endpoint calls this when the client
+        disconnects so abandoned generations stop consuming batch slots.)
+        """
+        with self._cv:
+            for i, s in enumerate(self._queue):
+                if s.rid == rid:
+                    del self._queue[i]
+                    return True
+            for s in self._slots:
+                if s.rid == rid:
+                    s.remaining = 0     # retires at the next step boundary
+                    self._cancelled.add(rid)
+                    return True
+            return self._results.pop(rid, None) is not None
 
     def try_result(self, rid: int) -> Optional[List[int]]:
         with self._cv:
@@ -210,7 +229,10 @@ class ContinuousBatcher:
             return
         for i in done:
             s = self._slots[i]
-            self._results[s.rid] = s.tokens
+            if s.rid in self._cancelled:
+                self._cancelled.discard(s.rid)   # drop, don't store
+            else:
+                self._results[s.rid] = s.tokens
         if keep:
             idx = torch.tensor(keep, device=self.device)
             self._state = [(h.index_select(0, idx), c.index_select(0, idx))

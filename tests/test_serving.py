@@ -249,3 +249,26 @@ def test_batcher_bench_tool_degenerate():
               "--requests", "6", "--solo-baseline", "1"])
     assert r["requests"] == 6 and r["req_per_s"] > 0
     assert r["p50_ms"] > 0 and r["solo_ms"] > 0
+
+
+def test_cancel_queued_live_and_finished():
+    """cancel(): dequeues a waiting request, retires a live one at the
+    next step, drops an unclaimed result -- neighbors unaffected."""
+    m = _model(seed=22)
+    eng = ContinuousBatcher(m, max_batch=2)
+    ra = eng.submit([1, 2], 8)
+    rb = eng.submit([3], 6)
+    rq = eng.submit([5], 4)            # queued (max_batch=2)
+    eng.step()
+    assert eng.cancel(rq)              # still queued: dequeued
+    eng.step()
+    assert eng.cancel(ra)              # live: retires next boundary
+    eng.drain()
+    assert eng.try_result(ra) is None  # cancelled -> no result stored
+    assert eng.try_result(rq) is None
+    assert eng.result(rb, timeout=0) == _solo(m, [3], 6)  # unaffected
+    rc = eng.submit([7], 2)
+    eng.drain()
+    assert eng.cancel(rc)              # finished: unclaimed result dropped
+    assert eng.try_result(rc) is None
+    assert not eng.cancel(12345)       # unknown rid
