@@ -637,3 +637,34 @@ def test_finetune_example_end_to_end(tmp_path):
     assert r["eval_loss"] == r["eval_loss"] and "eval_acc" in r
     import os as _os
     assert _os.path.exists(str(tmp_path / "ck" / "ckpt_final.pt"))
+
+
+def test_all_features_compose(tmp_path):
+    """Composition regression: lm_file + held-out split + bptt + accum +
+    cosine/warmup + dropout + periodic eval/ckpt in ONE run must train
+    and land the LR on min_lr."""
+    import numpy as np
+    from lstm_tensorspark_amd.config import DataConfig
+
+    (np.arange(32768) % 97).astype("uint16").tofile(str(tmp_path / "c.bin"))
+    cfg = Config(
+        model=ModelConfig(input_size=48, hidden_size=48, num_layers=2,
+                          head="lm", vocab_size=128, dropout=0.1),
+        train=TrainConfig(batch_size=8, seq_len=32, bptt=16, lr=0.5,
+                          momentum=0.9, grad_clip=1.0, steps=150,
+                          accum_steps=2, lr_schedule="cosine",
+                          warmup_steps=10, min_lr=0.05, eval_interval=50,
+                          eval_steps=2, ckpt_interval=100,
+                          ckpt_dir=str(tmp_path / "ck"), log_interval=0,
+                          seed=5),
+        data=DataConfig(task="lm_file", path=str(tmp_path / "c.bin"),
+                        eval_frac=0.1),
+    )
+    t = Trainer(cfg)
+    t.fit()
+    evals = [x for x in t.history if "eval_loss" in x]
+    assert len(evals) == 3
+    assert evals[-1]["eval_loss"] < evals[0]["eval_loss"]   # learning
+    assert abs(t.optimizer.lr - 0.05) < 1e-9                # schedule end
+    import os as _os
+    assert _os.path.exists(str(tmp_path / "ck" / "ckpt_final.pt"))
