@@ -133,3 +133,32 @@ def test_lm_file_windows_always_in_bounds(tmp_path_factory, n, t, b, step,
     assert x.shape == (t, b) and y.shape == (t, b)
     assert 0 <= int(x.min()) and int(x.max()) < 256
     assert torch.equal(x[1:], y[:-1])
+
+
+@given(lr=st.floats(1e-4, 10.0), min_lr_frac=st.floats(0.0, 0.9),
+       warmup=st.integers(0, 50), total=st.integers(1, 500),
+       kind=st.sampled_from(["cosine", "linear"]))
+@settings(max_examples=40, deadline=None)
+def test_lr_schedule_invariants(lr, min_lr_frac, warmup, total, kind):
+    """Any schedule config: values stay in [min(min_lr, warmup ramp), lr],
+    warmup is non-decreasing, decay is non-increasing, and the budget
+    boundary lands exactly on min_lr."""
+    from lstm_tensorspark_amd.config import Config, ModelConfig, TrainConfig
+    from lstm_tensorspark_amd.trainer import Trainer
+    min_lr = lr * min_lr_frac
+    cfg = Config(model=ModelConfig(input_size=8, hidden_size=8,
+                                   num_classes=4),
+                 train=TrainConfig(batch_size=2, seq_len=4, lr=lr,
+                                   min_lr=min_lr, warmup_steps=warmup,
+                                   lr_schedule=kind, steps=total,
+                                   log_interval=0))
+    t = Trainer(cfg)
+    vals = [t._lr_at(s) for s in range(0, max(total, warmup) + 2)]
+    assert all(0.0 <= v <= lr + 1e-12 for v in vals)
+    for a, b in zip(vals[:warmup], vals[1:warmup]):
+        assert a <= b + 1e-12                      # warmup non-decreasing
+    decay = vals[warmup:]
+    for a, b in zip(decay, decay[1:]):
+        assert b <= a + 1e-9 * lr                  # decay non-increasing
+    if total > warmup:
+        assert abs(t._lr_at(total) - min_lr) <= 1e-9 * max(lr, 1.0)
