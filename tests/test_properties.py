@@ -162,3 +162,44 @@ def test_lr_schedule_invariants(lr, min_lr_frac, warmup, total, kind):
         assert b <= a + 1e-9 * lr                  # decay non-increasing
     if total > warmup:
         assert abs(t._lr_at(total) - min_lr) <= 1e-9 * max(lr, 1.0)
+
+
+@given(accum=st.integers(1, 4), bptt=st.sampled_from([0, 4, 8]),
+       seed=st.integers(0, 100))
+@settings(max_examples=12, deadline=None)
+def test_accum_bptt_grad_scale_invariant(accum, bptt, seed):
+    """For any (accum_steps, bptt) combination the accumulated grad must
+    equal the mean of the per-micro-batch full-window grads (the 1/n and
+    1/windows scalings must compose correctly)."""
+    from lstm_tensorspark_amd.config import (Config, ModelConfig,
+                                             TrainConfig, DataConfig)
+    from lstm_tensorspark_amd.trainer import Trainer
+    def cfg():
+        return Config(
+            model=ModelConfig(input_size=6, hidden_size=8, num_classes=5),
+            train=TrainConfig(batch_size=3, seq_len=8, bptt=bptt,
+                              momentum=0.0, lr=0.1, seed=seed,
+                              log_interval=0),
+            data=DataConfig(task="copy"),
+        )
+    tA = Trainer(cfg())
+    batches = [tA.data.batch(i) for i in range(accum)]
+    tA.train_step(batches if accum > 1 else batches[0])
+    gA = {n: p.grad.detach().clone()
+          for n, p in zip(tA.model.weight_dict(), tA.model.parameters())}
+
+    tB = Trainer(cfg())
+    acc = {}
+    for b in batches:
+        tB.optimizer.zero_grad()
+        tB.train_step(b)                      # full-window single batch
+        for n, p in zip(tB.model.weight_dict(), tB.model.parameters()):
+            acc[n] = acc.get(n, 0) + p.grad.detach().clone() / accum
+        # undo the weight update so every micro-batch sees the same weights
+        tB2 = Trainer(cfg())
+        for p, q in zip(tB.model.parameters(), tB2.model.parameters()):
+            with torch.no_grad():
+                p.copy_(q)
+        tB.optimizer.sync_masters()
+    for n in gA:
+        assert torch.allclose(gA[n], acc[n], atol=2e-6, rtol=1e-4), n
