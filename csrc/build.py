@@ -56,6 +56,11 @@ def digest() -> str:
         with open(os.path.join(CSRC, s), "rb") as f:
             h.update(f.read())
     h.update(ARCH.encode())
+    # build flavor is part of the identity: otherwise a debug/ASAN .so
+    # would pass the up-to-date check for a release build (and a debug
+    # session would silently reuse the release .so)
+    h.update(os.environ.get("LSTM_TS_BUILD_MODE", "release").encode())
+    h.update(os.environ.get("LSTM_TS_ASAN", "0").encode())
     return h.hexdigest()
 
 
