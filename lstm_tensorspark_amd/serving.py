@@ -12,10 +12,12 @@ requests joining and leaving mid-flight:
 
 Everything is a row operation on the engine's native state layout (a
 list of (h, c) tensors with batch as dim 0), so no request ever pays for
-another's prompt or waits for the batch to drain.  Greedy-only by
-design: batched continuous decode must be REPRODUCIBLE per request --
-the test oracle is exact token-for-token parity with a solo
-`model.generate()` call.
+another's prompt or waits for the batch to drain.  Per-request decoding
+knobs (temperature/top-k/top-p with an OWN seeded generator, EOS
+early-stop) keep every request REPRODUCIBLE regardless of batch
+composition -- the test oracle is exact token-for-token parity with a
+solo `model.generate()` call, for greedy and seeded-sampled requests
+alike.
 
 This engine targets multi-tenant serving latency; the fixed-batch
 hipGraph decode (`model.generate(use_graph=True)`) remains the peak
