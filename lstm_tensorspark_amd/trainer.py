@@ -13,6 +13,11 @@ Truncated BPTT (SURVEY.md §5 long-context row): when train.bptt > 0 and
 < seq_len, each batch is streamed in windows of bptt steps with the (h, c)
 carry detached between windows -- activation memory is O(bptt*B*H), not
 O(T*B*H).
+
+Composable knobs on top (docs/TRAINING.md has the full contracts):
+gradient accumulation (accum_steps), LR warmup + cosine/linear decay,
+periodic held-out eval (eval_interval), auto-resume, fault injection,
+opt-in whole-step hipGraph capture.
 """
 from __future__ import annotations
 
