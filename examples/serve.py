@@ -9,9 +9,12 @@ see README "Serving / decode") behind a FastAPI endpoint:
          -d '{"tokens": [1, 2, 3], "steps": 16}'
 
 With --demo (no checkpoint) it serves a random-init model -- useful for
-latency/throughput testing of the serving stack itself.  Single-GPU,
-single-worker: the decode step is already graph-replayed, so the Python
-layer only pays tensor marshalling per request.
+latency/throughput testing of the serving stack itself.  Endpoints:
+POST /generate (greedy/temperature/top-k/top-p/beam/EOS), POST
+/generate_stream (SSE token stream; needs --continuous), GET /health,
+GET /metrics (Prometheus).  --continuous turns on the continuous
+batcher (lstm_tensorspark_amd/serving.py): concurrent requests share
+batched decode steps.  .pt and .safetensors checkpoints both serve.
 
 (No `from __future__ import annotations` here: FastAPI resolves the
 request-model type hints at runtime, and postponed annotations of a
